@@ -1,0 +1,195 @@
+"""Differentiable ProGen ops with CPU-reference / HIP-kernel dispatch.
+
+GPU path: hand-written CDNA4 HIP kernels (progen_amd/ops/hip/) via the
+in-tree extension ``progen_amd._C`` — mandatory on GPU (no silent eager
+fallback; see ops/dispatch.py). Plain GEMMs (QKV/FF/out projections, the
+SGU spatial matmul) go through torch.matmul = hipBLASLt, which is the
+library-GEMM path, not a compatibility layer.
+
+CPU path: the pure-PyTorch fp32 reference (ops/reference.py), which is
+also the numerics oracle for the kernels.
+"""
+
+from __future__ import annotations
+
+from typing import Optional, Tuple
+
+import torch
+
+from . import dispatch, reference
+
+
+# ---------------------------------------------------------------------------
+# fused LayerNorm(scale-only) + token shift
+# ---------------------------------------------------------------------------
+
+class _LnShiftFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, shift: bool, eps: float):
+        C = dispatch.ext()
+        y, mean, rstd = C.ln_shift_fwd(x, weight, bool(shift), float(eps))
+        ctx.save_for_backward(x, weight, mean, rstd)
+        ctx.shift = bool(shift)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        x, weight, mean, rstd = ctx.saved_tensors
+        C = dispatch.ext()
+        dx, dweight = C.ln_shift_bwd(dy.contiguous(), x, weight, mean, rstd, ctx.shift)
+        return dx, dweight, None, None
+
+
+def ln_shift(x: torch.Tensor, weight: torch.Tensor, shift: bool = True,
+             eps: float = 1e-5) -> torch.Tensor:
+    """LN (scale-only, reference: progen.py:22) + optional token shift
+    (reference: progen.py:43-46), fused on GPU."""
+    if dispatch.use_hip(x):
+        return _LnShiftFn.apply(x, weight, shift, eps)
+    return reference.ln_shift(x, weight, shift, eps)
+
+
+# ---------------------------------------------------------------------------
+# fused local window attention (rotary + window + softmax + AV)
+# ---------------------------------------------------------------------------
+
+class _LocalAttnFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, qkv, sin, cos, heads: int, window_size: int):
+        C = dispatch.ext()
+        out, lse = C.attn_fwd(qkv, sin, cos, int(heads), int(window_size))
+        ctx.save_for_backward(qkv, sin, cos, out, lse)
+        ctx.heads = int(heads)
+        ctx.window_size = int(window_size)
+        return out
+
+    @staticmethod
+    def backward(ctx, dout):
+        qkv, sin, cos, out, lse = ctx.saved_tensors
+        C = dispatch.ext()
+        dqkv = C.attn_bwd(dout.contiguous(), qkv, sin, cos, out, lse,
+                          ctx.heads, ctx.window_size)
+        return dqkv, None, None, None, None
+
+
+def local_attention(qkv: torch.Tensor, sin: torch.Tensor, cos: torch.Tensor,
+                    heads: int, window_size: int) -> torch.Tensor:
+    """Fused windowed-causal attention core (reference: progen.py:83-103).
+
+    Applies interleaved rotary to q, k AND v (quirk, progen.py:87), windows
+    the sequence with one-window lookback (window 0's zero lookback keys
+    UNMASKED, progen.py:90-96), runs online-softmax attention on MFMA and
+    returns the merged (B, N, h*dh) context."""
+    if dispatch.use_hip(qkv):
+        return _LocalAttnFn.apply(qkv, sin, cos, heads, window_size)
+    return reference.local_attention(qkv, sin, cos, heads, window_size)
+
+
+# ---------------------------------------------------------------------------
+# GLU-GELU epilogue
+# ---------------------------------------------------------------------------
+
+class _GluGeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, h):
+        C = dispatch.ext()
+        y = C.glu_fwd(h)
+        ctx.save_for_backward(h)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (h,) = ctx.saved_tensors
+        C = dispatch.ext()
+        return C.glu_bwd(dy.contiguous(), h)
+
+
+def glu_gelu(h: torch.Tensor) -> torch.Tensor:
+    """x, gate = split(h, 2); x * gelu(gate)  (reference: progen.py:139-141)."""
+    if dispatch.use_hip(h):
+        return _GluGeluFn.apply(h)
+    return reference.glu_gelu(h)
+
+
+class _GeluFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, h):
+        C = dispatch.ext()
+        y = C.gelu_fwd(h)
+        ctx.save_for_backward(h)
+        return y
+
+    @staticmethod
+    def backward(ctx, dy):
+        (h,) = ctx.saved_tensors
+        C = dispatch.ext()
+        return C.gelu_bwd(dy.contiguous(), h)
+
+
+def gelu(h: torch.Tensor) -> torch.Tensor:
+    if dispatch.use_hip(h):
+        return _GeluFn.apply(h)
+    return reference.gelu(h)
+
+
+# ---------------------------------------------------------------------------
+# masked cross-entropy (fused log_softmax + NLL gather; mask math on host)
+# ---------------------------------------------------------------------------
+
+class _CERowFn(torch.autograd.Function):
+    """Per-row fused -log_softmax(logits)[target]; returns nll (B, N) fp32.
+
+    The EOS/pad mask reduction (reference: utils.py:54-59) is cheap
+    (B, N)-shaped tensor math and stays in torch; the (B, N, V) work is
+    the kernel."""
+
+    @staticmethod
+    def forward(ctx, logits, targets):
+        C = dispatch.ext()
+        nll, lse = C.ce_fwd(logits, targets)
+        ctx.save_for_backward(logits, targets, lse)
+        return nll
+
+    @staticmethod
+    def backward(ctx, dnll):
+        logits, targets, lse = ctx.saved_tensors
+        C = dispatch.ext()
+        dlogits = C.ce_bwd(dnll.contiguous(), logits, targets, lse)
+        return dlogits, None
+
+
+def cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
+                  ignore_index: int = 0) -> torch.Tensor:
+    """Masked CE with first-pad-as-EOS (reference: utils.py:45-59); the
+    per-sequence masked mean then batch mean reduction order is preserved
+    (reference: utils.py:67,75-76)."""
+    if dispatch.use_hip(logits):
+        nll = _CERowFn.apply(logits, targets)
+        mask = targets != ignore_index
+        eos_mask = (~mask).long().cumsum(dim=-1) == 1
+        mask = (mask | eos_mask).to(nll.dtype)
+        ce_per_seq = (nll * mask).sum(dim=-1) / mask.sum(dim=-1)
+        return ce_per_seq.mean()
+    return reference.cross_entropy(logits, targets, ignore_index)
+
+
+# ---------------------------------------------------------------------------
+# SGU spatial gating
+# ---------------------------------------------------------------------------
+
+def sgu_gate(x: torch.Tensor, norm_weight: torch.Tensor,
+             spatial_weights: torch.Tensor, spatial_biases: torch.Tensor,
+             eps: float = 1e-5) -> torch.Tensor:
+    """gMLP spatial gating unit core (reference: progen.py:166-183).
+
+    GPU path: fused LN kernel on the gate half + causal (n, n) spatial
+    matmul on hipBLASLt (library GEMM) + elementwise gate-multiply. The
+    tril mask is applied to the weight once per forward (cheap: n^2)."""
+    if dispatch.use_hip(x):
+        xa, gate = x.chunk(2, dim=-1)
+        gate = ln_shift(gate.contiguous(), norm_weight, shift=False, eps=eps)
+        n = x.shape[1]
+        w = spatial_weights[:n, :n].tril().to(gate.dtype)
+        gate = torch.einsum("bnd,mn->bmd", gate, w) + spatial_biases[:n].to(gate.dtype)
+        return xa * gate
+    return reference.sgu_gate(x, norm_weight, spatial_weights, spatial_biases, eps)

@@ -1,0 +1,128 @@
+"""Training math and sampling (reference: progen_transformer/utils.py).
+
+get_loss_fn / cross_entropy / sample / select_top_k with the reference's
+semantics preserved; batching is native (torch batch dim) instead of
+vmap, and data parallelism is explicit RCCL (progen_amd/parallel/) instead
+of the reference's pmap (utils.py:70).
+"""
+
+from __future__ import annotations
+
+import os
+import shutil
+from typing import Callable, Optional, Tuple
+
+import torch
+
+from .ops import functional as OF
+from .ops import reference as R
+
+
+# -- fs helpers (reference: utils.py:23-38) ---------------------------------
+
+def exists(v) -> bool:
+    return v is not None
+
+
+def confirm(question: str) -> bool:
+    while True:
+        resp = input(f"{question} (y/n) ").lower()
+        if resp in ("y", "n"):
+            return resp == "y"
+
+
+def clear_directory_(path) -> None:
+    shutil.rmtree(str(path), ignore_errors=True)
+    path.mkdir(exist_ok=True, parents=True)
+
+
+def silentremove(filename) -> None:
+    try:
+        os.remove(filename)
+    except OSError:
+        pass
+
+
+# -- loss --------------------------------------------------------------------
+
+masked_mean = R.masked_mean
+cross_entropy = OF.cross_entropy
+
+
+def compute_loss(model: torch.nn.Module, data: torch.Tensor) -> torch.Tensor:
+    """One training loss on a (B, seq_len+1) int batch.
+
+    ids = data[:, :-1], labels = data[:, 1:]  (reference: utils.py:63);
+    per-sequence masked CE then batch mean (reference: utils.py:45-59,67).
+    """
+    ids, labels = data[:, :-1], data[:, 1:]
+    logits = model(ids)
+    return cross_entropy(logits, labels)
+
+
+def get_loss_fn(model: torch.nn.Module, data_parallel: bool = False) -> Callable:
+    """Returns loss_fn(data) -> (loss, None); gradients are produced by
+    loss.backward() (torch autograd) rather than returned — the explicit
+    analog of the reference's value_and_grad (utils.py:61-93). Under data
+    parallelism the gradient all-reduce is performed by the DDP wrapper
+    (progen_amd/parallel/ddp.py), not here."""
+
+    def loss_fn(data: torch.Tensor) -> torch.Tensor:
+        return compute_loss(model, data)
+
+    return loss_fn
+
+
+# -- sampling (reference: utils.py:97-135) -----------------------------------
+
+select_top_k = R.select_top_k
+gumbel_noise = R.gumbel_noise
+
+
+@torch.no_grad()
+def sample(
+    fn: Callable[[torch.Tensor], torch.Tensor],
+    prime: torch.Tensor,
+    length: int,
+    top_k: Optional[int] = None,
+    add_bos: bool = False,
+    generator: Optional[torch.Generator] = None,
+    device=None,
+) -> torch.Tensor:
+    """Gumbel-max top-k autoregressive decoding, reference semantics
+    (reference: utils.py:106-135):
+
+      - the sequence is padded to full ``length`` and every step runs a
+        full-length forward (no KV cache) — parity path;
+      - top-k uses a strict `>` mask vs the k-th value and sets excluded
+        logits to 0, not -inf (utils.py:97-100);
+      - everything after the second pad/EOS token is zeroed (utils.py:132-133).
+
+    fn: (n,) int64 tensor -> (n, V) logits (e.g. a closure over
+    TransformedProGen.apply or the bare module).
+    """
+    prime = torch.as_tensor(prime, device=device).long().flatten()
+    start_pos = prime.shape[-1]
+    pad = (0, length - start_pos) if not add_bos else (1, length - start_pos - 1)
+    seq = torch.nn.functional.pad(prime, pad)
+    if add_bos:
+        start_pos += 1
+
+    for curr_pos in range(start_pos, length):
+        logits = fn(seq)
+        logits = logits[curr_pos - 1].float()
+
+        noise = gumbel_noise(logits.shape, generator=generator,
+                             device=logits.device)
+
+        if top_k is not None:
+            mask, logits = select_top_k(logits, top_k)
+            noise = noise * mask
+
+        sampled = (logits + noise).argmax(dim=-1)
+        seq[curr_pos] = sampled
+
+    # zero after 2nd pad token (the 1st learned pad acts as EOS)
+    remove_after_eos = (seq == 0).long().cumsum(dim=-1) > 1
+    seq = seq * (~remove_after_eos).long()
+    return seq

@@ -1,0 +1,67 @@
+"""Sampler semantics tests (reference: utils.py:97-135)."""
+
+import numpy as np
+import torch
+
+from progen_amd import ProGen, ProGenBase, ProGenConfig
+from progen_amd.utils import sample
+
+TINY = dict(num_tokens=32, dim=16, seq_len=32, depth=2, window_size=8,
+            global_mlp_depth=1, heads=2, dim_head=8)
+
+
+def _fn(module):
+    def fn(seq):
+        with torch.no_grad():
+            return module(seq)[0]
+    return fn
+
+
+def test_sample_shapes_and_prime_preserved():
+    m = ProGenBase(ProGenConfig(**TINY))
+    prime = torch.tensor([5, 6, 7])
+    g = torch.Generator().manual_seed(0)
+    out = sample(_fn(m), prime, length=32, top_k=5, generator=g)
+    assert out.shape == (32,)
+    assert out[:3].tolist() == [5, 6, 7]
+
+
+def test_sample_add_bos():
+    m = ProGenBase(ProGenConfig(**TINY))
+    prime = torch.tensor([5, 6, 7])
+    g = torch.Generator().manual_seed(0)
+    out = sample(_fn(m), prime, length=32, top_k=5, add_bos=True, generator=g)
+    assert out[0].item() == 0      # BOS
+    assert out[1:4].tolist() == [5, 6, 7]  # prime intact (fixes the
+    # reference's add_bos off-by-one that adds the first sample onto the
+    # last prime token, utils.py:110-116)
+
+
+def test_sample_truncates_after_second_pad():
+    """Everything after the 2nd pad/EOS is zeroed (utils.py:132-133)."""
+    calls = {"n": 0}
+
+    def fn(seq):
+        # deterministic fake model: always put all mass on token 0 (pad)
+        logits = torch.full((seq.shape[0], 32), -100.0)
+        logits[:, 0] = 100.0
+        return logits
+
+    prime = torch.tensor([5, 6])
+    out = sample(fn, prime, length=10, top_k=1)
+    assert out[:2].tolist() == [5, 6]
+    assert (out[2:] == 0).all()
+
+
+def test_sample_respects_top_k():
+    def fn(seq):
+        logits = torch.zeros((seq.shape[0], 32))
+        logits[:, 3] = 5.0
+        logits[:, 4] = 4.0
+        logits[:, 5] = 3.0
+        return logits
+
+    g = torch.Generator().manual_seed(0)
+    out = sample(fn, torch.tensor([1]), length=16, top_k=2, generator=g)
+    # with top_k=2 only tokens {3, 4} may appear after the prime
+    assert set(out[1:].tolist()) <= {3, 4}
