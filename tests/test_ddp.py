@@ -1,0 +1,124 @@
+"""Multi-process DP tests over gloo (world_size=2, CPU).
+
+Verifies the bucketed all-reduce path gives the same gradients/params as
+single-process training on the combined batch — the DP-loss-parity test
+recommended by SURVEY.md §4.
+"""
+
+import os
+
+import numpy as np
+import pytest
+import torch
+import torch.distributed as dist
+import torch.multiprocessing as mp
+
+from progen_amd import ProGenBase, ProGenConfig
+from progen_amd.optim import ProGenAdamW
+from progen_amd.parallel.ddp import DistributedTrainer
+from progen_amd.utils import compute_loss
+
+TINY = dict(num_tokens=32, dim=16, seq_len=32, depth=2, window_size=8,
+            global_mlp_depth=1, heads=2, dim_head=8)
+
+
+def _worker(rank, world, port, result_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    os.environ["WORLD_SIZE"] = str(world)
+    os.environ["RANK"] = str(rank)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+
+    torch.manual_seed(0)
+    model = ProGenBase(ProGenConfig(**TINY))
+    opt = ProGenAdamW(model, max_grad_norm=None, grad_accum_every=2)
+    ddp = DistributedTrainer(opt.space, bucket_mb=0.01)  # tiny buckets
+
+    torch.manual_seed(1000)
+    all_data = torch.randint(0, 32, (4 * world, 33))
+
+    # 2 micro-batches of 2 seqs per rank
+    for micro in range(2):
+        lo = micro * 2 * world + rank * 2
+        data = all_data[lo:lo + 2]
+        if micro < 1:
+            with ddp.no_sync():
+                compute_loss(model, data).backward()
+        else:
+            compute_loss(model, data).backward()
+            ddp.finish_backward()
+        opt.micro_step()
+
+    if rank == 0:
+        result_q.put({
+            "params": opt.master.detach().clone(),
+            "grad": opt.space.flat_grad.detach().clone(),
+        })
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_dp2_matches_single_process():
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    port = 29511
+    procs = [ctx.Process(target=_worker, args=(r, world, port, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    got = q.get()
+    for p in procs:
+        p.join(120)
+        assert p.exitcode == 0
+
+    # single-process oracle on the full batch
+    torch.manual_seed(0)
+    model = ProGenBase(ProGenConfig(**TINY))
+    opt = ProGenAdamW(model, max_grad_norm=None, grad_accum_every=2)
+    torch.manual_seed(1000)
+    all_data = torch.randint(0, 32, (4 * world, 33))
+    for micro in range(2):
+        lo = micro * 2 * world
+        data = all_data[lo:lo + 2 * world]
+        compute_loss(model, data).backward()
+        opt.micro_step()
+
+    np.testing.assert_allclose(got["params"].numpy(), opt.master.numpy(),
+                               atol=1e-5)
+
+
+def _worker_bucket_overlap(rank, world, port, result_q):
+    os.environ["MASTER_ADDR"] = "127.0.0.1"
+    os.environ["MASTER_PORT"] = str(port)
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    torch.manual_seed(0)
+    model = ProGenBase(ProGenConfig(**TINY))
+    opt = ProGenAdamW(model, max_grad_norm=None)
+    ddp = DistributedTrainer(opt.space, bucket_mb=0.005)
+    assert len(ddp.buckets) > 1  # actually bucketed
+    # rank-dependent grads: after reduce every rank holds the mean
+    data = torch.randint(0, 32, (1 + rank, 33))
+    compute_loss(model, data).backward()
+    ddp.finish_backward()
+    result_q.put((rank, opt.space.flat_grad.clone()))
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_grads_identical_across_ranks():
+    world = 2
+    ctx = mp.get_context("spawn")
+    q = ctx.SimpleQueue()
+    procs = [ctx.Process(target=_worker_bucket_overlap, args=(r, world, 29512, q))
+             for r in range(world)]
+    for p in procs:
+        p.start()
+    res = {}
+    for _ in range(world):
+        r, g = q.get()
+        res[r] = g
+    for p in procs:
+        p.join(120)
+        assert p.exitcode == 0
+    np.testing.assert_allclose(res[0].numpy(), res[1].numpy(), atol=1e-7)
