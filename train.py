@@ -1,0 +1,252 @@
+"""Training CLI (reference parity: /root/reference/train.py).
+
+Flag surface mirrors the reference (train.py:36-57). Differences, by
+design (MI355X-native):
+  - data parallelism: launch with torchrun (one process per GPU, RCCL
+    over xGMI); the --data_parallel flag is accepted and implied by
+    WORLD_SIZE>1 (the reference used jax.pmap, train.py:42 / utils.py:70);
+  - --mixed_precision selects bf16 compute (MFMA-native) instead of the
+    reference's jmp fp16 policy;
+  - --grad_accum_mode {sum,apply_every} exposes the reference's
+    apply_every optimizer quirk (train.py:117-121) behind a flag,
+    defaulting to standard summed accumulation;
+  - wandb is optional (not installed in the offline image): --wandb_off
+    or a missing wandb package degrade to stdout logging.
+"""
+
+import os
+import time
+from pathlib import Path
+
+import click
+import numpy as np
+import torch
+
+try:
+    import tomllib
+except ModuleNotFoundError:
+    import tomli as tomllib
+
+from progen_amd import ProGenBase, ProGenConfig
+from progen_amd.checkpoint import get_checkpoint_fns, numpy_to_tensors, tensors_to_numpy
+from progen_amd.data import decode_tokens, iterator_from_tfrecords_folder
+from progen_amd.optim import ProGenAdamW
+from progen_amd.parallel import DistributedTrainer, init_distributed, is_distributed
+from progen_amd.utils import compute_loss, confirm, exists, sample
+
+
+def _wandb(wandb_off):
+    if wandb_off:
+        return None
+    try:
+        import wandb  # type: ignore
+        return wandb
+    except ImportError:
+        return None
+
+
+@click.command()
+@click.option('--seed', default=42)
+@click.option('--batch_size', default=4)
+@click.option('--grad_accum_every', default=4)
+@click.option('--learning_rate', default=2e-4)
+@click.option('--weight_decay', default=1e-3)
+@click.option('--data_parallel', default=False, is_flag=True)
+@click.option('--max_grad_norm', default=0.5)
+@click.option('--validate_every', default=100)
+@click.option('--sample_every', default=500)
+@click.option('--checkpoint_every', default=1000)
+@click.option('--checkpoint_path', default='./ckpts')
+@click.option('--checkpoint_keep_n', default=500)
+@click.option('--config_path', default='./configs/model')
+@click.option('--model_name', default='default')
+@click.option('--prime_length', default=25)
+@click.option('--seq_len', default=1024)
+@click.option('--mixed_precision', default=False, is_flag=True)
+@click.option('--grad_accum_mode', default='sum',
+              type=click.Choice(['sum', 'apply_every']))
+@click.option('--data_path', default='./train_data')
+@click.option('--wandb_off', default=False, is_flag=True)
+@click.option('--wandb_project_name', default='progen-training')
+@click.option('--new', default=False, is_flag=True)
+@click.option('--max_steps', default=0, help='stop after N effective batches (0 = full epoch)')
+@click.option('--yes', default=False, is_flag=True, help='skip the --new confirmation prompt')
+def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
+         data_parallel, max_grad_norm, validate_every, sample_every,
+         checkpoint_every, checkpoint_path, checkpoint_keep_n, config_path,
+         model_name, prime_length, seq_len, mixed_precision, grad_accum_mode,
+         data_path, wandb_off, wandb_project_name, new, max_steps, yes):
+    local_rank = init_distributed()
+    world = int(os.environ.get('WORLD_SIZE', '1'))
+    rank = int(os.environ.get('RANK', '0'))
+    is_main = rank == 0
+    device = torch.device('cuda', local_rank) if torch.cuda.is_available() \
+        else torch.device('cpu')
+
+    torch.manual_seed(seed + rank)
+
+    # checkpoints (reference: train.py:83-92)
+    reset_checkpoint, get_last_checkpoint, save_checkpoint = \
+        get_checkpoint_fns(checkpoint_path)
+    if new and is_main:
+        if not yes and not confirm('are you sure you want to clear all your '
+                                   'checkpoints and restart training?'):
+            raise SystemExit
+        reset_checkpoint()
+
+    last_checkpoint = get_last_checkpoint()
+
+    # model config: checkpoint beats toml (reference: train.py:95-100)
+    if not exists(last_checkpoint):
+        cfg_file = Path(config_path) / f'{model_name}.toml'
+        assert cfg_file.exists(), \
+            f'path to your model config {cfg_file} does not exist'
+        model_kwargs = tomllib.loads(cfg_file.read_text())
+    else:
+        model_kwargs = last_checkpoint['model_config']
+
+    cfg = ProGenConfig.from_dict(model_kwargs)
+    seq_len = cfg.seq_len
+
+    module = ProGenBase(cfg)
+    dtype = torch.bfloat16 if (mixed_precision and device.type == 'cuda') \
+        else torch.float32
+    module = module.to(device=device, dtype=dtype)
+    # rotary tables stay fp32 for accuracy
+    module.rotary_sin = module.rotary_sin.float()
+    module.rotary_cos = module.rotary_cos.float()
+
+    optim = ProGenAdamW(module, lr=learning_rate, weight_decay=weight_decay,
+                        max_grad_norm=max_grad_norm,
+                        accum_mode=grad_accum_mode,
+                        grad_accum_every=grad_accum_every)
+    ddp = DistributedTrainer(optim.space)
+
+    start_seq_index = 0
+    if exists(last_checkpoint):
+        module.load_state_dict({
+            k: torch.as_tensor(v).to(device=device)
+            for k, v in numpy_to_tensors(last_checkpoint['params']).items()
+        }, strict=False)  # copies in place -> flat param buffer updated
+        optim.master.copy_(optim.space.flat.float())
+        if exists(last_checkpoint.get('optim_state')):
+            optim.load_state_dict(numpy_to_tensors(last_checkpoint['optim_state']))
+        start_seq_index = last_checkpoint['next_seq_index']
+
+    num_params = module.num_params()
+
+    run_id = last_checkpoint.get('run_id') if exists(last_checkpoint) else None
+    wandb = _wandb(wandb_off) if is_main else None
+    if wandb is not None:
+        kwargs = {'id': run_id, 'resume': 'allow'} if exists(run_id) else {}
+        wandb.init(project=wandb_project_name, **kwargs)
+        wandb.config.num_params = num_params
+        run_id = wandb.run.id
+
+    # data (reference: train.py:153-172); each rank reads the same global
+    # batch and slices its shard (pmap-reshape parity, utils.py:89)
+    global_batch = batch_size * world
+    total_train_seqs, get_train_dataset = \
+        iterator_from_tfrecords_folder(data_path, data_type='train')
+    total_valid_seqs, get_valid_dataset = \
+        iterator_from_tfrecords_folder(data_path, data_type='valid')
+    assert total_train_seqs > 0, 'no protein sequences found for training'
+    assert total_valid_seqs > 0, 'no protein sequences found for validation'
+
+    train_dataset = get_train_dataset(seq_len=seq_len, batch_size=global_batch,
+                                      skip=start_seq_index)
+    valid_dataset = get_valid_dataset(seq_len=seq_len, batch_size=global_batch,
+                                      loop=True)
+
+    if is_main:
+        print(f'params: {num_params}')
+        print(f'sequence length: {seq_len}')
+        print(f'num sequences: {total_train_seqs}')
+        print(f'starting from sequence {start_seq_index}')
+
+    def my_shard(batch_np):
+        t = torch.from_numpy(batch_np.astype(np.int64))
+        lo = rank * batch_size
+        shard = t[lo:lo + batch_size]
+        if shard.shape[0] == 0:  # tail batch smaller than world*bs
+            shard = t[:1]
+        return shard.to(device)
+
+    effective_batch_size = global_batch * grad_accum_every
+    seq_index_ranges = range(start_seq_index, total_train_seqs,
+                             effective_batch_size)
+
+    step_times = []
+    for i, seq_index in enumerate(seq_index_ranges):
+        if max_steps and i >= max_steps:
+            break
+        t0 = time.perf_counter()
+        loss = None
+        for micro in range(grad_accum_every):
+            try:
+                data = my_shard(next(train_dataset))
+            except StopIteration:
+                break
+            last_micro = micro == grad_accum_every - 1
+            if last_micro:
+                loss = compute_loss(module, data)
+                loss.backward()
+                ddp.finish_backward()
+            else:
+                with ddp.no_sync():
+                    loss = compute_loss(module, data)
+                    loss.backward()
+            optim.micro_step()
+        if loss is None:
+            break
+        step_times.append(time.perf_counter() - t0)
+
+        # NOTE: like the reference (train.py:192), the logged loss is the
+        # last micro-batch's
+        loss_val = ddp.all_reduce_scalar(loss).item()
+        if is_main:
+            print(f'loss: {loss_val}')
+            if wandb is not None:
+                wandb.log({'loss': loss_val})
+
+        if i % checkpoint_every == 0 and is_main:
+            package = {
+                'next_seq_index': seq_index + effective_batch_size,
+                'params': tensors_to_numpy(
+                    {k: v for k, v in module.state_dict().items()}),
+                'optim_state': tensors_to_numpy(optim.state_dict()),
+                'model_config': model_kwargs,
+                'run_id': run_id,
+            }
+            save_checkpoint(package, checkpoint_keep_n)
+            print(f"checkpoint to start at sequence index of "
+                  f"{package['next_seq_index']}")
+
+        if i % validate_every == 0:
+            valid_data = my_shard(next(valid_dataset))
+            with torch.no_grad():
+                vloss = compute_loss(module, valid_data)
+            vloss_val = ddp.all_reduce_scalar(vloss).item()
+            if is_main:
+                print(f'valid_loss: {vloss_val}')
+                if wandb is not None:
+                    wandb.log({'valid_loss': vloss_val})
+
+        if i % sample_every == 0 and is_main:
+            valid_data = my_shard(next(valid_dataset))[0]
+            prime = valid_data[:prime_length]
+            prime_str = decode_tokens(prime.cpu().numpy())
+
+            def fwd(seq):
+                with torch.no_grad():
+                    return module(seq.to(device))[0].float().cpu()
+
+            sampled = sample(fwd, prime.cpu(), seq_len, top_k=25)
+            sampled_str = decode_tokens(sampled[prime_length:].numpy())
+            print(prime_str, '\n', '*' * 40, '\n', sampled_str)
+            if wandb is not None:
+                wandb.log({'samples': sampled_str})
+
+
+if __name__ == '__main__':
+    main()
