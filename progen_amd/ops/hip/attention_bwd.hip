@@ -1,0 +1,496 @@
+// Fused local-window attention, backward.
+//
+// Recompute-based flash backward of ops/hip/attention_fwd.hip (the
+// reference differentiates progen.py:83-103 through XLA; here the math
+// is explicit):
+//   D_i = rowsum(dO_i * O_i)
+//   P   = exp(S_masked - lse)          (S recomputed from q', scaled k')
+//   dV' = P^T dO
+//   dP  = dO V'^T
+//   dS  = P o (dP - D)
+//   dQ' = dS k_s                       (scale folded into staged K)
+//   dK' = dS^T q_s                     (scale folded into staged Q^T)
+// followed by attn_bwd_finalize_kernel, which applies the inverse rotary
+// rotation (rotary is linear, so it commutes with accumulation) and
+// casts the fp32 accumulator to bf16 dqkv.
+//
+// Geometry mirrors the forward: block = one (batch, head, window), 4
+// waves, each wave owns a 64-row q-chunk. dQ rows are exclusively owned
+// -> plain fp32 stores. dK/dV go through atomicAdd because adjacent
+// windows' key bands overlap (a key is seen by its own window AND the
+// next window's lookback, progen.py:90-91); window 0's lookback keys are
+// the zero pad and their grads are discarded.
+//
+// MFMA operand LDS images (all XOR-swizzled, byte ^= (row&7)<<4):
+//   k_lds  [key][dh] scaled k'   (S B-fragments)
+//   kt_lds [dh][key] scaled k'   (dQ B-fragments)
+//   v_lds  [key][dh] v'          (dP B-fragments)
+//   qt_lds [dh][row] scaled q'   (dK B-fragments, per wave)
+//   dot_lds[dh][row] dO          (dV B-fragments, per wave)
+//   pds_lds [key][row]           P then dS^T, b64-written from the MFMA
+//                                C-layout (4 consecutive rows at a fixed
+//                                key = one 8-B write), per wave
+//   dsrl_lds [row][key]          dS, scattered b16 writes, per wave
+
+#include "common.h"
+
+#define DH 64
+#define KT 64
+#define ATTN_WAVES 4
+#define ATTN_BLOCK (ATTN_WAVES * WAVE)
+#define NEG_INF (-1e30f)
+
+__device__ __forceinline__ int swz(int row, int byte_in_row) {
+  return (byte_in_row ^ ((row & 7) << 4));
+}
+
+__device__ __forceinline__ void rope8(float* x, const float* sinv,
+                                      const float* cosv) {
+#pragma unroll
+  for (int p = 0; p < 4; ++p) {
+    float x0 = x[2 * p], x1 = x[2 * p + 1];
+    float s = sinv[2 * p], c = cosv[2 * p];
+    x[2 * p] = x0 * c - x1 * s;
+    x[2 * p + 1] = x1 * c + x0 * s;
+  }
+}
+
+__device__ __forceinline__ void load_rope(const float* rsin,
+                                          const float* rcos, long long pos,
+                                          int d0, float* sv, float* cv) {
+  *(f32x4*)(sv) = *(const f32x4*)(rsin + pos * DH + d0);
+  *(f32x4*)(sv + 4) = *(const f32x4*)(rsin + pos * DH + d0 + 4);
+  *(f32x4*)(cv) = *(const f32x4*)(rcos + pos * DH + d0);
+  *(f32x4*)(cv + 4) = *(const f32x4*)(rcos + pos * DH + d0 + 4);
+}
+
+__global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
+    const short* __restrict__ dout,  // (B, N, H*DH) bf16
+    const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16
+    const float* __restrict__ rsin, const float* __restrict__ rcos,
+    const short* __restrict__ out,   // (B, N, H*DH) bf16 (fwd output)
+    const float* __restrict__ lse,   // (B, H, N)
+    float* __restrict__ dacc,        // (B, N, 3*H*DH) fp32, zero-init
+    int B, int N, int H, int wsz) {
+  const int window = blockIdx.x;
+  const int head = blockIdx.y;
+  const int batch = blockIdx.z;
+
+  const int lane = threadIdx.x % WAVE;
+  const int wid = threadIdx.x / WAVE;
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+
+  const long long HD3 = 3LL * H * DH;
+  const long long HD = (long long)H * DH;
+  const long long qkv_bn = (long long)batch * N * HD3;
+  const long long o_bn = (long long)batch * N * HD;
+  const int q_off = head * DH;
+  const int k_off = H * DH + head * DH;
+  const int v_off = 2 * H * DH + head * DH;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* k_lds = smem;                                   // 8 KiB
+  char* kt_lds = smem + 8192;                           // 8 KiB
+  char* v_lds = smem + 16384;                           // 8 KiB
+  char* qt_lds = smem + 24576 + wid * 8192;             // 32 KiB (4 waves)
+  char* dot_lds = smem + 24576 + 32768 + wid * 8192;    // 32 KiB
+  char* pds_lds = smem + 24576 + 65536 + wid * 8192;    // 32 KiB
+  char* dsrl_lds = smem + 24576 + 98304 + wid * 8192;   // 32 KiB
+  float* d_lds = (float*)(smem + 24576 + 131072 + wid * 256);      // 1 KiB
+  float* lse_lds = (float*)(smem + 24576 + 131072 + 1024 + wid * 256);
+
+  const float scale = rsqrtf((float)DH);
+  const int tiles = 2 * wsz / KT;
+  const int chunks = wsz / 64;
+  const int rounds = (chunks + ATTN_WAVES - 1) / ATTN_WAVES;
+
+  for (int round = 0; round < rounds; ++round) {
+    const int chunk = round * ATTN_WAVES + wid;
+    const bool active = chunk < chunks;
+    const int chunk_off = chunk * 64;
+    const int q0 = window * wsz + chunk_off;
+
+    bf16x8 qfrag[4][2];  // rotary q' (unscaled), S recompute A-fragments
+    f32x4 dqacc[4][4];
+#pragma unroll
+    for (int m = 0; m < 4; ++m)
+#pragma unroll
+      for (int d = 0; d < 4; ++d) dqacc[m][d] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    if (active) {
+#pragma unroll
+      for (int m = 0; m < 4; ++m) {
+        const int row = q0 + m * 16 + l15;
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          const int d0 = ks * 32 + 8 * l4;
+          bf16x8 v = *(const bf16x8*)(qkv + qkv_bn + (long long)row * HD3 +
+                                      q_off + d0);
+          float x[8], sv[8], cv[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) x[j] = bf2f(((short*)&v)[j]);
+          load_rope(rsin, rcos, row, d0, sv, cv);
+          rope8(x, sv, cv);
+          bf16x8 o;
+#pragma unroll
+          for (int j = 0; j < 8; ++j) ((short*)&o)[j] = f2bf(x[j]);
+          qfrag[m][ks] = o;
+        }
+      }
+
+      // per-round wave-local staging: Q^T (scaled), dO^T, D, lse.
+      // One lane per row of the chunk.
+      {
+        const int row = lane;
+        const long long gq = qkv_bn + (long long)(q0 + row) * HD3 + q_off;
+        const long long go = o_bn + (long long)(q0 + row) * HD + head * DH;
+        float dsum = 0.f;
+#pragma unroll
+        for (int g = 0; g < 8; ++g) {
+          const int d0 = g * 8;
+          bf16x8 qv = *(const bf16x8*)(qkv + gq + d0);
+          bf16x8 ov = *(const bf16x8*)(out + go + d0);
+          bf16x8 dov = *(const bf16x8*)(dout + go + d0);
+          float x[8], sv[8], cv[8];
+#pragma unroll
+          for (int j = 0; j < 8; ++j) x[j] = bf2f(((short*)&qv)[j]);
+          load_rope(rsin, rcos, q0 + row, d0, sv, cv);
+          rope8(x, sv, cv);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int d = d0 + j;
+            *(short*)(qt_lds + d * 128 + swz(d, row * 2)) = f2bf(x[j] * scale);
+            *(short*)(dot_lds + d * 128 + swz(d, row * 2)) = ((short*)&dov)[j];
+            dsum += bf2f(((short*)&ov)[j]) * bf2f(((short*)&dov)[j]);
+          }
+        }
+        d_lds[row] = dsum;
+        lse_lds[row] = lse[((long long)batch * H + head) * N + q0 + row];
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    }
+
+    const int max_tile = active ? ((chunk_off + 63 + wsz) / KT) : -1;
+
+    for (int t = 0; t < tiles; ++t) {
+      __syncthreads();
+#pragma unroll
+      for (int pass = 0; pass < 2; ++pass) {
+        const int flat = pass * ATTN_BLOCK + threadIdx.x;
+        const int key = flat >> 3;
+        const int d0 = (flat & 7) * 8;
+        const int kpos = (window - 1) * wsz + t * KT + key;
+        float kx[8], vx[8];
+        if (kpos >= 0) {
+          bf16x8 kvec = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 +
+                                         k_off + d0);
+          bf16x8 vvec = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 +
+                                         v_off + d0);
+          float sv[8], cv[8];
+          load_rope(rsin, rcos, kpos, d0, sv, cv);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) kx[j] = bf2f(((short*)&kvec)[j]);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) vx[j] = bf2f(((short*)&vvec)[j]);
+          rope8(kx, sv, cv);
+          rope8(vx, sv, cv);  // rotary on V too (progen.py:87)
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) kx[j] = vx[j] = 0.f;
+        }
+        bf16x8 kb, vb;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          ((short*)&kb)[j] = f2bf(kx[j] * scale);
+          ((short*)&vb)[j] = f2bf(vx[j]);
+        }
+        *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kb;
+        *(bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2)) = vb;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int d = d0 + j;
+          *(short*)(kt_lds + d * 128 + swz(d, key * 2)) = ((short*)&kb)[j];
+        }
+      }
+      __syncthreads();
+
+      if (active && t <= max_tile) {
+        const int kb = t * KT;
+
+        // ---- S = q' k_s^T ----
+        f32x4 s[4][4];
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) s[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            const int key = n * 16 + l15;
+            bf16x8 kf = *(const bf16x8*)(k_lds + key * 128 +
+                                         swz(key, (ks * 32 + 8 * l4) * 2));
+#pragma unroll
+            for (int m = 0; m < 4; ++m)
+              s[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  qfrag[m][ks], kf, s[m][n], 0, 0, 0);
+          }
+
+        // ---- P = exp(S - lse) masked; b64-write P^T into pds_lds ----
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int rowiw = chunk_off + m * 16 + l4 * 4 + r;
+            const float l = lse_lds[m * 16 + l4 * 4 + r];
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int kpos_band = kb + n * 16 + l15;
+              float v = ((float*)&s[m][n])[r];
+              v = (kpos_band > rowiw + wsz) ? 0.f : __expf(v - l);
+              ((float*)&s[m][n])[r] = v;  // s now holds P
+            }
+          }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            const int key = n * 16 + l15;
+            const int row0 = m * 16 + l4 * 4;
+            short pk[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) pk[r] = f2bf(((float*)&s[m][n])[r]);
+            *(unsigned long long*)(pds_lds + key * 128 + swz(key, row0 * 2)) =
+                *(unsigned long long*)pk;
+          }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+        // ---- dV = P^T dO : A = P^T (pds), B = dO^T image (dot_lds) ----
+        {
+          f32x4 dv[4][4];
+#pragma unroll
+          for (int m = 0; m < 4; ++m)
+#pragma unroll
+            for (int n = 0; n < 4; ++n) dv[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+            for (int m = 0; m < 4; ++m) {  // m: key fragment rows
+              const int key = m * 16 + l15;
+              bf16x8 pf = *(const bf16x8*)(pds_lds + key * 128 +
+                                           swz(key, (ks * 32 + 8 * l4) * 2));
+#pragma unroll
+              for (int n = 0; n < 4; ++n) {  // n: dh fragment cols
+                const int d = n * 16 + l15;
+                bf16x8 dof = *(const bf16x8*)(dot_lds + d * 128 +
+                                              swz(d, (ks * 32 + 8 * l4) * 2));
+                dv[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    pf, dof, dv[m][n], 0, 0, 0);
+              }
+            }
+          // C: dV[key][dh]; atomicAdd into dacc v region (skip pad keys)
+#pragma unroll
+          for (int m = 0; m < 4; ++m)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int kpos = (window - 1) * wsz + kb + m * 16 + l4 * 4 + r;
+              if (kpos >= 0) {
+#pragma unroll
+                for (int n = 0; n < 4; ++n)
+                  atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + v_off +
+                                n * 16 + l15,
+                            ((float*)&dv[m][n])[r]);
+              }
+            }
+        }
+
+        // ---- dP = dO V'^T ; dS = P o (dP - D) ----
+        f32x4 dp[4][4];
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) dp[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int m = 0; m < 4; ++m) {
+            const int row = q0 + m * 16 + l15;
+            const int d0 = ks * 32 + 8 * l4;
+            bf16x8 dof = *(const bf16x8*)(dout + o_bn + (long long)row * HD +
+                                          head * DH + d0);
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int key = n * 16 + l15;
+              bf16x8 vf = *(const bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2));
+              dp[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  dof, vf, dp[m][n], 0, 0, 0);
+            }
+          }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const float dval = d_lds[m * 16 + l4 * 4 + r];
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              float p = ((float*)&s[m][n])[r];
+              float d = ((float*)&dp[m][n])[r];
+              ((float*)&dp[m][n])[r] = p * (d - dval);  // dp now holds dS
+            }
+          }
+
+        // ---- write dS^T into pds_lds (b64, overwrites P) and dS into
+        //      dsrl_lds (scattered b16) ----
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            const int key = n * 16 + l15;
+            const int row0 = m * 16 + l4 * 4;
+            short dk4[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) dk4[r] = f2bf(((float*)&dp[m][n])[r]);
+            *(unsigned long long*)(pds_lds + key * 128 + swz(key, row0 * 2)) =
+                *(unsigned long long*)dk4;
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              *(short*)(dsrl_lds + (row0 + r) * 128 + swz(row0 + r, key * 2)) =
+                  dk4[r];
+          }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+        // ---- dK = dS^T q_s : A = dS^T (pds), B = Q^T image (qt_lds) ----
+        {
+          f32x4 dk[4][4];
+#pragma unroll
+          for (int m = 0; m < 4; ++m)
+#pragma unroll
+            for (int n = 0; n < 4; ++n) dk[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+            for (int m = 0; m < 4; ++m) {
+              const int key = m * 16 + l15;
+              bf16x8 dsf = *(const bf16x8*)(pds_lds + key * 128 +
+                                            swz(key, (ks * 32 + 8 * l4) * 2));
+#pragma unroll
+              for (int n = 0; n < 4; ++n) {
+                const int d = n * 16 + l15;
+                bf16x8 qf = *(const bf16x8*)(qt_lds + d * 128 +
+                                             swz(d, (ks * 32 + 8 * l4) * 2));
+                dk[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                    dsf, qf, dk[m][n], 0, 0, 0);
+              }
+            }
+#pragma unroll
+          for (int m = 0; m < 4; ++m)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              const int kpos = (window - 1) * wsz + kb + m * 16 + l4 * 4 + r;
+              if (kpos >= 0) {
+#pragma unroll
+                for (int n = 0; n < 4; ++n)
+                  atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + k_off +
+                                n * 16 + l15,
+                            ((float*)&dk[m][n])[r]);
+              }
+            }
+        }
+
+        // ---- dQ += dS k_s : A = dS (dsrl), B = K^T image (kt_lds) ----
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int m = 0; m < 4; ++m) {
+            const int row = m * 16 + l15;
+            bf16x8 dsf = *(const bf16x8*)(dsrl_lds + row * 128 +
+                                          swz(row, (ks * 32 + 8 * l4) * 2));
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int d = n * 16 + l15;
+              bf16x8 kf = *(const bf16x8*)(kt_lds + d * 128 +
+                                           swz(d, (ks * 32 + 8 * l4) * 2));
+              dqacc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  dsf, kf, dqacc[m][n], 0, 0, 0);
+            }
+          }
+      }
+    }
+
+    // ---- store dQ (rows exclusively owned -> plain fp32 stores) ----
+    if (active) {
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = q0 + m * 16 + l4 * 4 + r;
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+            dacc[qkv_bn + (long long)row * HD3 + q_off + n * 16 + l15] =
+                ((float*)&dqacc[m][n])[r];
+        }
+    }
+  }
+}
+
+// ---------------------------------------------------------------------------
+// finalize: inverse rotary rotation on the fp32 accumulator -> bf16 dqkv
+// ---------------------------------------------------------------------------
+
+__global__ __launch_bounds__(256) void attn_bwd_finalize_kernel(
+    const float* __restrict__ dacc, const float* __restrict__ rsin,
+    const float* __restrict__ rcos, short* __restrict__ dqkv, int B, int N,
+    int H) {
+  // one thread per 8-element dh group of one (b, n, qkv-slot, head)
+  const long long HD3 = 3LL * H * DH;
+  const long long total = (long long)B * N * 3 * H * (DH / 8);
+  for (long long idx = blockIdx.x * 256LL + threadIdx.x; idx < total;
+       idx += (long long)gridDim.x * 256) {
+    const int g = idx % (DH / 8);
+    const long long rest = idx / (DH / 8);
+    const int hslot = rest % (3 * H);
+    const long long bn = rest / (3 * H);
+    const int n = bn % N;
+    const int d0 = g * 8;
+
+    const long long off = bn * HD3 + (long long)hslot * DH + d0;
+    float x[8], sv[8], cv[8];
+#pragma unroll
+    for (int j = 0; j < 8; ++j) x[j] = dacc[off + j];
+    load_rope(rsin, rcos, n, d0, sv, cv);
+    // inverse rotation: dx[2i] = dy[2i] c + dy[2i+1] s;
+    //                   dx[2i+1] = dy[2i+1] c - dy[2i] s
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      float y0 = x[2 * p], y1 = x[2 * p + 1];
+      float s = sv[2 * p], c = cv[2 * p];
+      x[2 * p] = y0 * c + y1 * s;
+      x[2 * p + 1] = y1 * c - y0 * s;
+    }
+    bf16x8 o;
+#pragma unroll
+    for (int j = 0; j < 8; ++j) ((short*)&o)[j] = f2bf(x[j]);
+    *(bf16x8*)(dqkv + off) = o;
+  }
+}
+
+extern "C" {
+
+void attn_bwd_launch(const void* dout, const void* qkv, const float* rsin,
+                     const float* rcos, const void* out, const float* lse,
+                     float* dacc, void* dqkv, int B, int N, int H, int wsz,
+                     hipStream_t stream) {
+  dim3 grid(N / wsz, H, B), block(ATTN_BLOCK);
+  size_t lds = 24576 + 131072 + 2048;  // 154 KiB
+  attn_bwd_kernel<<<grid, block, lds, stream>>>(
+      (const short*)dout, (const short*)qkv, rsin, rcos, (const short*)out,
+      lse, dacc, B, N, H, wsz);
+  long long total = (long long)B * N * 3 * H * (DH / 8);
+  int fin_grid = (int)((total + 255) / 256);
+  if (fin_grid > 2048) fin_grid = 2048;
+  attn_bwd_finalize_kernel<<<fin_grid, 256, 0, stream>>>(dacc, rsin, rcos,
+                                                         (short*)dqkv, B, N, H);
+}
+
+}  // extern "C"

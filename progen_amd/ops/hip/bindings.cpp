@@ -1,0 +1,217 @@
+// PyTorch bindings for the ProGen CDNA4 HIP kernels (progen_amd._C).
+//
+// Tensor checks + output allocation here; all device code lives in the
+// .hip translation units (compiled for gfx950 only — no CUDA path).
+
+#include <torch/extension.h>
+
+#include <c10/hip/HIPStream.h>
+
+#include "kernels.h"
+
+namespace {
+
+inline hipStream_t cur_stream() {
+  return c10::hip::getCurrentHIPStream().stream();
+}
+
+inline bool check_dtype(const at::Tensor& t) {
+  TORCH_CHECK(t.scalar_type() == at::kBFloat16 || t.scalar_type() == at::kFloat,
+              "expected bf16 or fp32 tensor");
+  return t.scalar_type() == at::kBFloat16;
+}
+
+}  // namespace
+
+// ---------------------------------------------------------------------------
+// ln_shift
+// ---------------------------------------------------------------------------
+
+std::vector<at::Tensor> ln_shift_fwd(const at::Tensor& x, const at::Tensor& g,
+                                     bool shift, double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  bool bf = check_dtype(x);
+  const int B = x.size(0), N = x.size(1), D = x.size(2);
+  TORCH_CHECK(D % 16 == 0, "ln_shift: D must be a multiple of 16");
+  auto y = at::empty_like(x);
+  auto mean = at::empty({(long)B * N}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty_like(mean);
+  ln_shift_fwd_launch(x.data_ptr(), g.data_ptr(), y.data_ptr(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(), B * N, N,
+                      D, (float)eps, shift, bf, cur_stream());
+  return {y, mean, rstd};
+}
+
+std::vector<at::Tensor> ln_shift_bwd(const at::Tensor& dy, const at::Tensor& x,
+                                     const at::Tensor& g,
+                                     const at::Tensor& mean,
+                                     const at::Tensor& rstd, bool shift) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && x.is_contiguous());
+  bool bf = check_dtype(x);
+  const int B = x.size(0), N = x.size(1), D = x.size(2);
+  const int R = B * N;
+  int nblocks = std::min(R, 256);
+  auto dx = at::empty_like(x);
+  auto dw_part = at::empty({nblocks, D}, x.options().dtype(at::kFloat));
+  ln_shift_bwd_launch(dy.data_ptr(), x.data_ptr(), g.data_ptr(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                      dx.data_ptr(), dw_part.data_ptr<float>(), nblocks, R, N,
+                      D, shift, bf, cur_stream());
+  auto dw = dw_part.sum(0).to(x.scalar_type());
+  return {dx, dw};
+}
+
+// ---------------------------------------------------------------------------
+// glu / gelu
+// ---------------------------------------------------------------------------
+
+at::Tensor glu_fwd(const at::Tensor& h) {
+  TORCH_CHECK(h.is_cuda() && h.is_contiguous());
+  bool bf = check_dtype(h);
+  const int H2 = h.size(-1);
+  TORCH_CHECK(H2 % 16 == 0);
+  long long rows = h.numel() / H2;
+  auto sizes = h.sizes().vec();
+  sizes.back() = H2 / 2;
+  auto y = at::empty(sizes, h.options());
+  glu_fwd_launch(h.data_ptr(), y.data_ptr(), rows, H2 / 2, bf, cur_stream());
+  return y;
+}
+
+at::Tensor glu_bwd(const at::Tensor& dy, const at::Tensor& h) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && h.is_contiguous());
+  bool bf = check_dtype(h);
+  const int H2 = h.size(-1);
+  long long rows = h.numel() / H2;
+  auto dh = at::empty_like(h);
+  glu_bwd_launch(dy.data_ptr(), h.data_ptr(), dh.data_ptr(), rows, H2 / 2, bf,
+                 cur_stream());
+  return dh;
+}
+
+at::Tensor gelu_fwd(const at::Tensor& h) {
+  TORCH_CHECK(h.is_cuda() && h.is_contiguous());
+  bool bf = check_dtype(h);
+  TORCH_CHECK(h.numel() % 8 == 0);
+  auto y = at::empty_like(h);
+  gelu_fwd_launch(h.data_ptr(), y.data_ptr(), h.numel(), bf, cur_stream());
+  return y;
+}
+
+at::Tensor gelu_bwd(const at::Tensor& dy, const at::Tensor& h) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && h.is_contiguous());
+  bool bf = check_dtype(h);
+  auto dh = at::empty_like(h);
+  gelu_bwd_launch(dy.data_ptr(), h.data_ptr(), dh.data_ptr(), h.numel(), bf,
+                  cur_stream());
+  return dh;
+}
+
+// ---------------------------------------------------------------------------
+// cross entropy
+// ---------------------------------------------------------------------------
+
+std::vector<at::Tensor> ce_fwd(const at::Tensor& logits,
+                               const at::Tensor& targets) {
+  TORCH_CHECK(logits.is_cuda() && logits.is_contiguous());
+  TORCH_CHECK(targets.scalar_type() == at::kLong && targets.is_contiguous());
+  bool bf = check_dtype(logits);
+  const int V = logits.size(-1);
+  long long R = logits.numel() / V;
+  auto sizes = targets.sizes().vec();
+  auto nll = at::empty(sizes, logits.options().dtype(at::kFloat));
+  auto lse = at::empty_like(nll);
+  ce_fwd_launch(logits.data_ptr(), (const long long*)targets.data_ptr<long>(),
+                nll.data_ptr<float>(), lse.data_ptr<float>(), R, V, bf,
+                cur_stream());
+  return {nll, lse};
+}
+
+at::Tensor ce_bwd(const at::Tensor& dnll, const at::Tensor& logits,
+                  const at::Tensor& targets, const at::Tensor& lse) {
+  TORCH_CHECK(dnll.is_cuda() && dnll.is_contiguous());
+  bool bf = check_dtype(logits);
+  const int V = logits.size(-1);
+  long long R = logits.numel() / V;
+  auto dlogits = at::empty_like(logits);
+  ce_bwd_launch(dnll.data_ptr<float>(), logits.data_ptr(),
+                (const long long*)targets.data_ptr<long>(), lse.data_ptr<float>(),
+                dlogits.data_ptr(), R, V, bf, cur_stream());
+  return dlogits;
+}
+
+// ---------------------------------------------------------------------------
+// fused adamw
+// ---------------------------------------------------------------------------
+
+void fused_adamw(at::Tensor& master, at::Tensor& params,
+                 const at::Tensor& grads, at::Tensor& exp_avg,
+                 at::Tensor& exp_avg_sq, const at::Tensor& chunk_starts,
+                 const at::Tensor& chunk_ends, const at::Tensor& chunk_decay,
+                 double lr, double b1, double b2, double eps, double wd,
+                 long step, double grad_scale, const at::Tensor& clip_coef) {
+  TORCH_CHECK(master.is_cuda() && master.scalar_type() == at::kFloat);
+  bool bf = params.scalar_type() == at::kBFloat16;
+  fused_adamw_launch(master.data_ptr<float>(), params.data_ptr(),
+                     grads.data_ptr(), exp_avg.data_ptr<float>(),
+                     exp_avg_sq.data_ptr<float>(),
+                     (const long long*)chunk_starts.data_ptr<long>(), (const long long*)chunk_ends.data_ptr<long>(),
+                     chunk_decay.data_ptr<int>(), chunk_starts.size(0),
+                     (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
+                     (int)step, (float)grad_scale, clip_coef.data_ptr<float>(),
+                     bf, cur_stream());
+}
+
+// ---------------------------------------------------------------------------
+// local attention
+// ---------------------------------------------------------------------------
+
+std::vector<at::Tensor> attn_fwd(const at::Tensor& qkv, const at::Tensor& rsin,
+                                 const at::Tensor& rcos, long heads,
+                                 long window) {
+  TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 3);
+  TORCH_CHECK(qkv.scalar_type() == at::kBFloat16,
+              "attn_fwd: bf16 only (MFMA path)");
+  const int B = qkv.size(0), N = qkv.size(1);
+  const int H = (int)heads, wsz = (int)window;
+  TORCH_CHECK(qkv.size(2) == 3LL * H * 64, "attn: dim_head must be 64");
+  TORCH_CHECK(N % wsz == 0 && wsz % 64 == 0,
+              "attn: seq divisible by window, window divisible by 64");
+  TORCH_CHECK(rsin.scalar_type() == at::kFloat && rsin.is_contiguous());
+  auto out = at::empty({B, N, (long)H * 64}, qkv.options());
+  auto lse = at::empty({B, (long)H, N}, qkv.options().dtype(at::kFloat));
+  attn_fwd_launch(qkv.data_ptr(), rsin.data_ptr<float>(),
+                  rcos.data_ptr<float>(), out.data_ptr(),
+                  lse.data_ptr<float>(), B, N, H, wsz, cur_stream());
+  return {out, lse};
+}
+
+at::Tensor attn_bwd(const at::Tensor& dout, const at::Tensor& qkv,
+                    const at::Tensor& rsin, const at::Tensor& rcos,
+                    const at::Tensor& out, const at::Tensor& lse, long heads,
+                    long window) {
+  TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
+  const int B = qkv.size(0), N = qkv.size(1);
+  const int H = (int)heads, wsz = (int)window;
+  auto dacc = at::zeros_like(qkv, qkv.options().dtype(at::kFloat));
+  auto dqkv = at::empty_like(qkv);
+  attn_bwd_launch(dout.data_ptr(), qkv.data_ptr(), rsin.data_ptr<float>(),
+                  rcos.data_ptr<float>(), out.data_ptr(),
+                  lse.data_ptr<float>(), dacc.data_ptr<float>(),
+                  dqkv.data_ptr(), B, N, H, wsz, cur_stream());
+  return dqkv;
+}
+
+PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("ln_shift_fwd", &ln_shift_fwd, "fused LN+shift forward");
+  m.def("ln_shift_bwd", &ln_shift_bwd, "fused LN+shift backward");
+  m.def("glu_fwd", &glu_fwd, "GLU-GELU forward");
+  m.def("glu_bwd", &glu_bwd, "GLU-GELU backward");
+  m.def("gelu_fwd", &gelu_fwd, "GELU forward");
+  m.def("gelu_bwd", &gelu_bwd, "GELU backward");
+  m.def("ce_fwd", &ce_fwd, "fused CE forward (nll, lse)");
+  m.def("ce_bwd", &ce_bwd, "fused CE backward");
+  m.def("fused_adamw", &fused_adamw, "fused clip+AdamW over flat space");
+  m.def("attn_fwd", &attn_fwd, "fused local attention forward");
+  m.def("attn_bwd", &attn_bwd, "fused local attention backward");
+}

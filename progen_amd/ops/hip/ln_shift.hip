@@ -1,0 +1,282 @@
+// Fused scale-only LayerNorm + token shift, forward and backward.
+//
+// Implements the LN->shift prologue of both branches
+// (reference: progen.py:22,43-46,74-77,132-135) as one memory-bound pass:
+// the block normalizes row (b, n) and writes the first ceil(D/2) channels
+// into row n+1 (the token shift) and the rest into row n, so the shifted
+// activation never exists as a separate tensor.
+//
+// Memory-bound: vectorized 16 B/lane loads (bf16x8), fp32 statistics,
+// one block per row forward; grid-strided rows backward with per-block
+// dweight partials reduced on the host side.
+
+#include "common.h"
+
+#define LN_BLOCK 256
+#define LN_WAVES (LN_BLOCK / WAVE)
+
+// ---------------------------------------------------------------------------
+// forward
+// ---------------------------------------------------------------------------
+
+template <typename VEC, bool SHIFT, bool IS_BF16>
+__global__ __launch_bounds__(LN_BLOCK) void ln_shift_fwd_kernel(
+    const VEC* __restrict__ x, const VEC* __restrict__ g, VEC* __restrict__ y,
+    float* __restrict__ mean, float* __restrict__ rstd, int N, int Dv,
+    float eps) {
+  // Dv = D / VLEN (vector units); each VEC is 8 bf16 or 4 f32 (16 B)
+  constexpr int VLEN = IS_BF16 ? 8 : 4;
+  const int row = blockIdx.x;
+  const int n = row % N;
+  const long long base = (long long)row * Dv;
+
+  __shared__ float red[LN_WAVES];
+
+  float s = 0.f, ss = 0.f;
+  for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
+    VEC v = x[base + i];
+#pragma unroll
+    for (int j = 0; j < VLEN; ++j) {
+      float f = IS_BF16 ? bf2f(((short*)&v)[j]) : ((float*)&v)[j];
+      s += f;
+      ss += f * f;
+    }
+  }
+  s = wave_sum(s);
+  ss = wave_sum(ss);
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  if (lane == 0) red[wid] = s;
+  __syncthreads();
+  if (wid == 0) {
+    float t = (lane < LN_WAVES) ? red[lane] : 0.f;
+#pragma unroll
+    for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
+    if (lane == 0) red[0] = t;
+  }
+  __syncthreads();
+  s = red[0];
+  __syncthreads();
+  if (lane == 0) red[wid] = ss;
+  __syncthreads();
+  if (wid == 0) {
+    float t = (lane < LN_WAVES) ? red[lane] : 0.f;
+#pragma unroll
+    for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
+    if (lane == 0) red[0] = t;
+  }
+  __syncthreads();
+  ss = red[0];
+
+  const int D = Dv * VLEN;
+  const float mu = s / D;
+  const float var = fmaxf(ss / D - mu * mu, 0.f);
+  const float rs = rsqrtf(var + eps);
+  if (threadIdx.x == 0) {
+    mean[row] = mu;
+    rstd[row] = rs;
+  }
+
+  const int halfv = (D / 2) / VLEN;  // D even, half % VLEN == 0 (checked host-side)
+  for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
+    VEC v = x[base + i];
+    VEC gw = g[i];
+    VEC o;
+#pragma unroll
+    for (int j = 0; j < VLEN; ++j) {
+      float f = IS_BF16 ? bf2f(((short*)&v)[j]) : ((float*)&v)[j];
+      float gj = IS_BF16 ? bf2f(((short*)&gw)[j]) : ((float*)&gw)[j];
+      float r = (f - mu) * rs * gj;
+      if (IS_BF16) ((short*)&o)[j] = f2bf(r);
+      else ((float*)&o)[j] = r;
+    }
+    if (!SHIFT || i >= halfv) {
+      y[base + i] = o;
+    } else if (n + 1 < N) {  // shift half -> next row (within the sequence)
+      y[base + Dv + i] = o;
+    }
+  }
+  if (SHIFT && n == 0) {  // first row's shift half is the zero pad
+    VEC z;
+#pragma unroll
+    for (int j = 0; j < VLEN; ++j) {
+      if (IS_BF16) ((short*)&z)[j] = 0; else ((float*)&z)[j] = 0.f;
+    }
+    for (int i = threadIdx.x; i < halfv; i += LN_BLOCK) y[base + i] = z;
+  }
+}
+
+// ---------------------------------------------------------------------------
+// backward
+// ---------------------------------------------------------------------------
+// e[d] = upstream grad routed back through the shift:
+//   e[d < half] = dy[row+1][d] (0 at the last row of a sequence)
+//   e[d >= half] = dy[row][d]
+// dx = rs * (e*g - mean(e*g) - xhat * mean(e*g*xhat));  dw += e * xhat
+
+template <typename VEC, bool SHIFT, bool IS_BF16>
+__global__ __launch_bounds__(LN_BLOCK) void ln_shift_bwd_kernel(
+    const VEC* __restrict__ dy, const VEC* __restrict__ x,
+    const VEC* __restrict__ g, const float* __restrict__ mean,
+    const float* __restrict__ rstd, VEC* __restrict__ dx,
+    float* __restrict__ dw_part, int R, int N, int Dv) {
+  constexpr int VLEN = IS_BF16 ? 8 : 4;
+  const int D = Dv * VLEN;
+  const int halfv = (D / 2) / VLEN;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  float* dw_lds = (float*)smem;             // D floats
+  float* red = (float*)(smem + (size_t)D * 4);  // LN_WAVES floats
+
+  for (int i = threadIdx.x; i < D; i += LN_BLOCK) dw_lds[i] = 0.f;
+  __syncthreads();
+
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+
+  for (int row = blockIdx.x; row < R; row += gridDim.x) {
+    const int n = row % N;
+    const long long base = (long long)row * Dv;
+    const float mu = mean[row];
+    const float rs = rstd[row];
+
+    float a = 0.f, bsum = 0.f;  // a = sum(e*g*xhat), bsum = sum(e*g)
+    for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
+      const bool shifted = SHIFT && (i < halfv);
+      VEC e;
+      bool have_e = true;
+      if (shifted) {
+        if (n + 1 < N) e = dy[base + Dv + i];
+        else have_e = false;
+      } else {
+        e = dy[base + i];
+      }
+      VEC v = x[base + i];
+      VEC gw = g[i];
+#pragma unroll
+      for (int j = 0; j < VLEN; ++j) {
+        float ej = have_e ? (IS_BF16 ? bf2f(((short*)&e)[j]) : ((float*)&e)[j]) : 0.f;
+        float f = IS_BF16 ? bf2f(((short*)&v)[j]) : ((float*)&v)[j];
+        float gj = IS_BF16 ? bf2f(((short*)&gw)[j]) : ((float*)&gw)[j];
+        float xh = (f - mu) * rs;
+        a += ej * gj * xh;
+        bsum += ej * gj;
+      }
+    }
+    a = wave_sum(a);
+    bsum = wave_sum(bsum);
+    if (lane == 0) red[wid] = a;
+    __syncthreads();
+    if (wid == 0) {
+      float t = (lane < LN_WAVES) ? red[lane] : 0.f;
+#pragma unroll
+      for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
+      if (lane == 0) red[0] = t;
+    }
+    __syncthreads();
+    a = red[0];
+    __syncthreads();
+    if (lane == 0) red[wid] = bsum;
+    __syncthreads();
+    if (wid == 0) {
+      float t = (lane < LN_WAVES) ? red[lane] : 0.f;
+#pragma unroll
+      for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
+      if (lane == 0) red[0] = t;
+    }
+    __syncthreads();
+    bsum = red[0];
+
+    const float inv_d = 1.0f / D;
+    for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
+      const bool shifted = SHIFT && (i < halfv);
+      VEC e;
+      bool have_e = true;
+      if (shifted) {
+        if (n + 1 < N) e = dy[base + Dv + i];
+        else have_e = false;
+      } else {
+        e = dy[base + i];
+      }
+      VEC v = x[base + i];
+      VEC gw = g[i];
+      VEC o;
+#pragma unroll
+      for (int j = 0; j < VLEN; ++j) {
+        float ej = have_e ? (IS_BF16 ? bf2f(((short*)&e)[j]) : ((float*)&e)[j]) : 0.f;
+        float f = IS_BF16 ? bf2f(((short*)&v)[j]) : ((float*)&v)[j];
+        float gj = IS_BF16 ? bf2f(((short*)&gw)[j]) : ((float*)&gw)[j];
+        float xh = (f - mu) * rs;
+        float dxv = rs * (ej * gj - bsum * inv_d - xh * a * inv_d);
+        if (IS_BF16) ((short*)&o)[j] = f2bf(dxv);
+        else ((float*)&o)[j] = dxv;
+        dw_lds[i * VLEN + j] += ej * xh;  // thread-private index: no race
+      }
+      dx[base + i] = o;
+    }
+    __syncthreads();
+  }
+
+  float* out = dw_part + (size_t)blockIdx.x * D;
+  for (int i = threadIdx.x; i < D; i += LN_BLOCK) out[i] = dw_lds[i];
+}
+
+// ---------------------------------------------------------------------------
+// launchers
+// ---------------------------------------------------------------------------
+
+extern "C" {
+
+void ln_shift_fwd_launch(const void* x, const void* g, void* y, float* mean,
+                         float* rstd, int R, int N, int D, float eps,
+                         bool shift, bool is_bf16, hipStream_t stream) {
+  dim3 grid(R), block(LN_BLOCK);
+  if (is_bf16) {
+    int Dv = D / 8;
+    if (shift)
+      ln_shift_fwd_kernel<bf16x8, true, true><<<grid, block, 0, stream>>>(
+          (const bf16x8*)x, (const bf16x8*)g, (bf16x8*)y, mean, rstd, N, Dv, eps);
+    else
+      ln_shift_fwd_kernel<bf16x8, false, true><<<grid, block, 0, stream>>>(
+          (const bf16x8*)x, (const bf16x8*)g, (bf16x8*)y, mean, rstd, N, Dv, eps);
+  } else {
+    int Dv = D / 4;
+    if (shift)
+      ln_shift_fwd_kernel<f32x4, true, false><<<grid, block, 0, stream>>>(
+          (const f32x4*)x, (const f32x4*)g, (f32x4*)y, mean, rstd, N, Dv, eps);
+    else
+      ln_shift_fwd_kernel<f32x4, false, false><<<grid, block, 0, stream>>>(
+          (const f32x4*)x, (const f32x4*)g, (f32x4*)y, mean, rstd, N, Dv, eps);
+  }
+}
+
+void ln_shift_bwd_launch(const void* dy, const void* x, const void* g,
+                         const float* mean, const float* rstd, void* dx,
+                         float* dw_part, int nblocks, int R, int N, int D,
+                         bool shift, bool is_bf16, hipStream_t stream) {
+  dim3 grid(nblocks), block(LN_BLOCK);
+  size_t lds = (size_t)D * 4 + LN_WAVES * 4 + 16;
+  if (is_bf16) {
+    int Dv = D / 8;
+    if (shift)
+      ln_shift_bwd_kernel<bf16x8, true, true><<<grid, block, lds, stream>>>(
+          (const bf16x8*)dy, (const bf16x8*)x, (const bf16x8*)g, mean, rstd,
+          (bf16x8*)dx, dw_part, R, N, Dv);
+    else
+      ln_shift_bwd_kernel<bf16x8, false, true><<<grid, block, lds, stream>>>(
+          (const bf16x8*)dy, (const bf16x8*)x, (const bf16x8*)g, mean, rstd,
+          (bf16x8*)dx, dw_part, R, N, Dv);
+  } else {
+    int Dv = D / 4;
+    if (shift)
+      ln_shift_bwd_kernel<f32x4, true, false><<<grid, block, lds, stream>>>(
+          (const f32x4*)dy, (const f32x4*)x, (const f32x4*)g, mean, rstd,
+          (f32x4*)dx, dw_part, R, N, Dv);
+    else
+      ln_shift_bwd_kernel<f32x4, false, false><<<grid, block, lds, stream>>>(
+          (const f32x4*)dy, (const f32x4*)x, (const f32x4*)g, mean, rstd,
+          (f32x4*)dx, dw_part, R, N, Dv);
+  }
+}
+
+}  // extern "C"
