@@ -154,6 +154,12 @@ class ProGenBase(nn.Module):
         if x.dim() == 1:
             x = x.unsqueeze(0)
         n = x.shape[1]
+        if self.rotary_sin.dtype != torch.float32:
+            # module.to(bf16) casts buffers; rotary tables must stay fp32
+            sin, cos = R.fixed_pos_embedding(self.cfg.seq_len,
+                                             self.cfg.dim_head,
+                                             device=self.rotary_sin.device)
+            self.rotary_sin, self.rotary_cos = sin, cos
         sin = self.rotary_sin[:n]
         cos = self.rotary_cos[:n]
         h = self.embed(x.long())

@@ -164,7 +164,8 @@ def cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
     per-sequence masked mean then batch mean reduction order is preserved
     (reference: utils.py:67,75-76)."""
     if dispatch.use_hip(logits):
-        nll = _CERowFn.apply(logits, targets)
+        targets = targets.long().contiguous()
+        nll = _CERowFn.apply(logits.contiguous(), targets)
         mask = targets != ignore_index
         eos_mask = (~mask).long().cumsum(dim=-1) == 1
         mask = (mask | eos_mask).to(nll.dtype)

@@ -24,6 +24,31 @@ def exists(v) -> bool:
     return v is not None
 
 
+def load_dotenv(path: str = ".env") -> None:
+    """Minimal .env loader (parity with the reference's dotenv usage,
+    reference: train.py:1-2, .env:1-2 — there it carries XLA flags; here
+    it can carry ROCm/RCCL knobs like NCCL_MIN_NCHANNELS)."""
+    try:
+        with open(path) as f:
+            for line in f:
+                line = line.strip()
+                if not line or line.startswith("#") or "=" not in line:
+                    continue
+                k, _, v = line.partition("=")
+                os.environ.setdefault(k.strip(), v.strip())
+    except FileNotFoundError:
+        pass
+
+
+def set_hardware_rng_(*_args, **_kwargs) -> None:
+    """API-parity stub. The reference monkey-patches jax.random with a
+    key-ignoring hardware RNG for TPU throughput (reference:
+    utils.py:139-158). The PyTorch-ROCm stack's default generator is
+    already the fast on-device Philox path, so there is nothing to
+    patch; kept so reference-style call sites keep working."""
+    return None
+
+
 def confirm(question: str) -> bool:
     while True:
         resp = input(f"{question} (y/n) ").lower()

@@ -32,7 +32,11 @@ from progen_amd.checkpoint import get_checkpoint_fns, numpy_to_tensors, tensors_
 from progen_amd.data import decode_tokens, iterator_from_tfrecords_folder
 from progen_amd.optim import ProGenAdamW
 from progen_amd.parallel import DistributedTrainer, init_distributed, is_distributed
-from progen_amd.utils import compute_loss, confirm, exists, sample
+from progen_amd.utils import (compute_loss, confirm, exists, load_dotenv,
+                              sample, set_hardware_rng_)
+
+load_dotenv()        # reference: train.py:1-2
+set_hardware_rng_()  # reference: train.py:32 (no-op on ROCm, see utils)
 
 
 def _wandb(wandb_off):
@@ -205,9 +209,12 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
         # last micro-batch's
         loss_val = ddp.all_reduce_scalar(loss).item()
         if is_main:
+            toks_per_sec = effective_batch_size * seq_len / step_times[-1]
             print(f'loss: {loss_val}')
+            if i % 10 == 0:
+                print(f'tokens/sec (whole job): {toks_per_sec:.0f}')
             if wandb is not None:
-                wandb.log({'loss': loss_val})
+                wandb.log({'loss': loss_val, 'tokens_per_sec': toks_per_sec})
 
         if i % checkpoint_every == 0 and is_main:
             package = {
