@@ -75,7 +75,19 @@ def main():
     data[:, 0] = 0
     data[:, -8:] = 0  # pad tail so the EOS-mask path is exercised
 
+    graphed = None
+    if on_gpu and os.environ.get("PROGEN_NO_GRAPH") != "1":
+        from progen_amd.runtime import GraphedTrainStep
+        try:
+            graphed = GraphedTrainStep(module, optim, ddp, B, N, device)
+        except Exception as e:  # noqa: BLE001 — eager fallback, report it
+            import sys
+            print(f"[bench] hipGraph capture failed ({e}); eager fallback",
+                  file=sys.stderr)
+
     def step():
+        if graphed is not None:
+            return graphed.run(data)
         loss = compute_loss(module, data)
         loss.backward()
         ddp.finish_backward()

@@ -11,6 +11,14 @@
 
 #define ADAMW_BLOCK 256
 
+// step counter lives ON DEVICE so the kernel computes its own bias
+// correction — required for hipGraph capture (a host-baked step would be
+// frozen at capture time). step_inc_kernel runs just before on the same
+// stream.
+__global__ void step_inc_kernel(int* step) {
+  if (threadIdx.x == 0 && blockIdx.x == 0) *step += 1;
+}
+
 template <bool IS_BF16>
 __global__ __launch_bounds__(ADAMW_BLOCK) void fused_adamw_kernel(
     float* __restrict__ master, void* __restrict__ params,
@@ -18,9 +26,12 @@ __global__ __launch_bounds__(ADAMW_BLOCK) void fused_adamw_kernel(
     float* __restrict__ exp_avg_sq, const long long* __restrict__ starts,
     const long long* __restrict__ ends, const int* __restrict__ decay_flags,
     int nchunks, float lr, float b1, float b2, float eps, float wd,
-    float bc1, float bc2, float grad_scale,
+    const int* __restrict__ step_ptr, float grad_scale,
     const float* __restrict__ clip_coef) {
   const float clip = *clip_coef;
+  const int step = *step_ptr;
+  const float bc1 = 1.f - powf(b1, (float)step);
+  const float bc2 = 1.f - powf(b2, (float)step);
   for (int c = blockIdx.x; c < nchunks; c += gridDim.x) {
     const long long s = starts[c];
     const long long e = ends[c];
@@ -41,26 +52,74 @@ __global__ __launch_bounds__(ADAMW_BLOCK) void fused_adamw_kernel(
   }
 }
 
+// grad norm: one-pass sum of squares over the flat grad buffer (fp32
+// accumulation), partials combined with a device-scope atomicAdd — the
+// clip coefficient then stays on device (no host sync, no fp32 copy of
+// the 2.7 GB bf16 grad buffer).
+template <bool IS_BF16>
+__global__ __launch_bounds__(ADAMW_BLOCK) void grad_sumsq_kernel(
+    const void* __restrict__ grads, float* __restrict__ out, long long nv) {
+  constexpr int VLEN = IS_BF16 ? 8 : 4;
+  float acc = 0.f;
+  for (long long i = blockIdx.x * (long long)ADAMW_BLOCK + threadIdx.x;
+       i < nv; i += (long long)gridDim.x * ADAMW_BLOCK) {
+    if (IS_BF16) {
+      bf16x8 v = ((const bf16x8*)grads)[i];
+#pragma unroll
+      for (int j = 0; j < VLEN; ++j) {
+        float f = bf2f(((short*)&v)[j]);
+        acc += f * f;
+      }
+    } else {
+      f32x4 v = ((const f32x4*)grads)[i];
+#pragma unroll
+      for (int j = 0; j < VLEN; ++j) acc += ((float*)&v)[j] * ((float*)&v)[j];
+    }
+  }
+  acc = wave_sum(acc);
+  __shared__ float red[ADAMW_BLOCK / WAVE];
+  const int wid = threadIdx.x / WAVE, lane = threadIdx.x % WAVE;
+  if (lane == 0) red[wid] = acc;
+  __syncthreads();
+  if (threadIdx.x == 0) {
+    float t = 0.f;
+#pragma unroll
+    for (int w = 0; w < ADAMW_BLOCK / WAVE; ++w) t += red[w];
+    atomicAdd(out, t);
+  }
+}
+
 extern "C" {
+
+void grad_sumsq_launch(const void* grads, float* out, long long numel,
+                       bool is_bf16, hipStream_t stream) {
+  long long nv = numel / (is_bf16 ? 8 : 4);
+  int grid = (int)((nv + ADAMW_BLOCK - 1) / ADAMW_BLOCK);
+  if (grid > 2048) grid = 2048;
+  if (is_bf16)
+    grad_sumsq_kernel<true><<<grid, ADAMW_BLOCK, 0, stream>>>(grads, out, nv);
+  else
+    grad_sumsq_kernel<false><<<grid, ADAMW_BLOCK, 0, stream>>>(grads, out, nv);
+}
 
 void fused_adamw_launch(float* master, void* params, const void* grads,
                         float* exp_avg, float* exp_avg_sq,
                         const long long* starts, const long long* ends,
                         const int* decay_flags, int nchunks, float lr,
-                        float b1, float b2, float eps, float wd, int step,
-                        float grad_scale, const float* clip_coef,
-                        bool is_bf16, hipStream_t stream) {
-  float bc1 = 1.f - powf(b1, (float)step);
-  float bc2 = 1.f - powf(b2, (float)step);
+                        float b1, float b2, float eps, float wd,
+                        int* step_dev, float grad_scale,
+                        const float* clip_coef, bool is_bf16,
+                        hipStream_t stream) {
+  step_inc_kernel<<<1, 1, 0, stream>>>(step_dev);
   int grid = nchunks < 2048 ? nchunks : 2048;
   if (is_bf16)
     fused_adamw_kernel<true><<<grid, ADAMW_BLOCK, 0, stream>>>(
         master, params, grads, exp_avg, exp_avg_sq, starts, ends, decay_flags,
-        nchunks, lr, b1, b2, eps, wd, bc1, bc2, grad_scale, clip_coef);
+        nchunks, lr, b1, b2, eps, wd, step_dev, grad_scale, clip_coef);
   else
     fused_adamw_kernel<false><<<grid, ADAMW_BLOCK, 0, stream>>>(
         master, params, grads, exp_avg, exp_avg_sq, starts, ends, decay_flags,
-        nchunks, lr, b1, b2, eps, wd, bc1, bc2, grad_scale, clip_coef);
+        nchunks, lr, b1, b2, eps, wd, step_dev, grad_scale, clip_coef);
 }
 
 }  // extern "C"

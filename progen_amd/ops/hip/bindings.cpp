@@ -144,12 +144,24 @@ at::Tensor ce_bwd(const at::Tensor& dnll, const at::Tensor& logits,
 // fused adamw
 // ---------------------------------------------------------------------------
 
+at::Tensor grad_sumsq(const at::Tensor& grads) {
+  TORCH_CHECK(grads.is_cuda() && grads.is_contiguous());
+  bool bf = grads.scalar_type() == at::kBFloat16;
+  TORCH_CHECK(grads.numel() % (bf ? 8 : 4) == 0);
+  auto out = at::zeros({1}, grads.options().dtype(at::kFloat));
+  grad_sumsq_launch(grads.data_ptr(), out.data_ptr<float>(), grads.numel(), bf,
+                    cur_stream());
+  return out;
+}
+
 void fused_adamw(at::Tensor& master, at::Tensor& params,
                  const at::Tensor& grads, at::Tensor& exp_avg,
                  at::Tensor& exp_avg_sq, const at::Tensor& chunk_starts,
                  const at::Tensor& chunk_ends, const at::Tensor& chunk_decay,
                  double lr, double b1, double b2, double eps, double wd,
-                 long step, double grad_scale, const at::Tensor& clip_coef) {
+                 at::Tensor& step_dev, double grad_scale,
+                 const at::Tensor& clip_coef) {
+  TORCH_CHECK(step_dev.scalar_type() == at::kInt && step_dev.is_cuda());
   TORCH_CHECK(master.is_cuda() && master.scalar_type() == at::kFloat);
   bool bf = params.scalar_type() == at::kBFloat16;
   fused_adamw_launch(master.data_ptr<float>(), params.data_ptr(),
@@ -158,8 +170,8 @@ void fused_adamw(at::Tensor& master, at::Tensor& params,
                      (const long long*)chunk_starts.data_ptr<long>(), (const long long*)chunk_ends.data_ptr<long>(),
                      chunk_decay.data_ptr<int>(), chunk_starts.size(0),
                      (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
-                     (int)step, (float)grad_scale, clip_coef.data_ptr<float>(),
-                     bf, cur_stream());
+                     step_dev.data_ptr<int>(), (float)grad_scale,
+                     clip_coef.data_ptr<float>(), bf, cur_stream());
 }
 
 // ---------------------------------------------------------------------------
@@ -211,6 +223,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("gelu_bwd", &gelu_bwd, "GELU backward");
   m.def("ce_fwd", &ce_fwd, "fused CE forward (nll, lse)");
   m.def("ce_bwd", &ce_bwd, "fused CE backward");
+  m.def("grad_sumsq", &grad_sumsq, "sum of squares of flat grads");
   m.def("fused_adamw", &fused_adamw, "fused clip+AdamW over flat space");
   m.def("attn_fwd", &attn_fwd, "fused local attention forward");
   m.def("attn_bwd", &attn_bwd, "fused local attention backward");

@@ -28,13 +28,16 @@ void ce_bwd_launch(const float* dnll, const void* logits,
                    const long long* targets, const float* lse, void* dlogits,
                    long long R, int V, bool is_bf16, hipStream_t stream);
 
+void grad_sumsq_launch(const void* grads, float* out, long long numel,
+                       bool is_bf16, hipStream_t stream);
 void fused_adamw_launch(float* master, void* params, const void* grads,
                         float* exp_avg, float* exp_avg_sq,
                         const long long* starts, const long long* ends,
                         const int* decay_flags, int nchunks, float lr,
-                        float b1, float b2, float eps, float wd, int step,
-                        float grad_scale, const float* clip_coef,
-                        bool is_bf16, hipStream_t stream);
+                        float b1, float b2, float eps, float wd,
+                        int* step_dev, float grad_scale,
+                        const float* clip_coef, bool is_bf16,
+                        hipStream_t stream);
 
 void attn_fwd_launch(const void* qkv, const float* rsin, const float* rcos,
                      void* out, float* lse, int B, int N, int H, int wsz,

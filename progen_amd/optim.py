@@ -108,6 +108,11 @@ class ProGenAdamW:
         self.step_count = 0
         self._micro = 0
         self._update_acc: Optional[torch.Tensor] = None  # apply_every mode
+        # device-side step counter: the fused kernel increments it and
+        # computes bias correction from it, so hipGraph replays keep
+        # correct Adam bias correction without host involvement
+        self.step_dev = torch.zeros(1, dtype=torch.int32, device=flat.device) \
+            if flat.is_cuda else None
 
         # per-chunk decay flags: weight decay only on ndim>1 params
         # (reference: train.py:115 exclude_norm_and_bias_params)
@@ -206,19 +211,28 @@ class ProGenAdamW:
 
     def _step_hip(self, grad_scale: float) -> None:
         C = dispatch.ext()
-        clip_coef = self._clip_coef(self.space.flat_grad.float() * grad_scale)
+        if self.max_grad_norm is None:
+            clip_coef = torch.ones(1, device=self.master.device)
+        else:
+            # norm(scale*g) = scale*norm(g): fused sumsq kernel, no fp32
+            # grad copy, clip coefficient stays on device (no host sync)
+            norm = C.grad_sumsq(self.space.flat_grad).sqrt_() * grad_scale
+            clip_coef = self.max_grad_norm / torch.clamp_min(norm, self.max_grad_norm)
         C.fused_adamw(
             self.master, self.space.flat, self.space.flat_grad,
             self.exp_avg, self.exp_avg_sq,
             self.chunk_starts, self.chunk_ends, self.chunk_decay,
             float(self.lr), float(self.betas[0]), float(self.betas[1]),
-            float(self.eps), float(self.weight_decay), int(self.step_count),
+            float(self.eps), float(self.weight_decay), self.step_dev,
             float(grad_scale), clip_coef,
         )
 
     # -- checkpoint state ----------------------------------------------------
 
     def state_dict(self) -> Dict:
+        if self.step_dev is not None:
+            # graph replays advance only the device counter; sync back
+            self.step_count = max(self.step_count, int(self.step_dev.item()))
         return {
             "step_count": self.step_count,
             "micro": self._micro,
@@ -229,6 +243,8 @@ class ProGenAdamW:
 
     def load_state_dict(self, sd: Dict) -> None:
         self.step_count = int(sd["step_count"])
+        if self.step_dev is not None:
+            self.step_dev.fill_(self.step_count)
         self._micro = int(sd.get("micro", 0))
         with torch.no_grad():
             self.master.copy_(torch.as_tensor(sd["master"]).to(self.master.device))
