@@ -80,9 +80,14 @@ def main():
         from progen_amd.runtime import GraphedTrainStep
         try:
             graphed = GraphedTrainStep(module, optim, ddp, B, N, device)
-        except Exception as e:  # noqa: BLE001 — eager fallback, report it
+            if rank == 0:
+                import sys
+                print("[bench] hipGraph step captured", file=sys.stderr)
+        except Exception:  # noqa: BLE001 — eager fallback, report it
             import sys
-            print(f"[bench] hipGraph capture failed ({e}); eager fallback",
+            import traceback
+            traceback.print_exc()
+            print("[bench] hipGraph capture failed; eager fallback",
                   file=sys.stderr)
 
     def step():
