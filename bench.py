@@ -48,6 +48,8 @@ def main():
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
     args = p.parse_args()
 
+    from progen_amd.tuning import enable_tuned_gemms
+    enable_tuned_gemms()
     local_rank = init_distributed()
     world = int(os.environ.get("WORLD_SIZE", "1"))
     rank = int(os.environ.get("RANK", "0"))

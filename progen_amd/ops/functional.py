@@ -57,17 +57,21 @@ class _LocalAttnFn(torch.autograd.Function):
     @staticmethod
     def forward(ctx, qkv, sin, cos, heads: int, window_size: int):
         C = dispatch.ext()
-        out, lse = C.attn_fwd(qkv, sin, cos, int(heads), int(window_size))
-        ctx.save_for_backward(qkv, sin, cos, out, lse)
+        # pre-rotation pass (rotary on q, k AND v — progen.py:87); the
+        # attention kernels then stage pure bf16 copies
+        qkv_rot = C.rope_qkv(qkv.contiguous(), sin, cos)
+        out, lse = C.attn_fwd(qkv_rot, int(heads), int(window_size))
+        ctx.save_for_backward(qkv_rot, sin, cos, out, lse)
         ctx.heads = int(heads)
         ctx.window_size = int(window_size)
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        qkv, sin, cos, out, lse = ctx.saved_tensors
+        qkv_rot, sin, cos, out, lse = ctx.saved_tensors
         C = dispatch.ext()
-        dqkv = C.attn_bwd(dout.contiguous(), qkv, sin, cos, out, lse,
+        # bwd finalize applies the inverse rotation (rotary is linear)
+        dqkv = C.attn_bwd(dout.contiguous(), qkv_rot, sin, cos, out, lse,
                           ctx.heads, ctx.window_size)
         return dqkv, None, None, None, None
 

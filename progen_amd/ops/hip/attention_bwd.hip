@@ -66,8 +66,7 @@ __device__ __forceinline__ void load_rope(const float* rsin,
 
 __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
     const short* __restrict__ dout,  // (B, N, H*DH) bf16
-    const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16
-    const float* __restrict__ rsin, const float* __restrict__ rcos,
+    const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16, PRE-ROTATED
     const short* __restrict__ out,   // (B, N, H*DH) bf16 (fwd output)
     const float* __restrict__ lse,   // (B, H, N)
     float* __restrict__ dacc,        // (B, N, 3*H*DH) fp32, zero-init
@@ -127,14 +126,10 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           const int d0 = ks * 32 + 8 * l4;
           bf16x8 v = *(const bf16x8*)(qkv + qkv_bn + (long long)row * HD3 +
                                       q_off + d0);
-          float x[8], sv[8], cv[8];
+          bf16x8 o;  // pre-rotated input; fold in the softmax scale
 #pragma unroll
-          for (int j = 0; j < 8; ++j) x[j] = bf2f(((short*)&v)[j]);
-          load_rope(rsin, rcos, row, d0, sv, cv);
-          rope8(x, sv, cv);
-          bf16x8 o;
-#pragma unroll
-          for (int j = 0; j < 8; ++j) ((short*)&o)[j] = f2bf(x[j]);
+          for (int j = 0; j < 8; ++j)
+            ((short*)&o)[j] = f2bf(bf2f(((short*)&v)[j]) * scale);
           qfrag[m][ks] = o;
         }
       }
@@ -152,15 +147,12 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           bf16x8 qv = *(const bf16x8*)(qkv + gq + d0);
           bf16x8 ov = *(const bf16x8*)(out + go + d0);
           bf16x8 dov = *(const bf16x8*)(dout + go + d0);
-          float x[8], sv[8], cv[8];
-#pragma unroll
-          for (int j = 0; j < 8; ++j) x[j] = bf2f(((short*)&qv)[j]);
-          load_rope(rsin, rcos, q0 + row, d0, sv, cv);
-          rope8(x, sv, cv);
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             const int d = d0 + j;
-            *(short*)(qt_lds + d * 128 + swz(d, row * 2)) = f2bf(x[j] * scale);
+            // qt holds SCALED q' (dK = dS^T q_s)
+            *(short*)(qt_lds + d * 128 + swz(d, row * 2)) =
+                f2bf(bf2f(((short*)&qv)[j]) * scale);
             *(short*)(dot_lds + d * 128 + swz(d, row * 2)) = ((short*)&dov)[j];
             dsum += bf2f(((short*)&ov)[j]) * bf2f(((short*)&dov)[j]);
           }
@@ -181,36 +173,27 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
         const int key = flat >> 3;
         const int d0 = (flat & 7) * 8;
         const int kpos = (window - 1) * wsz + t * KT + key;
-        float kx[8], vx[8];
+        bf16x8 kb, vb;
         if (kpos >= 0) {
-          bf16x8 kvec = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 +
-                                         k_off + d0);
-          bf16x8 vvec = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 +
-                                         v_off + d0);
-          float sv[8], cv[8];
-          load_rope(rsin, rcos, kpos, d0, sv, cv);
-#pragma unroll
-          for (int j = 0; j < 8; ++j) kx[j] = bf2f(((short*)&kvec)[j]);
-#pragma unroll
-          for (int j = 0; j < 8; ++j) vx[j] = bf2f(((short*)&vvec)[j]);
-          rope8(kx, sv, cv);
-          rope8(vx, sv, cv);  // rotary on V too (progen.py:87)
+          kb = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 + k_off +
+                                d0);
+          vb = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 + v_off +
+                                d0);
         } else {
 #pragma unroll
-          for (int j = 0; j < 8; ++j) kx[j] = vx[j] = 0.f;
-        }
-        bf16x8 kb, vb;
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          ((short*)&kb)[j] = f2bf(kx[j] * scale);
-          ((short*)&vb)[j] = f2bf(vx[j]);
+          for (int j = 0; j < 8; ++j) {
+            ((short*)&kb)[j] = 0;
+            ((short*)&vb)[j] = 0;
+          }
         }
         *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kb;
         *(bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2)) = vb;
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const int d = d0 + j;
-          *(short*)(kt_lds + d * 128 + swz(d, key * 2)) = ((short*)&kb)[j];
+          // kt holds SCALED k' (dQ = dS k_s)
+          *(short*)(kt_lds + d * 128 + swz(d, key * 2)) =
+              f2bf(bf2f(((short*)&kb)[j]) * scale);
         }
       }
       __syncthreads();
@@ -218,7 +201,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
       if (active && t <= max_tile) {
         const int kb = t * KT;
 
-        // ---- S = q' k_s^T ----
+        // ---- S = q_s k'^T ----
         f32x4 s[4][4];
 #pragma unroll
         for (int m = 0; m < 4; ++m)
@@ -484,8 +467,8 @@ void attn_bwd_launch(const void* dout, const void* qkv, const float* rsin,
   dim3 grid(N / wsz, H, B), block(ATTN_BLOCK);
   size_t lds = 24576 + 131072 + 2048;  // 154 KiB
   attn_bwd_kernel<<<grid, block, lds, stream>>>(
-      (const short*)dout, (const short*)qkv, rsin, rcos, (const short*)out,
-      lse, dacc, B, N, H, wsz);
+      (const short*)dout, (const short*)qkv, (const short*)out, lse, dacc, B,
+      N, H, wsz);
   long long total = (long long)B * N * 3 * H * (DH / 8);
   int fin_grid = (int)((total + 255) / 256);
   if (fin_grid > 2048) fin_grid = 2048;

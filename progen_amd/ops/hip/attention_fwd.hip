@@ -1,8 +1,10 @@
 // Fused local-window attention, forward (reference: progen.py:83-103).
 //
+// Input is the PRE-ROTATED qkv (ops/hip/rope_qkv.hip applies the
+// reference's rotary-on-q/k/v quirk, progen.py:87, in a separate
+// memory-bound pass), so this kernel's staging is a pure bf16 copy.
+//
 // One kernel fuses, per (batch, head, window):
-//   - GPT-J interleaved rotary applied to q, k AND v (quirk preserved,
-//     reference: progen.py:87) during register/LDS staging;
 //   - the one-window-lookback key band [prev window ‖ own window] with
 //     window 0's lookback keys ZERO and UNMASKED (progen.py:90-96):
 //     zero K rows give logit 0 into the softmax denominator and zero V
@@ -16,9 +18,12 @@
 // 64-row Q chunk (round-robins chunks when wsz > 256). Per 64-key tile:
 // K staged in LDS [key][dh] and V transposed [dh][key], both
 // XOR-swizzled (byte ^= (row&7)<<4) so the mfma_f32_16x16x32_bf16
-// B-fragment ds_read_b128s are <=2-way bank conflicted. Q fragments and
-// the fp32 softmax state live in registers for the whole block.
-// Saves per-row logsumexp (B, h, N) fp32 for the backward's recompute.
+// B-fragment ds_read_b128s are <=2-way bank conflicted. Staging uses the
+// async-STAGE split (T14): tile t+1's global loads are issued before
+// tile t's compute, the LDS writes land after the barrier — hiding HBM
+// latency at 1 wave/SIMD occupancy. Q fragments (pre-scaled) and the
+// fp32 softmax state live in registers for the whole block. Saves
+// per-row logsumexp (B, h, N) fp32 for the backward's recompute.
 //
 // dim_head is fixed at 64 (the ProGen family's head size).
 
@@ -30,37 +35,18 @@
 #define ATTN_BLOCK (ATTN_WAVES * WAVE)
 #define NEG_INF (-1e30f)
 
-using f32x4v = f32x4;
-
 __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return (byte_in_row ^ ((row & 7) << 4));
 }
 
-// rotary: interleaved pairs (progen.py:30-41); sin/cos tables are
-// repeat-interleaved so sin[2i] == sin[2i+1]
-__device__ __forceinline__ void rope8(float* x, const float* sinv,
-                                      const float* cosv) {
-#pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    float x0 = x[2 * p], x1 = x[2 * p + 1];
-    float s = sinv[2 * p], c = cosv[2 * p];
-    x[2 * p] = x0 * c - x1 * s;
-    x[2 * p + 1] = x1 * c + x0 * s;
-  }
-}
-
 __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
-    const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16
-    const float* __restrict__ rsin,  // (N, DH)
-    const float* __restrict__ rcos,  // (N, DH)
+    const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16, PRE-ROTATED
     short* __restrict__ out,         // (B, N, H*DH) bf16
     float* __restrict__ lse_out,     // (B, H, N)
     int B, int N, int H, int wsz) {
   const int window = blockIdx.x;
   const int head = blockIdx.y;
   const int batch = blockIdx.z;
-  const int nwin = N / wsz;
-  (void)nwin;
 
   const int lane = threadIdx.x % WAVE;
   const int wid = threadIdx.x / WAVE;
@@ -73,25 +59,29 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
   const int k_off = H * DH + head * DH;
   const int v_off = 2 * H * DH + head * DH;
 
-  // LDS: K tile [KT][DH] swizzled + V^T tile [DH][KT] swizzled + per-wave
-  // P [64][KT] swizzled
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_lds = smem;                          // KT*DH*2 = 8 KiB
-  char* v_lds = smem + KT * DH * 2;            // 8 KiB
+  char* k_lds = smem;                          // 8 KiB
+  char* v_lds = smem + KT * DH * 2;            // 8 KiB (V^T image)
   char* p_lds = smem + 2 * KT * DH * 2 + wid * 64 * KT * 2;  // 8 KiB/wave
 
   const float scale = rsqrtf((float)DH);
   const int tiles = 2 * wsz / KT;
-  const int chunks = wsz / 64;  // 64-row q chunks in this window
+  const int chunks = wsz / 64;
   const int rounds = (chunks + ATTN_WAVES - 1) / ATTN_WAVES;
+
+  // staging geometry: 512 (key, d0) units per tile, 2 per thread
+  const int su_key[2] = {(int)threadIdx.x >> 3,
+                         (int)(threadIdx.x + ATTN_BLOCK) >> 3};
+  const int su_d0[2] = {((int)threadIdx.x & 7) * 8,
+                        (((int)threadIdx.x + ATTN_BLOCK) & 7) * 8};
 
   for (int round = 0; round < rounds; ++round) {
     const int chunk = round * ATTN_WAVES + wid;
     const bool active = chunk < chunks;
-    const int chunk_off = chunk * 64;  // q-row offset within the window
-    const int q0 = window * wsz + chunk_off;  // global q row of this wave
+    const int chunk_off = chunk * 64;
+    const int q0 = window * wsz + chunk_off;
 
-    // ---- load Q fragments (+rotary, *scale) into registers ----
+    // ---- Q fragments (pre-rotated; fold in the softmax scale) ----
     bf16x8 qfrag[4][2];
     if (active) {
 #pragma unroll
@@ -102,23 +92,15 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
           const int d0 = ks * 32 + 8 * l4;
           bf16x8 v = *(const bf16x8*)(qkv + qkv_bn + (long long)row * HD3 +
                                       q_off + d0);
-          float x[8], sv[8], cv[8];
-#pragma unroll
-          for (int j = 0; j < 8; ++j) x[j] = bf2f(((short*)&v)[j]);
-          *(f32x4*)(sv) = *(const f32x4*)(rsin + (long long)row * DH + d0);
-          *(f32x4*)(sv + 4) = *(const f32x4*)(rsin + (long long)row * DH + d0 + 4);
-          *(f32x4*)(cv) = *(const f32x4*)(rcos + (long long)row * DH + d0);
-          *(f32x4*)(cv + 4) = *(const f32x4*)(rcos + (long long)row * DH + d0 + 4);
-          rope8(x, sv, cv);
           bf16x8 o;
 #pragma unroll
-          for (int j = 0; j < 8; ++j) ((short*)&o)[j] = f2bf(x[j] * scale);
+          for (int j = 0; j < 8; ++j)
+            ((short*)&o)[j] = f2bf(bf2f(((short*)&v)[j]) * scale);
           qfrag[m][ks] = o;
         }
       }
     }
 
-    // ---- softmax state ----
     float m_run[4][4], l_run[4][4];
 #pragma unroll
     for (int m = 0; m < 4; ++m)
@@ -127,7 +109,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         m_run[m][r] = NEG_INF;
         l_run[m][r] = 0.f;
       }
-    f32x4 oacc[4][4];  // [m][dh fragment] per-lane 4 rows x 1 col
+    f32x4 oacc[4][4];
 #pragma unroll
     for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -135,51 +117,47 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 
     const int max_tile = active ? ((chunk_off + 63 + wsz) / KT) : -1;
 
-    for (int t = 0; t < tiles; ++t) {
-      // ---- cooperative stage of K tile and V^T tile (+rotary) ----
-      // 256 threads x 16 B = 4 KiB per pass; tile is 8 KiB -> 2 passes
-      __syncthreads();
+    // ---- T14 staging: issue loads early, write LDS after barrier ----
+    bf16x8 kreg[2], vreg[2];
+
+    auto issue_loads = [&](int t) {
 #pragma unroll
-      for (int pass = 0; pass < 2; ++pass) {
-        const int flat = pass * ATTN_BLOCK + threadIdx.x;  // 0..511
-        const int key = flat >> 3;           // 0..63 within tile
-        const int d0 = (flat & 7) * 8;       // dh group of 8
-        const int kpos_band = t * KT + key;  // 0..2wsz
-        const int kpos = (window - 1) * wsz + kpos_band;  // global key pos
-        float kx[8], vx[8];
+      for (int u = 0; u < 2; ++u) {
+        const int kpos = (window - 1) * wsz + t * KT + su_key[u];
         if (kpos >= 0) {
-          bf16x8 kvec = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 +
-                                         k_off + d0);
-          bf16x8 vvec = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 +
-                                         v_off + d0);
-          float sv[8], cv[8];
-          *(f32x4*)(sv) = *(const f32x4*)(rsin + (long long)kpos * DH + d0);
-          *(f32x4*)(sv + 4) = *(const f32x4*)(rsin + (long long)kpos * DH + d0 + 4);
-          *(f32x4*)(cv) = *(const f32x4*)(rcos + (long long)kpos * DH + d0);
-          *(f32x4*)(cv + 4) = *(const f32x4*)(rcos + (long long)kpos * DH + d0 + 4);
-#pragma unroll
-          for (int j = 0; j < 8; ++j) kx[j] = bf2f(((short*)&kvec)[j]);
-#pragma unroll
-          for (int j = 0; j < 8; ++j) vx[j] = bf2f(((short*)&vvec)[j]);
-          rope8(kx, sv, cv);
-          rope8(vx, sv, cv);  // rotary on V too (progen.py:87)
+          const long long base = qkv_bn + (long long)kpos * HD3;
+          kreg[u] = *(const bf16x8*)(qkv + base + k_off + su_d0[u]);
+          vreg[u] = *(const bf16x8*)(qkv + base + v_off + su_d0[u]);
         } else {
-          // window 0 lookback: the zero-pad window (progen.py:90-91)
 #pragma unroll
-          for (int j = 0; j < 8; ++j) kx[j] = vx[j] = 0.f;
+          for (int j = 0; j < 8; ++j) {
+            ((short*)&kreg[u])[j] = 0;
+            ((short*)&vreg[u])[j] = 0;
+          }
         }
-        bf16x8 kb;
+      }
+    };
+
+    auto write_lds = [&]() {
 #pragma unroll
-        for (int j = 0; j < 8; ++j) ((short*)&kb)[j] = f2bf(kx[j]);
-        *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kb;
-        // V^T: scatter 8 bf16 to [d][key]
+      for (int u = 0; u < 2; ++u) {
+        const int key = su_key[u];
+        const int d0 = su_d0[u];
+        *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const int d = d0 + j;
-          *(short*)(v_lds + d * 128 + swz(d, key * 2)) = f2bf(vx[j]);
+          *(short*)(v_lds + d * 128 + swz(d, key * 2)) = ((short*)&vreg[u])[j];
         }
       }
-      __syncthreads();
+    };
+
+    issue_loads(0);
+    write_lds();
+    __syncthreads();
+
+    for (int t = 0; t < tiles; ++t) {
+      if (t + 1 < tiles) issue_loads(t + 1);  // in flight during compute
 
       if (active && t <= max_tile) {
         // ---- S = Q K^T ----
@@ -204,7 +182,6 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 
         // ---- mask + online softmax ----
         const int kb = t * KT;
-        float tile_max[4][4];
 #pragma unroll
         for (int m = 0; m < 4; ++m) {
 #pragma unroll
@@ -220,14 +197,8 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
               mx = fmaxf(mx, v);
             }
             mx = group16_max(mx);
-            tile_max[m][r] = mx;
-          }
-        }
-#pragma unroll
-        for (int m = 0; m < 4; ++m) {
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const float mnew = fmaxf(m_run[m][r], tile_max[m][r]);
+
+            const float mnew = fmaxf(m_run[m][r], mx);
             const float alpha =
                 (m_run[m][r] == NEG_INF) ? 0.f : __expf(m_run[m][r] - mnew);
             float psum = 0.f;
@@ -241,14 +212,13 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
             psum = group16_sum(psum);
             l_run[m][r] = l_run[m][r] * alpha + psum;
             m_run[m][r] = mnew;
-            // rescale O rows
 #pragma unroll
             for (int d = 0; d < 4; ++d)
               ((float*)&oacc[m][d])[r] *= alpha;
           }
         }
 
-        // ---- P -> bf16 -> LDS (per-wave region) ----
+        // ---- P -> bf16 -> per-wave LDS ----
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -261,9 +231,6 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
                   f2bf(((float*)&s[m][n])[r]);
             }
           }
-        // wave-local LDS write->read (no cross-wave sharing of p_lds):
-        // drain DS writes before the fragment reads; a block barrier is
-        // illegal here (divergent path), a wave-local wait suffices.
         asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
 
         // ---- O += P V ----
@@ -283,6 +250,12 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
             }
           }
         }
+      }
+
+      __syncthreads();  // all waves done reading LDS tile t
+      if (t + 1 < tiles) {
+        write_lds();    // compiler inserts the vmcnt wait at first use
+        __syncthreads();
       }
     }
 
@@ -308,18 +281,18 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         }
       }
     }
+    if (round + 1 < rounds) __syncthreads();
   }
 }
 
 extern "C" {
 
-void attn_fwd_launch(const void* qkv, const float* rsin, const float* rcos,
-                     void* out, float* lse, int B, int N, int H, int wsz,
-                     hipStream_t stream) {
+void attn_fwd_launch(const void* qkv_rot, void* out, float* lse, int B, int N,
+                     int H, int wsz, hipStream_t stream) {
   dim3 grid(N / wsz, H, B), block(ATTN_BLOCK);
   size_t lds = (size_t)(2 * KT * DH * 2) + (size_t)ATTN_WAVES * 64 * KT * 2;
   attn_fwd_kernel<<<grid, block, lds, stream>>>(
-      (const short*)qkv, rsin, rcos, (short*)out, lse, B, N, H, wsz);
+      (const short*)qkv_rot, (short*)out, lse, B, N, H, wsz);
 }
 
 }  // extern "C"

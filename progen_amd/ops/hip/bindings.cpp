@@ -178,23 +178,35 @@ void fused_adamw(at::Tensor& master, at::Tensor& params,
 // local attention
 // ---------------------------------------------------------------------------
 
-std::vector<at::Tensor> attn_fwd(const at::Tensor& qkv, const at::Tensor& rsin,
-                                 const at::Tensor& rcos, long heads,
-                                 long window) {
+at::Tensor rope_qkv(const at::Tensor& qkv, const at::Tensor& rsin,
+                    const at::Tensor& rcos) {
   TORCH_CHECK(qkv.is_cuda() && qkv.is_contiguous() && qkv.dim() == 3);
-  TORCH_CHECK(qkv.scalar_type() == at::kBFloat16,
-              "attn_fwd: bf16 only (MFMA path)");
+  TORCH_CHECK(qkv.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(rsin.scalar_type() == at::kFloat && rsin.is_contiguous());
   const int B = qkv.size(0), N = qkv.size(1);
+  const int H = (int)(qkv.size(2) / (3 * 64));
+  TORCH_CHECK(qkv.size(2) == 3LL * H * 64, "rope_qkv: dim_head must be 64");
+  auto qkv_rot = at::empty_like(qkv);
+  rope_qkv_launch(qkv.data_ptr(), rsin.data_ptr<float>(),
+                  rcos.data_ptr<float>(), qkv_rot.data_ptr(), B, N, H,
+                  cur_stream());
+  return qkv_rot;
+}
+
+std::vector<at::Tensor> attn_fwd(const at::Tensor& qkv_rot, long heads,
+                                 long window) {
+  TORCH_CHECK(qkv_rot.is_cuda() && qkv_rot.is_contiguous() && qkv_rot.dim() == 3);
+  TORCH_CHECK(qkv_rot.scalar_type() == at::kBFloat16,
+              "attn_fwd: bf16 only (MFMA path)");
+  const int B = qkv_rot.size(0), N = qkv_rot.size(1);
   const int H = (int)heads, wsz = (int)window;
-  TORCH_CHECK(qkv.size(2) == 3LL * H * 64, "attn: dim_head must be 64");
+  TORCH_CHECK(qkv_rot.size(2) == 3LL * H * 64, "attn: dim_head must be 64");
   TORCH_CHECK(N % wsz == 0 && wsz % 64 == 0,
               "attn: seq divisible by window, window divisible by 64");
-  TORCH_CHECK(rsin.scalar_type() == at::kFloat && rsin.is_contiguous());
-  auto out = at::empty({B, N, (long)H * 64}, qkv.options());
-  auto lse = at::empty({B, (long)H, N}, qkv.options().dtype(at::kFloat));
-  attn_fwd_launch(qkv.data_ptr(), rsin.data_ptr<float>(),
-                  rcos.data_ptr<float>(), out.data_ptr(),
-                  lse.data_ptr<float>(), B, N, H, wsz, cur_stream());
+  auto out = at::empty({B, N, (long)H * 64}, qkv_rot.options());
+  auto lse = at::empty({B, (long)H, N}, qkv_rot.options().dtype(at::kFloat));
+  attn_fwd_launch(qkv_rot.data_ptr(), out.data_ptr(), lse.data_ptr<float>(),
+                  B, N, H, wsz, cur_stream());
   return {out, lse};
 }
 
@@ -225,6 +237,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_bwd", &ce_bwd, "fused CE backward");
   m.def("grad_sumsq", &grad_sumsq, "sum of squares of flat grads");
   m.def("fused_adamw", &fused_adamw, "fused clip+AdamW over flat space");
+  m.def("rope_qkv", &rope_qkv, "pre-rotation of qkv (rotary on q,k,v)");
   m.def("attn_fwd", &attn_fwd, "fused local attention forward");
   m.def("attn_bwd", &attn_bwd, "fused local attention backward");
 }

@@ -39,9 +39,10 @@ void fused_adamw_launch(float* master, void* params, const void* grads,
                         const float* clip_coef, bool is_bf16,
                         hipStream_t stream);
 
-void attn_fwd_launch(const void* qkv, const float* rsin, const float* rcos,
-                     void* out, float* lse, int B, int N, int H, int wsz,
-                     hipStream_t stream);
+void rope_qkv_launch(const void* qkv, const float* rsin, const float* rcos,
+                     void* qkv_rot, int B, int N, int H, hipStream_t stream);
+void attn_fwd_launch(const void* qkv_rot, void* out, float* lse, int B, int N,
+                     int H, int wsz, hipStream_t stream);
 void attn_bwd_launch(const void* dout, const void* qkv, const float* rsin,
                      const float* rcos, const void* out, const float* lse,
                      float* dacc, void* dqkv, int B, int N, int H, int wsz,

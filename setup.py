@@ -20,6 +20,7 @@ HIP_SOURCES = [
     "progen_amd/ops/hip/glu.hip",
     "progen_amd/ops/hip/cross_entropy.hip",
     "progen_amd/ops/hip/adamw.hip",
+    "progen_amd/ops/hip/rope_qkv.hip",
     "progen_amd/ops/hip/attention_fwd.hip",
     "progen_amd/ops/hip/attention_bwd.hip",
 ]

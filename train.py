@@ -80,6 +80,8 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
          checkpoint_every, checkpoint_path, checkpoint_keep_n, config_path,
          model_name, prime_length, seq_len, mixed_precision, grad_accum_mode,
          data_path, wandb_off, wandb_project_name, new, max_steps, yes):
+    from progen_amd.tuning import enable_tuned_gemms
+    enable_tuned_gemms()
     local_rank = init_distributed()
     world = int(os.environ.get('WORLD_SIZE', '1'))
     rank = int(os.environ.get('RANK', '0'))
