@@ -182,19 +182,71 @@ def cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
 # SGU spatial gating
 # ---------------------------------------------------------------------------
 
+_TRI_CACHE = {}
+
+
+def _tri_tiles(n: int, device) -> tuple:
+    """Lower-triangle 64x64 tile index lists for sgu_dw (cached per n)."""
+    key = (n, str(device))
+    if key not in _TRI_CACHE:
+        ms, ks = [], []
+        for mt in range(n // 64):
+            for kt in range(mt + 1):
+                ms.append(mt)
+                ks.append(kt)
+        _TRI_CACHE[key] = (
+            torch.tensor(ms, dtype=torch.int32, device=device),
+            torch.tensor(ks, dtype=torch.int32, device=device),
+        )
+    return _TRI_CACHE[key]
+
+
+class _SGUFn(torch.autograd.Function):
+    """Causal spatial matmul + bias + gate multiply on the hand-written
+    CDNA4 kernels (ops/hip/sgu.hip)."""
+
+    @staticmethod
+    def forward(ctx, xa, g_ln, w, bias):
+        C = dispatch.ext()
+        out, gate_out = C.sgu_fwd(xa, g_ln, w, bias)
+        ctx.save_for_backward(xa, g_ln, w, gate_out)
+        return out
+
+    @staticmethod
+    def backward(ctx, dy):
+        xa, g_ln, w, gate_out = ctx.saved_tensors
+        C = dispatch.ext()
+        dy = dy.contiguous()
+        dxa = dy * gate_out
+        t = (dy * xa).contiguous()
+        dg_ln = C.sgu_dgate(t, w)
+        n = xa.shape[1]
+        tri_m, tri_k = _tri_tiles(n, xa.device)
+        dw = C.sgu_dw(t, g_ln, tri_m, tri_k).to(w.dtype)
+        db = t.float().sum(dim=(0, 2)).unsqueeze(-1).to(w.dtype)
+        return dxa, dg_ln, dw, db
+
+
 def sgu_gate(x: torch.Tensor, norm_weight: torch.Tensor,
              spatial_weights: torch.Tensor, spatial_biases: torch.Tensor,
              eps: float = 1e-5) -> torch.Tensor:
     """gMLP spatial gating unit core (reference: progen.py:166-183).
 
-    GPU path: fused LN kernel on the gate half + causal (n, n) spatial
-    matmul on hipBLASLt (library GEMM) + elementwise gate-multiply. The
-    tril mask is applied to the weight once per forward (cheap: n^2)."""
+    GPU path: fused LN kernel on the gate half + the hand-written causal
+    spatial-matmul kernels (ops/hip/sgu.hip — the tril mask is baked into
+    the tile iteration). Sequences shorter than 256 (toy configs) use the
+    hipBLASLt composite path."""
     if dispatch.use_hip(x):
         xa, gate = x.chunk(2, dim=-1)
-        gate = ln_shift(gate.contiguous(), norm_weight, shift=False, eps=eps)
+        gate_ln = ln_shift(gate.contiguous(), norm_weight, shift=False, eps=eps)
         n = x.shape[1]
-        w = spatial_weights[:n, :n].tril().to(gate.dtype)
-        gate = torch.einsum("bnd,mn->bmd", gate, w) + spatial_biases[:n].to(gate.dtype)
-        return xa * gate
+        d = xa.shape[-1]
+        if (x.dtype == torch.bfloat16 and n % 256 == 0 and d % 64 == 0):
+            w = spatial_weights[:n, :n].contiguous()
+            return _SGUFn.apply(xa.contiguous(), gate_ln, w,
+                                spatial_biases[:n])
+        w = spatial_weights[:n, :n].tril().to(gate_ln.dtype)
+        gate_ln = torch.einsum("bnd,mn->bmd", gate_ln, w) + \
+            spatial_biases[:n].to(gate_ln.dtype)
+        return xa * gate_ln
     return reference.sgu_gate(x, norm_weight, spatial_weights, spatial_biases, eps)

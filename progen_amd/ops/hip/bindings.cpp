@@ -226,7 +226,48 @@ at::Tensor attn_bwd(const at::Tensor& dout, const at::Tensor& qkv,
   return dqkv;
 }
 
+std::vector<at::Tensor> sgu_fwd(const at::Tensor& xa, const at::Tensor& g_ln,
+                                const at::Tensor& w, const at::Tensor& bias) {
+  TORCH_CHECK(xa.is_cuda() && xa.is_contiguous() && xa.dim() == 3);
+  TORCH_CHECK(xa.scalar_type() == at::kBFloat16 &&
+              g_ln.scalar_type() == at::kBFloat16 &&
+              w.scalar_type() == at::kBFloat16);
+  const int B = xa.size(0), N = xa.size(1), D = xa.size(2);
+  TORCH_CHECK(N % 256 == 0 && D % 64 == 0, "sgu_fwd: N%256, D%64 required");
+  TORCH_CHECK(w.is_contiguous() && w.size(0) == N && w.size(1) == N);
+  auto bias_f = bias.to(at::kFloat).contiguous().view({-1});
+  auto out = at::empty_like(xa);
+  auto gate_out = at::empty_like(xa);
+  sgu_fwd_launch(xa.data_ptr(), g_ln.data_ptr(), w.data_ptr(),
+                 bias_f.data_ptr<float>(), out.data_ptr(),
+                 gate_out.data_ptr(), B, N, D, cur_stream());
+  return {out, gate_out};
+}
+
+at::Tensor sgu_dgate(const at::Tensor& t_in, const at::Tensor& w) {
+  const int B = t_in.size(0), N = t_in.size(1), D = t_in.size(2);
+  auto dg = at::empty_like(t_in);
+  sgu_dgate_launch(t_in.data_ptr(), w.data_ptr(), dg.data_ptr(), B, N, D,
+                   cur_stream());
+  return dg;
+}
+
+at::Tensor sgu_dw(const at::Tensor& t_in, const at::Tensor& g_ln,
+                  const at::Tensor& tri_m, const at::Tensor& tri_k) {
+  const int B = t_in.size(0), N = t_in.size(1), D = t_in.size(2);
+  TORCH_CHECK(tri_m.scalar_type() == at::kInt && tri_m.is_cuda());
+  auto dw = at::zeros({(long)N, (long)N},
+                      t_in.options().dtype(at::kFloat));
+  sgu_dw_launch(t_in.data_ptr(), g_ln.data_ptr(), dw.data_ptr<float>(),
+                tri_m.data_ptr<int>(), tri_k.data_ptr<int>(),
+                tri_m.size(0), B, N, D, cur_stream());
+  return dw;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
+  m.def("sgu_fwd", &sgu_fwd, "SGU causal spatial matmul forward");
+  m.def("sgu_dgate", &sgu_dgate, "SGU backward: dgate");
+  m.def("sgu_dw", &sgu_dw, "SGU backward: dW");
   m.def("ln_shift_fwd", &ln_shift_fwd, "fused LN+shift forward");
   m.def("ln_shift_bwd", &ln_shift_bwd, "fused LN+shift backward");
   m.def("glu_fwd", &glu_fwd, "GLU-GELU forward");

@@ -48,4 +48,13 @@ void attn_bwd_launch(const void* dout, const void* qkv, const float* rsin,
                      float* dacc, void* dqkv, int B, int N, int H, int wsz,
                      hipStream_t stream);
 
+void sgu_fwd_launch(const void* xa, const void* g_ln, const void* w,
+                    const float* bias, void* out, void* gate_out, int B,
+                    int N, int D, hipStream_t stream);
+void sgu_dgate_launch(const void* t_in, const void* w, void* dg, int B, int N,
+                      int D, hipStream_t stream);
+void sgu_dw_launch(const void* t_in, const void* g_ln, float* dw,
+                   const int* tri_m, const int* tri_k, int ntri, int B, int N,
+                   int D, hipStream_t stream);
+
 }  // extern "C"

@@ -23,6 +23,7 @@ HIP_SOURCES = [
     "progen_amd/ops/hip/rope_qkv.hip",
     "progen_amd/ops/hip/attention_fwd.hip",
     "progen_amd/ops/hip/attention_bwd.hip",
+    "progen_amd/ops/hip/sgu.hip",
 ]
 
 setup(

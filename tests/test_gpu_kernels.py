@@ -248,3 +248,53 @@ def test_training_loss_decreases():
         opt.micro_step()
         losses.append(loss.item())
     assert losses[-1] < losses[0] * 0.7, losses[::10]
+
+
+# ---------------------------------------------------------------------------
+# SGU spatial gating kernels
+# ---------------------------------------------------------------------------
+
+def test_sgu_kernels_fwd_bwd():
+    torch.manual_seed(11)
+    B, N, H = 2, 256, 256  # H = full hidden (split into xa/gate halves)
+    x = (torch.randn(B, N, H, device=dev()) / 4.0).to(torch.bfloat16)
+    x.requires_grad_(True)
+    g = torch.randn(H // 2, device=dev(), dtype=torch.bfloat16)
+    g.requires_grad_(True)
+    W = (torch.randn(N, N, device=dev()) * 1e-3).to(torch.bfloat16)
+    W.requires_grad_(True)
+    b = torch.ones(N, 1, device=dev(), dtype=torch.bfloat16)
+    b.requires_grad_(True)
+
+    out = OF.sgu_gate(x, g, W, b)
+    dy = (torch.randn_like(out) / 4.0).to(torch.bfloat16)
+    out.backward(dy)
+
+    x32 = x.detach().float().cpu().requires_grad_(True)
+    g32 = g.detach().float().cpu().requires_grad_(True)
+    W32 = W.detach().float().cpu().requires_grad_(True)
+    b32 = b.detach().float().cpu().requires_grad_(True)
+    out32 = R.sgu_gate(x32, g32, W32, b32)
+    out32.backward(dy.float().cpu())
+
+    assert rel_err(out, out32) < 2e-2
+    assert rel_err(x.grad, x32.grad) < 3e-2
+    assert rel_err(g.grad, g32.grad) < 3e-2
+    assert rel_err(W.grad, W32.grad) < 3e-2
+    assert rel_err(b.grad, b32.grad) < 3e-2
+
+
+def test_sgu_causality_gpu():
+    torch.manual_seed(12)
+    B, N, H = 1, 256, 128
+    x = (torch.randn(B, N, H, device=dev()) / 4.0).to(torch.bfloat16)
+    g = torch.ones(H // 2, device=dev(), dtype=torch.bfloat16)
+    W = (torch.randn(N, N, device=dev()) * 1e-2).to(torch.bfloat16)
+    b = torch.ones(N, 1, device=dev(), dtype=torch.bfloat16)
+    base = OF.sgu_gate(x, g, W, b)
+    x2 = x.clone()
+    p = 100
+    x2[0, p] += 1.0
+    out2 = OF.sgu_gate(x2, g, W, b)
+    d = (base[0, :p].float() - out2[0, :p].float()).abs().max().item()
+    assert d == 0.0, d
