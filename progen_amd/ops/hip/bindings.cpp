@@ -50,7 +50,7 @@ std::vector<at::Tensor> ln_shift_bwd(const at::Tensor& dy, const at::Tensor& x,
   bool bf = check_dtype(x);
   const int B = x.size(0), N = x.size(1), D = x.size(2);
   const int R = B * N;
-  int nblocks = std::min(R, 256);
+  int nblocks = std::min(R, 1024);  // >=4 blocks/CU for latency hiding
   auto dx = at::empty_like(x);
   auto dw_part = at::empty({nblocks, D}, x.options().dtype(at::kFloat));
   ln_shift_bwd_launch(dy.data_ptr(), x.data_ptr(), g.data_ptr(),

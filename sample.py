@@ -12,14 +12,16 @@ import torch
 from progen_amd import ProGenBase, ProGenConfig
 from progen_amd.checkpoint import get_checkpoint_fns, numpy_to_tensors
 from progen_amd.data import decode_tokens, encode_tokens
-from progen_amd.utils import sample
+from progen_amd.utils import sample, sample_fast
 
 
 @click.command()
 @click.option('--seed', default=42)
 @click.option('--checkpoint_path', default='./ckpts')
 @click.option('--prime', default='')
-def main(seed, checkpoint_path, prime):
+@click.option('--fast', default=False, is_flag=True,
+              help='length-growing decode with EOS early-exit (identical tokens)')
+def main(seed, checkpoint_path, prime, fast):
     _, get_last_checkpoint, _ = get_checkpoint_fns(checkpoint_path)
     last_checkpoint = get_last_checkpoint()
     if last_checkpoint is None:
@@ -57,8 +59,13 @@ def main(seed, checkpoint_path, prime):
         with torch.no_grad():
             return module(seq.to(device))[0].float().cpu()
 
-    sampled = sample(fwd, prime_tensor, seq_len, top_k=25, add_bos=True,
-                     generator=g)
+    if fast:
+        sampled = sample_fast(fwd, prime_tensor, seq_len, top_k=25,
+                              add_bos=True, generator=g,
+                              window_size=cfg.window_size)
+    else:
+        sampled = sample(fwd, prime_tensor, seq_len, top_k=25, add_bos=True,
+                         generator=g)
     sampled_str = decode_tokens(sampled[prime_length:].numpy())
 
     print('\n', prime, '\n', '*' * 40, '\n', sampled_str)

@@ -65,3 +65,38 @@ def test_sample_respects_top_k():
     out = sample(fn, torch.tensor([1]), length=16, top_k=2, generator=g)
     # with top_k=2 only tokens {3, 4} may appear after the prime
     assert set(out[1:].tolist()) <= {3, 4}
+
+
+def test_sample_fast_matches_reference():
+    """sample_fast must emit IDENTICAL tokens to the parity sampler (the
+    model is causal, so the shorter padded forwards see the same
+    prefixes)."""
+    from progen_amd.utils import sample_fast
+
+    m = ProGenBase(ProGenConfig(**TINY))
+    prime = torch.tensor([5, 6, 7])
+    out_ref = sample(_fn(m), prime, length=32, top_k=5,
+                     generator=torch.Generator().manual_seed(3))
+    out_fast = sample_fast(_fn(m), prime, length=32, top_k=5,
+                           generator=torch.Generator().manual_seed(3),
+                           window_size=8)
+    assert out_ref.tolist() == out_fast.tolist()
+
+
+def test_sample_fast_early_exit_bos():
+    from progen_amd.utils import sample_fast
+
+    calls = {"n": 0}
+
+    def fn(seq):
+        calls["n"] += 1
+        logits = torch.full((seq.shape[0], 32), -100.0)
+        logits[:, 0] = 100.0  # always emit pad
+        return logits
+
+    out = sample_fast(fn, torch.tensor([5, 6]), length=30, top_k=1,
+                      add_bos=True, window_size=8)
+    # BOS is the first zero; the first sampled pad is EOS -> one step only
+    assert calls["n"] == 1
+    assert out[1:3].tolist() == [5, 6]
+    assert (out[3:] == 0).all()
