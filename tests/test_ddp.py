@@ -62,7 +62,8 @@ def test_dp2_matches_single_process():
     world = 2
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = 29511
+    import random
+    port = random.randint(29600, 29999)
     procs = [ctx.Process(target=_worker, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
@@ -107,10 +108,12 @@ def _worker_bucket_overlap(rank, world, port, result_q):
 
 @pytest.mark.timeout(120)
 def test_grads_identical_across_ranks():
+    import random
     world = 2
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    procs = [ctx.Process(target=_worker_bucket_overlap, args=(r, world, 29512, q))
+    port = random.randint(29600, 29999)
+    procs = [ctx.Process(target=_worker_bucket_overlap, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
         p.start()
