@@ -6,6 +6,7 @@ recommended by SURVEY.md §4.
 """
 
 import os
+import socket
 
 import numpy as np
 import pytest
@@ -20,6 +21,12 @@ from progen_amd.utils import compute_loss
 
 TINY = dict(num_tokens=32, dim=16, seq_len=32, depth=2, window_size=8,
             global_mlp_depth=1, heads=2, dim_head=8)
+
+
+def _free_port() -> int:
+    with socket.socket() as s:
+        s.bind(("127.0.0.1", 0))
+        return s.getsockname()[1]
 
 
 def _worker(rank, world, port, result_q):
@@ -62,8 +69,7 @@ def test_dp2_matches_single_process():
     world = 2
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    import random
-    port = random.randint(29600, 29999)
+    port = _free_port()
     procs = [ctx.Process(target=_worker, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
@@ -108,11 +114,10 @@ def _worker_bucket_overlap(rank, world, port, result_q):
 
 @pytest.mark.timeout(120)
 def test_grads_identical_across_ranks():
-    import random
     world = 2
     ctx = mp.get_context("spawn")
     q = ctx.SimpleQueue()
-    port = random.randint(29600, 29999)
+    port = _free_port()
     procs = [ctx.Process(target=_worker_bucket_overlap, args=(r, world, port, q))
              for r in range(world)]
     for p in procs:
