@@ -4,33 +4,39 @@
 // reference differentiates progen.py:83-103 through XLA; here the math
 // is explicit):
 //   D_i = rowsum(dO_i * O_i)
-//   P   = exp(S_masked - lse)          (S recomputed from q', scaled k')
+//   P   = exp(S_masked - lse)          (S recomputed from scaled q', k')
 //   dV' = P^T dO
 //   dP  = dO V'^T
 //   dS  = P o (dP - D)
-//   dQ' = dS k_s                       (scale folded into staged K)
+//   dQ' = dS k_s                       (scale folded into staged K^T)
 //   dK' = dS^T q_s                     (scale folded into staged Q^T)
 // followed by attn_bwd_finalize_kernel, which applies the inverse rotary
 // rotation (rotary is linear, so it commutes with accumulation) and
-// casts the fp32 accumulator to bf16 dqkv.
+// casts the fp32 accumulator to bf16 dqkv. Input qkv is PRE-ROTATED
+// (ops/hip/rope_qkv.hip).
 //
-// Geometry mirrors the forward: block = one (batch, head, window), 4
-// waves, each wave owns a 64-row q-chunk. dQ rows are exclusively owned
-// -> plain fp32 stores. dK/dV go through atomicAdd because adjacent
-// windows' key bands overlap (a key is seen by its own window AND the
-// next window's lookback, progen.py:90-91); window 0's lookback keys are
-// the zero pad and their grads are discarded.
+// Geometry: block = one (batch, head, window), 4 waves, each wave owns a
+// 64-row q-chunk. Per 64-key tile the per-row phases (S, P, dS, dQ) are
+// wave-local; the per-KEY phases (dV, dK) are computed as per-wave
+// 16-key OUTPUT SLICES whose MFMA K-dim spans ALL chunks' P/dS/dO^T/Q^T
+// LDS regions — so every dV/dK element is produced by exactly ONE wave
+// and hits global memory with ONE atomicAdd per overlapping window
+// (adjacent windows share keys through the lookback, progen.py:90-91;
+// the naive per-wave accumulation was 4x more atomic traffic and was the
+// top kernel in the step profile). dQ rows are exclusively owned ->
+// plain fp32 stores. Window 0's lookback keys are the zero pad; their
+// gradients are discarded.
 //
-// MFMA operand LDS images (all XOR-swizzled, byte ^= (row&7)<<4):
-//   k_lds  [key][dh] scaled k'   (S B-fragments)
-//   kt_lds [dh][key] scaled k'   (dQ B-fragments)
-//   v_lds  [key][dh] v'          (dP B-fragments)
-//   qt_lds [dh][row] scaled q'   (dK B-fragments, per wave)
-//   dot_lds[dh][row] dO          (dV B-fragments, per wave)
-//   pds_lds [key][row]           P then dS^T, b64-written from the MFMA
-//                                C-layout (4 consecutive rows at a fixed
-//                                key = one 8-B write), per wave
-//   dsrl_lds [row][key]          dS, scattered b16 writes, per wave
+// MFMA operand LDS images (XOR-swizzled, byte ^= (row&7)<<4):
+//   k_lds  [key][dh]   k'           (S B-fragments)
+//   kt_lds [dh][key]   scaled k'    (dQ B-fragments)
+//   v_lds  [key][dh]   v'           (dP B-fragments)
+//   qt_lds [dh][row]   scaled q'    (dK B-fragments, per chunk)
+//   dot_lds[dh][row]   dO           (dV B-fragments, per chunk)
+//   pds_lds [key][row] P then dS^T  (b64-written from the MFMA C-layout:
+//                                    4 consecutive rows at a fixed key =
+//                                    one 8-B write; per chunk)
+//   dsrl_lds [row][key] dS          (scattered b16 writes; per chunk)
 
 #include "common.h"
 
@@ -42,17 +48,6 @@
 
 __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return (byte_in_row ^ ((row & 7) << 4));
-}
-
-__device__ __forceinline__ void rope8(float* x, const float* sinv,
-                                      const float* cosv) {
-#pragma unroll
-  for (int p = 0; p < 4; ++p) {
-    float x0 = x[2 * p], x1 = x[2 * p + 1];
-    float s = sinv[2 * p], c = cosv[2 * p];
-    x[2 * p] = x0 * c - x1 * s;
-    x[2 * p + 1] = x1 * c + x0 * s;
-  }
 }
 
 __device__ __forceinline__ void load_rope(const float* rsin,
@@ -92,25 +87,37 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
   char* k_lds = smem;                                   // 8 KiB
   char* kt_lds = smem + 8192;                           // 8 KiB
   char* v_lds = smem + 16384;                           // 8 KiB
-  char* qt_lds = smem + 24576 + wid * 8192;             // 32 KiB (4 waves)
-  char* dot_lds = smem + 24576 + 32768 + wid * 8192;    // 32 KiB
-  char* pds_lds = smem + 24576 + 65536 + wid * 8192;    // 32 KiB
-  char* dsrl_lds = smem + 24576 + 98304 + wid * 8192;   // 32 KiB
+  char* qt_base = smem + 24576;                         // 32 KiB (4 chunks)
+  char* dot_base = smem + 24576 + 32768;                // 32 KiB
+  char* pds_base = smem + 24576 + 65536;                // 32 KiB
+  char* dsrl_base = smem + 24576 + 98304;               // 32 KiB
   float* d_lds = (float*)(smem + 24576 + 131072 + wid * 256);      // 1 KiB
   float* lse_lds = (float*)(smem + 24576 + 131072 + 1024 + wid * 256);
+
+  char* qt_lds = qt_base + wid * 8192;
+  char* dot_lds = dot_base + wid * 8192;
+  char* pds_lds = pds_base + wid * 8192;
+  char* dsrl_lds = dsrl_base + wid * 8192;
 
   const float scale = rsqrtf((float)DH);
   const int tiles = 2 * wsz / KT;
   const int chunks = wsz / 64;
   const int rounds = (chunks + ATTN_WAVES - 1) / ATTN_WAVES;
 
+  // T14 staging registers (pure copies of pre-rotated k/v)
+  const int su_key[2] = {(int)threadIdx.x >> 3,
+                         (int)(threadIdx.x + ATTN_BLOCK) >> 3};
+  const int su_d0[2] = {((int)threadIdx.x & 7) * 8,
+                        (((int)threadIdx.x + ATTN_BLOCK) & 7) * 8};
+
   for (int round = 0; round < rounds; ++round) {
     const int chunk = round * ATTN_WAVES + wid;
     const bool active = chunk < chunks;
+    const int nactive = min(ATTN_WAVES, chunks - round * ATTN_WAVES);
     const int chunk_off = chunk * 64;
     const int q0 = window * wsz + chunk_off;
 
-    bf16x8 qfrag[4][2];  // rotary q' (unscaled), S recompute A-fragments
+    bf16x8 qfrag[4][2];  // scaled pre-rotated q fragments
     f32x4 dqacc[4][4];
 #pragma unroll
     for (int m = 0; m < 4; ++m)
@@ -126,7 +133,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           const int d0 = ks * 32 + 8 * l4;
           bf16x8 v = *(const bf16x8*)(qkv + qkv_bn + (long long)row * HD3 +
                                       q_off + d0);
-          bf16x8 o;  // pre-rotated input; fold in the softmax scale
+          bf16x8 o;
 #pragma unroll
           for (int j = 0; j < 8; ++j)
             ((short*)&o)[j] = f2bf(bf2f(((short*)&v)[j]) * scale);
@@ -134,8 +141,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
         }
       }
 
-      // per-round wave-local staging: Q^T (scaled), dO^T, D, lse.
-      // One lane per row of the chunk.
+      // per-round chunk staging: Q^T (scaled), dO^T, D, lse; one lane/row
       {
         const int row = lane;
         const long long gq = qkv_bn + (long long)(q0 + row) * HD3 + q_off;
@@ -150,7 +156,6 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             const int d = d0 + j;
-            // qt holds SCALED q' (dK = dS^T q_s)
             *(short*)(qt_lds + d * 128 + swz(d, row * 2)) =
                 f2bf(bf2f(((short*)&qv)[j]) * scale);
             *(short*)(dot_lds + d * 128 + swz(d, row * 2)) = ((short*)&dov)[j];
@@ -160,49 +165,59 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
         d_lds[row] = dsum;
         lse_lds[row] = lse[((long long)batch * H + head) * N + q0 + row];
       }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
     }
+    __syncthreads();  // qt/dot/pds regions ready & previous round done
 
     const int max_tile = active ? ((chunk_off + 63 + wsz) / KT) : -1;
 
-    for (int t = 0; t < tiles; ++t) {
-      __syncthreads();
+    // ---- staging prologue (T14): tile 0 ----
+    bf16x8 kreg[2], vreg[2];
+    auto issue_loads = [&](int t) {
 #pragma unroll
-      for (int pass = 0; pass < 2; ++pass) {
-        const int flat = pass * ATTN_BLOCK + threadIdx.x;
-        const int key = flat >> 3;
-        const int d0 = (flat & 7) * 8;
-        const int kpos = (window - 1) * wsz + t * KT + key;
-        bf16x8 kb, vb;
+      for (int u = 0; u < 2; ++u) {
+        const int kpos = (window - 1) * wsz + t * KT + su_key[u];
         if (kpos >= 0) {
-          kb = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 + k_off +
-                                d0);
-          vb = *(const bf16x8*)(qkv + qkv_bn + (long long)kpos * HD3 + v_off +
-                                d0);
+          const long long base = qkv_bn + (long long)kpos * HD3;
+          kreg[u] = *(const bf16x8*)(qkv + base + k_off + su_d0[u]);
+          vreg[u] = *(const bf16x8*)(qkv + base + v_off + su_d0[u]);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
-            ((short*)&kb)[j] = 0;
-            ((short*)&vb)[j] = 0;
+            ((short*)&kreg[u])[j] = 0;
+            ((short*)&vreg[u])[j] = 0;
           }
         }
-        *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kb;
-        *(bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2)) = vb;
+      }
+    };
+    auto write_lds = [&]() {
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const int key = su_key[u];
+        const int d0 = su_d0[u];
+        *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
+        *(bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2)) = vreg[u];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
           const int d = d0 + j;
-          // kt holds SCALED k' (dQ = dS k_s)
           *(short*)(kt_lds + d * 128 + swz(d, key * 2)) =
-              f2bf(bf2f(((short*)&kb)[j]) * scale);
+              f2bf(bf2f(((short*)&kreg[u])[j]) * scale);
         }
       }
-      __syncthreads();
+    };
 
-      if (active && t <= max_tile) {
-        const int kb = t * KT;
+    issue_loads(0);
+    write_lds();
+    __syncthreads();
 
-        // ---- S = q_s k'^T ----
-        f32x4 s[4][4];
+    for (int t = 0; t < tiles; ++t) {
+      if (t + 1 < tiles) issue_loads(t + 1);
+      const int kb = t * KT;
+      // chunks whose causal range covers this tile: chunk >= c_min
+      const int c_min = max(0, (t * KT - wsz - 63 + 63) / 64 - round * ATTN_WAVES);
+      const bool i_compute = active && t <= max_tile;
+
+      f32x4 s[4][4];  // S -> P for this wave's rows
+      if (i_compute) {
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -220,7 +235,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
                   qfrag[m][ks], kf, s[m][n], 0, 0, 0);
           }
 
-        // ---- P = exp(S - lse) masked; b64-write P^T into pds_lds ----
+        // P = exp(S - lse) masked; b64-write P^T into own pds region
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -232,7 +247,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
               const int kpos_band = kb + n * 16 + l15;
               float v = ((float*)&s[m][n])[r];
               v = (kpos_band > rowiw + wsz) ? 0.f : __expf(v - l);
-              ((float*)&s[m][n])[r] = v;  // s now holds P
+              ((float*)&s[m][n])[r] = v;
             }
           }
 #pragma unroll
@@ -247,48 +262,50 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             *(unsigned long long*)(pds_lds + key * 128 + swz(key, row0 * 2)) =
                 *(unsigned long long*)pk;
           }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      }
+      __syncthreads();  // all P regions ready
 
-        // ---- dV = P^T dO : A = P^T (pds), B = dO^T image (dot_lds) ----
-        {
-          f32x4 dv[4][4];
+      // ---- dV slice: this wave owns keys [wid*16, wid*16+16) of the
+      // tile; K-dim spans contributing chunks' rows ----
+      {
+        f32x4 dv[4];
 #pragma unroll
-          for (int m = 0; m < 4; ++m)
+        for (int n = 0; n < 4; ++n) dv[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        for (int c = c_min; c < nactive; ++c) {
+          char* pds_c = pds_base + c * 8192;
+          char* dot_c = dot_base + c * 8192;
 #pragma unroll
-            for (int n = 0; n < 4; ++n) dv[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+          for (int ks = 0; ks < 2; ++ks) {
+            const int key = wid * 16 + l15;
+            const int r0 = ks * 32 + 8 * l4;
+            bf16x8 pf = *(const bf16x8*)(pds_c + key * 128 + swz(key, r0 * 2));
 #pragma unroll
-          for (int ks = 0; ks < 2; ++ks)
-#pragma unroll
-            for (int m = 0; m < 4; ++m) {  // m: key fragment rows
-              const int key = m * 16 + l15;
-              bf16x8 pf = *(const bf16x8*)(pds_lds + key * 128 +
-                                           swz(key, (ks * 32 + 8 * l4) * 2));
-#pragma unroll
-              for (int n = 0; n < 4; ++n) {  // n: dh fragment cols
-                const int d = n * 16 + l15;
-                bf16x8 dof = *(const bf16x8*)(dot_lds + d * 128 +
-                                              swz(d, (ks * 32 + 8 * l4) * 2));
-                dv[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    pf, dof, dv[m][n], 0, 0, 0);
-              }
+            for (int n = 0; n < 4; ++n) {
+              const int d = n * 16 + l15;
+              bf16x8 dof = *(const bf16x8*)(dot_c + d * 128 + swz(d, r0 * 2));
+              dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
+                                                              0, 0, 0);
             }
-          // C: dV[key][dh]; atomicAdd into dacc v region (skip pad keys)
-#pragma unroll
-          for (int m = 0; m < 4; ++m)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              const int kpos = (window - 1) * wsz + kb + m * 16 + l4 * 4 + r;
-              if (kpos >= 0) {
-#pragma unroll
-                for (int n = 0; n < 4; ++n)
-                  atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + v_off +
-                                n * 16 + l15,
-                            ((float*)&dv[m][n])[r]);
-              }
-            }
+          }
         }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
+          if (kpos >= 0 && c_min < nactive) {
+#pragma unroll
+            for (int n = 0; n < 4; ++n)
+              atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + v_off +
+                            n * 16 + l15,
+                        ((float*)&dv[n])[r]);
+          }
+        }
+      }
 
-        // ---- dP = dO V'^T ; dS = P o (dP - D) ----
+      __syncthreads();  // dV reads of every pds region complete before
+                        // any wave overwrites its own with dS
+
+      // ---- dP = dO V'^T ; dS = P o (dP - D); write dS^T + dS ----
+      if (i_compute) {
         f32x4 dp[4][4];
 #pragma unroll
         for (int m = 0; m < 4; ++m)
@@ -319,12 +336,9 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             for (int n = 0; n < 4; ++n) {
               float p = ((float*)&s[m][n])[r];
               float d = ((float*)&dp[m][n])[r];
-              ((float*)&dp[m][n])[r] = p * (d - dval);  // dp now holds dS
+              ((float*)&dp[m][n])[r] = p * (d - dval);  // now dS
             }
           }
-
-        // ---- write dS^T into pds_lds (b64, overwrites P) and dS into
-        //      dsrl_lds (scattered b16) ----
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -341,47 +355,46 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
               *(short*)(dsrl_lds + (row0 + r) * 128 + swz(row0 + r, key * 2)) =
                   dk4[r];
           }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      }
+      __syncthreads();  // all dS regions ready
 
-        // ---- dK = dS^T q_s : A = dS^T (pds), B = Q^T image (qt_lds) ----
-        {
-          f32x4 dk[4][4];
+      // ---- dK slice (keys [wid*16, wid*16+16)): K spans chunks ----
+      {
+        f32x4 dk[4];
 #pragma unroll
-          for (int m = 0; m < 4; ++m)
+        for (int n = 0; n < 4; ++n) dk[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        for (int c = c_min; c < nactive; ++c) {
+          char* pds_c = pds_base + c * 8192;
+          char* qt_c = qt_base + c * 8192;
 #pragma unroll
-            for (int n = 0; n < 4; ++n) dk[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+          for (int ks = 0; ks < 2; ++ks) {
+            const int key = wid * 16 + l15;
+            const int r0 = ks * 32 + 8 * l4;
+            bf16x8 dsf = *(const bf16x8*)(pds_c + key * 128 + swz(key, r0 * 2));
 #pragma unroll
-          for (int ks = 0; ks < 2; ++ks)
-#pragma unroll
-            for (int m = 0; m < 4; ++m) {
-              const int key = m * 16 + l15;
-              bf16x8 dsf = *(const bf16x8*)(pds_lds + key * 128 +
-                                            swz(key, (ks * 32 + 8 * l4) * 2));
-#pragma unroll
-              for (int n = 0; n < 4; ++n) {
-                const int d = n * 16 + l15;
-                bf16x8 qf = *(const bf16x8*)(qt_lds + d * 128 +
-                                             swz(d, (ks * 32 + 8 * l4) * 2));
-                dk[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                    dsf, qf, dk[m][n], 0, 0, 0);
-              }
+            for (int n = 0; n < 4; ++n) {
+              const int d = n * 16 + l15;
+              bf16x8 qf = *(const bf16x8*)(qt_c + d * 128 + swz(d, r0 * 2));
+              dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, dk[n],
+                                                              0, 0, 0);
             }
-#pragma unroll
-          for (int m = 0; m < 4; ++m)
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              const int kpos = (window - 1) * wsz + kb + m * 16 + l4 * 4 + r;
-              if (kpos >= 0) {
-#pragma unroll
-                for (int n = 0; n < 4; ++n)
-                  atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + k_off +
-                                n * 16 + l15,
-                            ((float*)&dk[m][n])[r]);
-              }
-            }
+          }
         }
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
+          if (kpos >= 0 && c_min < nactive) {
+#pragma unroll
+            for (int n = 0; n < 4; ++n)
+              atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + k_off +
+                            n * 16 + l15,
+                        ((float*)&dk[n])[r]);
+          }
+        }
+      }
 
-        // ---- dQ += dS k_s : A = dS (dsrl), B = K^T image (kt_lds) ----
+      // ---- dQ += dS k_s (own rows; accumulates across tiles) ----
+      if (i_compute) {
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
@@ -399,6 +412,12 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             }
           }
       }
+
+      __syncthreads();  // done reading k/v/kt LDS for tile t
+      if (t + 1 < tiles) {
+        write_lds();
+        __syncthreads();
+      }
     }
 
     // ---- store dQ (rows exclusively owned -> plain fp32 stores) ----
@@ -414,6 +433,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
                 ((float*)&dqacc[m][n])[r];
         }
     }
+    __syncthreads();
   }
 }
 
@@ -425,7 +445,6 @@ __global__ __launch_bounds__(256) void attn_bwd_finalize_kernel(
     const float* __restrict__ dacc, const float* __restrict__ rsin,
     const float* __restrict__ rcos, short* __restrict__ dqkv, int B, int N,
     int H) {
-  // one thread per 8-element dh group of one (b, n, qkv-slot, head)
   const long long HD3 = 3LL * H * DH;
   const long long total = (long long)B * N * 3 * H * (DH / 8);
   for (long long idx = blockIdx.x * 256LL + threadIdx.x; idx < total;

@@ -10,20 +10,27 @@
 //     zero K rows give logit 0 into the softmax denominator and zero V
 //     rows contribute nothing — exact parity by zero-filling the tiles;
 //   - the offset-causal mask tril(ones(wsz, 2wsz), k=wsz) (progen.py:95)
-//     baked into the tile iteration (fully-masked tiles skipped);
+//     baked into the tile iteration (fully-masked tiles skipped; fully
+//     VISIBLE tiles skip the per-element compare);
 //   - fp32 online softmax (max-subtract parity with progen.py:98-99);
 //   - P·V accumulation and the '(w n) (h d)' output merge (progen.py:102).
 //
 // Geometry (CDNA4): block = 4 waves = one window; each wave owns a
 // 64-row Q chunk (round-robins chunks when wsz > 256). Per 64-key tile:
-// K staged in LDS [key][dh] and V transposed [dh][key], both
-// XOR-swizzled (byte ^= (row&7)<<4) so the mfma_f32_16x16x32_bf16
-// B-fragment ds_read_b128s are <=2-way bank conflicted. Staging uses the
-// async-STAGE split (T14): tile t+1's global loads are issued before
-// tile t's compute, the LDS writes land after the barrier — hiding HBM
-// latency at 1 wave/SIMD occupancy. Q fragments (pre-scaled) and the
-// fp32 softmax state live in registers for the whole block. Saves
-// per-row logsumexp (B, h, N) fp32 for the backward's recompute.
+//   - QK^T is computed SWAPPED — S^T = mfma(K, Q) — so the MFMA C-layout
+//     holds 4 consecutive KEYS of one q-row per register quad: the P
+//     tile is written to LDS [row][key] with ds_write_b64 (the
+//     non-swapped form needs 64 scattered b16 writes), and the row
+//     softmax reduce is 16 in-lane values + a 2-step cross-lane shuffle;
+//   - K staged in LDS [key][dh] and V transposed [dh][key], both
+//     XOR-swizzled (byte ^= (row&7)<<4) -> <=2-way bank conflicts on the
+//     ds_read_b128 fragments; staging uses the async-STAGE split (T14):
+//     tile t+1's global loads issue before tile t's compute, the LDS
+//     writes land after the barrier — hiding HBM latency at 1 wave/SIMD;
+//   - per-row softmax state (m, l) lives in the 4 lanes of the row's
+//     shuffle group; the O rescale factor crosses to the PV C-layout
+//     rows through a tiny per-wave LDS broadcast array.
+// Saves per-row logsumexp (B, h, N) fp32 for the backward's recompute.
 //
 // dim_head is fixed at 64 (the ProGen family's head size).
 
@@ -63,13 +70,14 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
   char* k_lds = smem;                          // 8 KiB
   char* v_lds = smem + KT * DH * 2;            // 8 KiB (V^T image)
   char* p_lds = smem + 2 * KT * DH * 2 + wid * 64 * KT * 2;  // 8 KiB/wave
+  float* bc_lds = (float*)(smem + 2 * KT * DH * 2 + ATTN_WAVES * 64 * KT * 2 +
+                           wid * 512);  // [64 alpha | 64 inv_l] per wave
 
   const float scale = rsqrtf((float)DH);
   const int tiles = 2 * wsz / KT;
   const int chunks = wsz / 64;
   const int rounds = (chunks + ATTN_WAVES - 1) / ATTN_WAVES;
 
-  // staging geometry: 512 (key, d0) units per tile, 2 per thread
   const int su_key[2] = {(int)threadIdx.x >> 3,
                          (int)(threadIdx.x + ATTN_BLOCK) >> 3};
   const int su_d0[2] = {((int)threadIdx.x & 7) * 8,
@@ -81,7 +89,9 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
     const int chunk_off = chunk * 64;
     const int q0 = window * wsz + chunk_off;
 
-    // ---- Q fragments (pre-rotated; fold in the softmax scale) ----
+    // ---- Q fragments (pre-rotated; fold in the softmax scale).
+    // The same per-lane data serves as the mfma B operand for the
+    // swapped S^T = K Q^T: lane holds Q[row l15+16n][dh 8*l4+j..]. ----
     bf16x8 qfrag[4][2];
     if (active) {
 #pragma unroll
@@ -101,15 +111,15 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
       }
     }
 
-    float m_run[4][4], l_run[4][4];
+    // per-lane softmax state: 4 q-rows (n*16 + l15), replicated in the
+    // row's 4-lane shuffle group (l4 = 0..3)
+    float m_run[4], l_run[4];
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
-#pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        m_run[m][r] = NEG_INF;
-        l_run[m][r] = 0.f;
-      }
-    f32x4 oacc[4][4];
+    for (int n = 0; n < 4; ++n) {
+      m_run[n] = NEG_INF;
+      l_run[n] = 0.f;
+    }
+    f32x4 oacc[4][4];  // [m rowblock][dh frag], C rows = l4*4+r
 #pragma unroll
     for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -117,9 +127,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 
     const int max_tile = active ? ((chunk_off + 63 + wsz) / KT) : -1;
 
-    // ---- T14 staging: issue loads early, write LDS after barrier ----
     bf16x8 kreg[2], vreg[2];
-
     auto issue_loads = [&](int t) {
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
@@ -137,7 +145,6 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         }
       }
     };
-
     auto write_lds = [&]() {
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
@@ -157,83 +164,98 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
     __syncthreads();
 
     for (int t = 0; t < tiles; ++t) {
-      if (t + 1 < tiles) issue_loads(t + 1);  // in flight during compute
+      if (t + 1 < tiles) issue_loads(t + 1);
 
       if (active && t <= max_tile) {
-        // ---- S = Q K^T ----
-        f32x4 s[4][4];
+        const int kb = t * KT;
+        // every key of the tile visible to every row of the chunk?
+        const bool tile_full = (kb + KT - 1) <= chunk_off + wsz;
+
+        // ---- S^T = K Q^T: st[km][n] rows=keys, cols=q-rows ----
+        f32x4 st[4][4];
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int km = 0; km < 4; ++km)
 #pragma unroll
-          for (int n = 0; n < 4; ++n) s[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+          for (int n = 0; n < 4; ++n) st[km][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
 #pragma unroll
-          for (int n = 0; n < 4; ++n) {
-            const int key = n * 16 + l15;
+          for (int km = 0; km < 4; ++km) {
+            const int key = km * 16 + l15;
             const int d0 = ks * 32 + 8 * l4;
             bf16x8 kfrag = *(const bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2));
 #pragma unroll
-            for (int m = 0; m < 4; ++m)
-              s[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  qfrag[m][ks], kfrag, s[m][n], 0, 0, 0);
+            for (int n = 0; n < 4; ++n)
+              st[km][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  kfrag, qfrag[n][ks], st[km][n], 0, 0, 0);
           }
         }
 
-        // ---- mask + online softmax ----
-        const int kb = t * KT;
+        // ---- mask + per-row max (in-lane over 16 keys, then x-lane) ----
+        float tile_max[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const int rowiw = chunk_off + n * 16 + l15;
+          float mx = NEG_INF;
+#pragma unroll
+          for (int km = 0; km < 4; ++km)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              float v = ((float*)&st[km][n])[r];
+              if (!tile_full) {
+                const int kpos_band = kb + km * 16 + l4 * 4 + r;
+                if (kpos_band > rowiw + wsz) v = NEG_INF;  // progen.py:95
+                ((float*)&st[km][n])[r] = v;
+              }
+              mx = fmaxf(mx, v);
+            }
+          // row data lives in lanes l15, l15+16, l15+32, l15+48
+          mx = fmaxf(mx, __shfl_xor(mx, 16, 64));
+          mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+          tile_max[n] = mx;
+        }
+
+        // ---- online softmax update; write P (bf16) with b64 ----
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const float mnew = fmaxf(m_run[n], tile_max[n]);
+          const float alpha =
+              (m_run[n] == NEG_INF) ? 0.f : __expf(m_run[n] - mnew);
+          float psum = 0.f;
+          const int row = n * 16 + l15;
+#pragma unroll
+          for (int km = 0; km < 4; ++km) {
+            short pk[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              float v = ((float*)&st[km][n])[r];
+              float p = (v <= NEG_INF) ? 0.f : __expf(v - mnew);
+              psum += p;
+              pk[r] = f2bf(p);
+            }
+            *(unsigned long long*)(p_lds + row * 128 +
+                                   swz(row, (km * 16 + l4 * 4) * 2)) =
+                *(unsigned long long*)pk;
+          }
+          psum += __shfl_xor(psum, 16, 64);
+          psum += __shfl_xor(psum, 32, 64);
+          l_run[n] = l_run[n] * alpha + psum;
+          m_run[n] = mnew;
+          if (l4 == 0) bc_lds[row] = alpha;  // broadcast to PV C-layout rows
+        }
+        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+        // ---- O = O * alpha + P V ----
 #pragma unroll
         for (int m = 0; m < 4; ++m) {
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            const int rowiw = chunk_off + m * 16 + l4 * 4 + r;
-            float mx = NEG_INF;
-#pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              const int kpos_band = kb + n * 16 + l15;
-              float v = ((float*)&s[m][n])[r];
-              if (kpos_band > rowiw + wsz) v = NEG_INF;  // progen.py:95
-              ((float*)&s[m][n])[r] = v;
-              mx = fmaxf(mx, v);
-            }
-            mx = group16_max(mx);
-
-            const float mnew = fmaxf(m_run[m][r], mx);
-            const float alpha =
-                (m_run[m][r] == NEG_INF) ? 0.f : __expf(m_run[m][r] - mnew);
-            float psum = 0.f;
-#pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              float v = ((float*)&s[m][n])[r];
-              float p = (v == NEG_INF) ? 0.f : __expf(v - mnew);
-              ((float*)&s[m][n])[r] = p;
-              psum += p;
-            }
-            psum = group16_sum(psum);
-            l_run[m][r] = l_run[m][r] * alpha + psum;
-            m_run[m][r] = mnew;
+            const float alpha = bc_lds[m * 16 + l4 * 4 + r];
 #pragma unroll
             for (int d = 0; d < 4; ++d)
               ((float*)&oacc[m][d])[r] *= alpha;
           }
         }
-
-        // ---- P -> bf16 -> per-wave LDS ----
-#pragma unroll
-        for (int m = 0; m < 4; ++m)
-#pragma unroll
-          for (int n = 0; n < 4; ++n) {
-            const int key = n * 16 + l15;
-#pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              const int row = m * 16 + l4 * 4 + r;
-              *(short*)(p_lds + row * 128 + swz(row, key * 2)) =
-                  f2bf(((float*)&s[m][n])[r]);
-            }
-          }
-        asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
-
-        // ---- O += P V ----
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
 #pragma unroll
@@ -252,36 +274,43 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         }
       }
 
-      __syncthreads();  // all waves done reading LDS tile t
+      __syncthreads();
       if (t + 1 < tiles) {
-        write_lds();    // compiler inserts the vmcnt wait at first use
+        write_lds();
         __syncthreads();
       }
     }
 
     // ---- epilogue: O /= l, store out + lse ----
     if (active) {
+      // broadcast inv_l and write lse from the softmax-state lanes
+#pragma unroll
+      for (int n = 0; n < 4; ++n) {
+        const int row = n * 16 + l15;
+        if (l4 == 0) {
+          bc_lds[64 + row] = 1.0f / l_run[n];
+          lse_out[((long long)batch * H + head) * N + q0 + row] =
+              m_run[n] + logf(l_run[n]);
+        }
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       const long long out_bn = ((long long)batch * N) * (long long)(H * DH);
 #pragma unroll
       for (int m = 0; m < 4; ++m) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = q0 + m * 16 + l4 * 4 + r;
-          const float inv_l = 1.0f / l_run[m][r];
+          const float inv_l = bc_lds[64 + m * 16 + l4 * 4 + r];
 #pragma unroll
           for (int d = 0; d < 4; ++d) {
             const int dcol = d * 16 + l15;
             out[out_bn + (long long)row * (H * DH) + head * DH + dcol] =
                 f2bf(((float*)&oacc[m][d])[r] * inv_l);
           }
-          if (l15 == 0) {
-            lse_out[((long long)batch * H + head) * N + row] =
-                m_run[m][r] + logf(l_run[m][r]);
-          }
         }
       }
     }
-    if (round + 1 < rounds) __syncthreads();
+    if (rounds > 1) __syncthreads();
   }
 }
 
@@ -290,7 +319,8 @@ extern "C" {
 void attn_fwd_launch(const void* qkv_rot, void* out, float* lse, int B, int N,
                      int H, int wsz, hipStream_t stream) {
   dim3 grid(N / wsz, H, B), block(ATTN_BLOCK);
-  size_t lds = (size_t)(2 * KT * DH * 2) + (size_t)ATTN_WAVES * 64 * KT * 2;
+  size_t lds = (size_t)(2 * KT * DH * 2) + (size_t)ATTN_WAVES * 64 * KT * 2 +
+               ATTN_WAVES * 512;
   attn_fwd_kernel<<<grid, block, lds, stream>>>(
       (const short*)qkv_rot, (short*)out, lse, B, N, H, wsz);
 }
