@@ -62,13 +62,17 @@ def main():
     cfg = ProGenConfig(**CONFIGS[args.model])
     dtype = torch.bfloat16 if (args.dtype == "bf16" and on_gpu) else torch.float32
 
-    torch.manual_seed(1234 + rank)
+    torch.manual_seed(1234)  # identical init on every rank (DP replicas)
     module = ProGenBase(cfg).to(device=device, dtype=dtype)
     module.rotary_sin = module.rotary_sin.float()
     module.rotary_cos = module.rotary_cos.float()
 
     optim = ProGenAdamW(module, lr=2e-4, weight_decay=1e-3, max_grad_norm=0.5)
     ddp = DistributedTrainer(optim.space)
+    if world > 1:  # belt-and-braces: bitwise-identical replicas
+        torch.distributed.broadcast(optim.space.flat, src=0)
+        optim.master.copy_(optim.space.flat.float())
+    torch.manual_seed(1234 + rank)  # rank-local data stream
 
     B, N = args.batch, cfg.seq_len
     # synthetic Uniref50-shaped batch: byte tokens 1..256 with zero BOS

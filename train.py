@@ -89,7 +89,7 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
     device = torch.device('cuda', local_rank) if torch.cuda.is_available() \
         else torch.device('cpu')
 
-    torch.manual_seed(seed + rank)
+    torch.manual_seed(seed)  # identical model init on every rank
 
     # checkpoints (reference: train.py:83-92)
     reset_checkpoint, get_last_checkpoint, save_checkpoint = \
@@ -127,6 +127,10 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
                         accum_mode=grad_accum_mode,
                         grad_accum_every=grad_accum_every)
     ddp = DistributedTrainer(optim.space)
+    if world > 1:  # DP replicas must start bitwise-identical
+        import torch.distributed as dist
+        dist.broadcast(optim.space.flat, src=0)
+        optim.master.copy_(optim.space.flat.float())
 
     start_seq_index = 0
     if exists(last_checkpoint):
