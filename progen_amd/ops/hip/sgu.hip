@@ -189,15 +189,27 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_dgate_kernel(
         }
       }
     }
-    // per-wave W^T tile: wt[k][m] = W[m][k], k in wave's rows.
-    // Each lane stages one k-row (64 m values = strided column of W).
-    if (t * 64 >= k0) {  // wave needs this m-tile
-      const int k = k0 + lane;
+    // W^T tiles for all 4 waves, staged cooperatively: the block reads
+    // W[t*64+mm][kblk*256 .. +256] with coalesced 16-B loads and
+    // scatter-transposes into the per-wave [k][m] images (the triu mask
+    // W[m][k]=0 for m<k applied at the write).
+    {
+      // 64 m-rows x 32 k-groups(8) = 2048 units / 256 threads = 8 iters
 #pragma unroll
-      for (int mm = 0; mm < 64; ++mm) {
+      for (int it = 0; it < 8; ++it) {
+        const int u = it * SGU_BLOCK + (int)threadIdx.x;
+        const int mm = u >> 5;           // 0..63
+        const int k8 = (u & 31) * 8;     // 0..255 step 8
         const int m = t * 64 + mm;
-        short wv = (m >= k) ? w[(long long)m * N + k] : (short)0;
-        *(short*)(wt_lds + lane * 128 + swz(lane, mm * 2)) = wv;
+        const int kglob = kblk * 256 + k8;
+        bf16x8 wv = *(const bf16x8*)(w + (long long)m * N + kglob);
+        char* wt_region = smem + 8192 + (k8 / 64) * 8192;
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          const int k = (k8 + j) & 63;  // k index within the wave region
+          short v = (m >= kglob + j) ? ((short*)&wv)[j] : (short)0;
+          *(short*)(wt_region + k * 128 + swz(k, mm * 2)) = v;
+        }
       }
     }
     __syncthreads();
