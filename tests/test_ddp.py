@@ -58,8 +58,8 @@ def _worker(rank, world, port, result_q):
 
     if rank == 0:
         result_q.put({
-            "params": opt.master.detach().clone(),
-            "grad": opt.space.flat_grad.detach().clone(),
+            "params": opt.master.detach().numpy().copy(),
+            "grad": opt.space.flat_grad.detach().numpy().copy(),
         })
     dist.destroy_process_group()
 
@@ -91,7 +91,7 @@ def test_dp2_matches_single_process():
         compute_loss(model, data).backward()
         opt.micro_step()
 
-    np.testing.assert_allclose(got["params"].numpy(), opt.master.numpy(),
+    np.testing.assert_allclose(got["params"], opt.master.numpy(),
                                atol=1e-5)
 
 
@@ -108,7 +108,7 @@ def _worker_bucket_overlap(rank, world, port, result_q):
     data = torch.randint(0, 32, (1 + rank, 33))
     compute_loss(model, data).backward()
     ddp.finish_backward()
-    result_q.put((rank, opt.space.flat_grad.clone()))
+    result_q.put((rank, opt.space.flat_grad.numpy().copy()))
     dist.destroy_process_group()
 
 
@@ -129,4 +129,4 @@ def test_grads_identical_across_ranks():
     for p in procs:
         p.join(120)
         assert p.exitcode == 0
-    np.testing.assert_allclose(res[0].numpy(), res[1].numpy(), atol=1e-7)
+    np.testing.assert_allclose(res[0], res[1], atol=1e-7)
