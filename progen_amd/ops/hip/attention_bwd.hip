@@ -64,7 +64,8 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
     const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16, PRE-ROTATED
     const short* __restrict__ out,   // (B, N, H*DH) bf16 (fwd output)
     const float* __restrict__ lse,   // (B, H, N)
-    float* __restrict__ dacc,        // (B, N, 3*H*DH) fp32, zero-init
+    float* __restrict__ dacc,        // (B, N, 3*H*DH) fp32 (own + dQ)
+    float* __restrict__ dlook,       // (B, N, 2*H*DH) fp32 (lookback k/v)
     int B, int N, int H, int wsz) {
   const int window = blockIdx.x;
   const int head = blockIdx.y;
@@ -82,6 +83,9 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
   const int q_off = head * DH;
   const int k_off = H * DH + head * DH;
   const int v_off = 2 * H * DH + head * DH;
+  const long long look_bn = (long long)batch * N * (2LL * H * DH);
+  const int lk_off = head * DH;            // k slot in dlook
+  const int lv_off = H * DH + head * DH;   // v slot in dlook
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                                   // 8 KiB
@@ -288,15 +292,20 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             }
           }
         }
+        const bool lookback = kb < wsz;  // tile-uniform half of the band
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
-          if (kpos >= 0 && c_min < nactive) {
+          if (kpos >= 0) {
+            float* dst = lookback
+                ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lv_off
+                : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
 #pragma unroll
-            for (int n = 0; n < 4; ++n)
-              atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + v_off +
-                            n * 16 + l15,
-                        ((float*)&dv[n])[r]);
+            for (int n = 0; n < 4; ++n) {
+              float v = ((float*)&dv[n])[r];
+              if (round > 0) v += dst[n * 16 + l15];  // later chunk rounds
+              dst[n * 16 + l15] = v;
+            }
           }
         }
       }
@@ -380,15 +389,20 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             }
           }
         }
+        const bool lookback = kb < wsz;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
-          if (kpos >= 0 && c_min < nactive) {
+          if (kpos >= 0) {
+            float* dst = lookback
+                ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lk_off
+                : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
 #pragma unroll
-            for (int n = 0; n < 4; ++n)
-              atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + k_off +
-                            n * 16 + l15,
-                        ((float*)&dk[n])[r]);
+            for (int n = 0; n < 4; ++n) {
+              float v = ((float*)&dk[n])[r];
+              if (round > 0) v += dst[n * 16 + l15];
+              dst[n * 16 + l15] = v;
+            }
           }
         }
       }
@@ -442,10 +456,11 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
 // ---------------------------------------------------------------------------
 
 __global__ __launch_bounds__(256) void attn_bwd_finalize_kernel(
-    const float* __restrict__ dacc, const float* __restrict__ rsin,
-    const float* __restrict__ rcos, short* __restrict__ dqkv, int B, int N,
-    int H) {
+    const float* __restrict__ dacc, const float* __restrict__ dlook,
+    const float* __restrict__ rsin, const float* __restrict__ rcos,
+    short* __restrict__ dqkv, int B, int N, int H, int wsz) {
   const long long HD3 = 3LL * H * DH;
+  const long long HD2 = 2LL * H * DH;
   const long long total = (long long)B * N * 3 * H * (DH / 8);
   for (long long idx = blockIdx.x * 256LL + threadIdx.x; idx < total;
        idx += (long long)gridDim.x * 256) {
@@ -460,6 +475,13 @@ __global__ __launch_bounds__(256) void attn_bwd_finalize_kernel(
     float x[8], sv[8], cv[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) x[j] = dacc[off + j];
+    // k/v slots: add the lookback contribution (exists unless this is
+    // the last window — its keys are nobody's lookback)
+    if (hslot >= H && (n / wsz) < (N / wsz) - 1) {
+      const long long loff = bn * HD2 + (long long)(hslot - H) * DH + d0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) x[j] += dlook[loff + j];
+    }
     load_rope(rsin, rcos, n, d0, sv, cv);
     // inverse rotation: dx[2i] = dy[2i] c + dy[2i+1] s;
     //                   dx[2i+1] = dy[2i+1] c - dy[2i] s
@@ -481,18 +503,18 @@ extern "C" {
 
 void attn_bwd_launch(const void* dout, const void* qkv, const float* rsin,
                      const float* rcos, const void* out, const float* lse,
-                     float* dacc, void* dqkv, int B, int N, int H, int wsz,
-                     hipStream_t stream) {
+                     float* dacc, float* dlook, void* dqkv, int B, int N,
+                     int H, int wsz, hipStream_t stream) {
   dim3 grid(N / wsz, H, B), block(ATTN_BLOCK);
   size_t lds = 24576 + 131072 + 2048;  // 154 KiB
   attn_bwd_kernel<<<grid, block, lds, stream>>>(
-      (const short*)dout, (const short*)qkv, (const short*)out, lse, dacc, B,
-      N, H, wsz);
+      (const short*)dout, (const short*)qkv, (const short*)out, lse, dacc,
+      dlook, B, N, H, wsz);
   long long total = (long long)B * N * 3 * H * (DH / 8);
   int fin_grid = (int)((total + 255) / 256);
   if (fin_grid > 2048) fin_grid = 2048;
-  attn_bwd_finalize_kernel<<<fin_grid, 256, 0, stream>>>(dacc, rsin, rcos,
-                                                         (short*)dqkv, B, N, H);
+  attn_bwd_finalize_kernel<<<fin_grid, 256, 0, stream>>>(
+      dacc, dlook, rsin, rcos, (short*)dqkv, B, N, H, wsz);
 }
 
 }  // extern "C"

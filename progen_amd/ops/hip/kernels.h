@@ -45,8 +45,8 @@ void attn_fwd_launch(const void* qkv_rot, void* out, float* lse, int B, int N,
                      int H, int wsz, hipStream_t stream);
 void attn_bwd_launch(const void* dout, const void* qkv, const float* rsin,
                      const float* rcos, const void* out, const float* lse,
-                     float* dacc, void* dqkv, int B, int N, int H, int wsz,
-                     hipStream_t stream);
+                     float* dacc, float* dlook, void* dqkv, int B, int N,
+                     int H, int wsz, hipStream_t stream);
 
 void sgu_fwd_launch(const void* xa, const void* g_ln, const void* w,
                     const float* bias, void* out, void* gate_out, int B,
