@@ -74,12 +74,14 @@ def _wandb(wandb_off):
 @click.option('--wandb_project_name', default='progen-training')
 @click.option('--new', default=False, is_flag=True)
 @click.option('--max_steps', default=0, help='stop after N effective batches (0 = full epoch)')
+@click.option('--graph/--no-graph', default=True,
+              help='hipGraph-capture the training step (GPU, grad_accum_every=1)')
 @click.option('--yes', default=False, is_flag=True, help='skip the --new confirmation prompt')
 def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
          data_parallel, max_grad_norm, validate_every, sample_every,
          checkpoint_every, checkpoint_path, checkpoint_keep_n, config_path,
          model_name, prime_length, seq_len, mixed_precision, grad_accum_mode,
-         data_path, wandb_off, wandb_project_name, new, max_steps, yes):
+         data_path, wandb_off, wandb_project_name, new, max_steps, yes, graph):
     from progen_amd.tuning import enable_tuned_gemms
     enable_tuned_gemms()
     local_rank = init_distributed()
@@ -168,6 +170,21 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
     valid_dataset = get_valid_dataset(seq_len=seq_len, batch_size=global_batch,
                                       loop=True)
 
+    # hipGraph-captured step (progen_amd/runtime.py): replays the whole
+    # fwd+bwd(+optimizer) sequence as one graph when shapes are static
+    graphed = None
+    if graph and device.type == 'cuda' and grad_accum_every == 1 \
+            and grad_accum_mode == 'sum':
+        from progen_amd.runtime import GraphedTrainStep
+        try:
+            graphed = GraphedTrainStep(module, optim, ddp, batch_size,
+                                       seq_len, device)
+            if is_main:
+                print('hipGraph training step captured')
+        except Exception as e:  # noqa: BLE001
+            if is_main:
+                print(f'hipGraph capture failed ({e}); running eager')
+
     if is_main:
         print(f'params: {num_params}')
         print(f'sequence length: {seq_len}')
@@ -197,6 +214,9 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
                 data = my_shard(next(train_dataset))
             except StopIteration:
                 break
+            if graphed is not None:
+                loss = graphed.run(data)
+                continue
             last_micro = micro == grad_accum_every - 1
             if last_micro:
                 loss = compute_loss(module, data)
