@@ -40,6 +40,8 @@
 #define KT 64                 // keys per tile
 #define ATTN_WAVES 4
 #define ATTN_BLOCK (ATTN_WAVES * WAVE)
+#define MF 2                  // 16-row m-fragments per wave (32-row chunks)
+#define QB (MF * 16)          // q rows per wave
 #define NEG_INF (-1e30f)
 
 __device__ __forceinline__ int swz(int row, int byte_in_row) {
@@ -51,7 +53,10 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
     short* __restrict__ out,         // (B, N, H*DH) bf16
     float* __restrict__ lse_out,     // (B, H, N)
     int B, int N, int H, int wsz) {
-  const int window = blockIdx.x;
+  // blocks per window: each covers ATTN_WAVES*QB q rows
+  const int sub_per_win = (wsz + ATTN_WAVES * QB - 1) / (ATTN_WAVES * QB);
+  const int window = blockIdx.x / sub_per_win;
+  const int sub = blockIdx.x % sub_per_win;
   const int head = blockIdx.y;
   const int batch = blockIdx.z;
 
@@ -69,14 +74,16 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
   char* k_lds = smem;                          // 8 KiB
   char* v_lds = smem + KT * DH * 2;            // 8 KiB (V^T image)
-  char* p_lds = smem + 2 * KT * DH * 2 + wid * 64 * KT * 2;  // 8 KiB/wave
-  float* bc_lds = (float*)(smem + 2 * KT * DH * 2 + ATTN_WAVES * 64 * KT * 2 +
-                           wid * 512);  // [64 alpha | 64 inv_l] per wave
+  char* p_lds = smem + 2 * KT * DH * 2 + wid * QB * KT * 2;  // 4 KiB/wave
+  float* bc_lds = (float*)(smem + 2 * KT * DH * 2 + ATTN_WAVES * QB * KT * 2 +
+                           wid * 2 * QB * 4);  // [QB alpha | QB inv_l]/wave
 
   const float scale = rsqrtf((float)DH);
   const int tiles = 2 * wsz / KT;
-  const int chunks = wsz / 64;
-  const int rounds = (chunks + ATTN_WAVES - 1) / ATTN_WAVES;
+  // chunks local to this block's 128-row slice of the window
+  const int chunks_all = wsz / QB;
+  const int chunks = min(ATTN_WAVES, chunks_all - sub * ATTN_WAVES);
+  const int rounds = 1;
 
   const int su_key[2] = {(int)threadIdx.x >> 3,
                          (int)(threadIdx.x + ATTN_BLOCK) >> 3};
@@ -84,18 +91,17 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
                         (((int)threadIdx.x + ATTN_BLOCK) & 7) * 8};
 
   for (int round = 0; round < rounds; ++round) {
-    const int chunk = round * ATTN_WAVES + wid;
-    const bool active = chunk < chunks;
-    const int chunk_off = chunk * 64;
+    const bool active = wid < chunks;
+    const int chunk_off = (sub * ATTN_WAVES + wid) * QB;  // within window
     const int q0 = window * wsz + chunk_off;
 
     // ---- Q fragments (pre-rotated; fold in the softmax scale).
     // The same per-lane data serves as the mfma B operand for the
     // swapped S^T = K Q^T: lane holds Q[row l15+16n][dh 8*l4+j..]. ----
-    bf16x8 qfrag[4][2];
+    bf16x8 qfrag[MF][2];
     if (active) {
 #pragma unroll
-      for (int m = 0; m < 4; ++m) {
+      for (int m = 0; m < MF; ++m) {
         const int row = q0 + m * 16 + l15;
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
@@ -113,19 +119,19 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 
     // per-lane softmax state: 4 q-rows (n*16 + l15), replicated in the
     // row's 4-lane shuffle group (l4 = 0..3)
-    float m_run[4], l_run[4];
+    float m_run[MF], l_run[MF];
 #pragma unroll
-    for (int n = 0; n < 4; ++n) {
+    for (int n = 0; n < MF; ++n) {
       m_run[n] = NEG_INF;
       l_run[n] = 0.f;
     }
-    f32x4 oacc[4][4];  // [m rowblock][dh frag], C rows = l4*4+r
+    f32x4 oacc[MF][4];  // [m rowblock][dh frag], C rows = l4*4+r
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
+    for (int m = 0; m < MF; ++m)
 #pragma unroll
       for (int d = 0; d < 4; ++d) oacc[m][d] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-    const int max_tile = active ? ((chunk_off + 63 + wsz) / KT) : -1;
+    const int max_tile = active ? ((chunk_off + QB - 1 + wsz) / KT) : -1;
 
     bf16x8 kreg[2], vreg[2];
     auto issue_loads = [&](int t) {
@@ -172,11 +178,11 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         const bool tile_full = (kb + KT - 1) <= chunk_off + wsz;
 
         // ---- S^T = K Q^T: st[km][n] rows=keys, cols=q-rows ----
-        f32x4 st[4][4];
+        f32x4 st[4][MF];
 #pragma unroll
         for (int km = 0; km < 4; ++km)
 #pragma unroll
-          for (int n = 0; n < 4; ++n) st[km][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+          for (int n = 0; n < MF; ++n) st[km][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
 #pragma unroll
@@ -185,16 +191,16 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
             const int d0 = ks * 32 + 8 * l4;
             bf16x8 kfrag = *(const bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2));
 #pragma unroll
-            for (int n = 0; n < 4; ++n)
+            for (int n = 0; n < MF; ++n)
               st[km][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                   kfrag, qfrag[n][ks], st[km][n], 0, 0, 0);
           }
         }
 
         // ---- mask + per-row max (in-lane over 16 keys, then x-lane) ----
-        float tile_max[4];
+        float tile_max[MF];
 #pragma unroll
-        for (int n = 0; n < 4; ++n) {
+        for (int n = 0; n < MF; ++n) {
           const int rowiw = chunk_off + n * 16 + l15;
           float mx = NEG_INF;
 #pragma unroll
@@ -217,7 +223,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 
         // ---- online softmax update; write P (bf16) with b64 ----
 #pragma unroll
-        for (int n = 0; n < 4; ++n) {
+        for (int n = 0; n < MF; ++n) {
           const float mnew = fmaxf(m_run[n], tile_max[n]);
           const float alpha =
               (m_run[n] == NEG_INF) ? 0.f : __expf(m_run[n] - mnew);
@@ -247,7 +253,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 
         // ---- O = O * alpha + P V ----
 #pragma unroll
-        for (int m = 0; m < 4; ++m) {
+        for (int m = 0; m < MF; ++m) {
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             const float alpha = bc_lds[m * 16 + l4 * 4 + r];
@@ -259,7 +265,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
 #pragma unroll
-          for (int m = 0; m < 4; ++m) {
+          for (int m = 0; m < MF; ++m) {
             const int row = m * 16 + l15;
             const int kk0 = ks * 32 + 8 * l4;
             bf16x8 pfrag = *(const bf16x8*)(p_lds + row * 128 + swz(row, kk0 * 2));
@@ -285,10 +291,10 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
     if (active) {
       // broadcast inv_l and write lse from the softmax-state lanes
 #pragma unroll
-      for (int n = 0; n < 4; ++n) {
+      for (int n = 0; n < MF; ++n) {
         const int row = n * 16 + l15;
         if (l4 == 0) {
-          bc_lds[64 + row] = 1.0f / l_run[n];
+          bc_lds[QB + row] = 1.0f / l_run[n];
           lse_out[((long long)batch * H + head) * N + q0 + row] =
               m_run[n] + logf(l_run[n]);
         }
@@ -296,11 +302,11 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       const long long out_bn = ((long long)batch * N) * (long long)(H * DH);
 #pragma unroll
-      for (int m = 0; m < 4; ++m) {
+      for (int m = 0; m < MF; ++m) {
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = q0 + m * 16 + l4 * 4 + r;
-          const float inv_l = bc_lds[64 + m * 16 + l4 * 4 + r];
+          const float inv_l = bc_lds[QB + m * 16 + l4 * 4 + r];
 #pragma unroll
           for (int d = 0; d < 4; ++d) {
             const int dcol = d * 16 + l15;
@@ -318,9 +324,10 @@ extern "C" {
 
 void attn_fwd_launch(const void* qkv_rot, void* out, float* lse, int B, int N,
                      int H, int wsz, hipStream_t stream) {
-  dim3 grid(N / wsz, H, B), block(ATTN_BLOCK);
-  size_t lds = (size_t)(2 * KT * DH * 2) + (size_t)ATTN_WAVES * 64 * KT * 2 +
-               ATTN_WAVES * 512;
+  const int sub_per_win = (wsz + ATTN_WAVES * QB - 1) / (ATTN_WAVES * QB);
+  dim3 grid((N / wsz) * sub_per_win, H, B), block(ATTN_BLOCK);
+  size_t lds = (size_t)(2 * KT * DH * 2) + (size_t)ATTN_WAVES * QB * KT * 2 +
+               ATTN_WAVES * 2 * QB * 4;
   attn_fwd_kernel<<<grid, block, lds, stream>>>(
       (const short*)qkv_rot, (short*)out, lse, B, N, H, wsz);
 }
