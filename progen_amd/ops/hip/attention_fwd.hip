@@ -157,8 +157,12 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         const int key = su_key[u];
         const int d0 = su_d0[u];
         *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
+        // stagger j by lane: concurrent lanes hit different (d&7) -> the
+        // XOR swizzle spreads the b16 writes over 8 bank groups instead
+        // of all landing on one (PMC: 16-way conflicts without this)
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
+        for (int jj = 0; jj < 8; ++jj) {
+          const int j = (jj + (int)threadIdx.x) & 7;
           const int d = d0 + j;
           *(short*)(v_lds + d * 128 + swz(d, key * 2)) = ((short*)&vreg[u])[j];
         }
