@@ -45,10 +45,16 @@ def _haiku_linear_init_(weight: torch.Tensor, bias: Optional[torch.Tensor],
 
 
 class Linear(nn.Linear):
-    """nn.Linear with haiku-style default init."""
+    """nn.Linear with haiku-style default init; on GPU the weight grad is
+    computed on a side stream concurrent with the backward chain
+    (ops/overlap.py)."""
 
     def reset_parameters(self) -> None:
         _haiku_linear_init_(self.weight, self.bias)
+
+    def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ..ops.overlap import overlap_linear
+        return overlap_linear(x, self.weight, self.bias)
 
 
 class LocalAttention(nn.Module):

@@ -146,6 +146,9 @@ class ProGenAdamW:
         parameters were actually updated this call."""
         self._micro += 1
         if self.accum_mode == "apply_every":
+            if self.space.flat_grad.is_cuda:
+                from .ops.overlap import WgradQueue
+                WgradQueue.sync()
             self._apply_every_micro()
             return self._micro % self.grad_accum_every == 0
         if self._micro % self.grad_accum_every == 0:
@@ -156,6 +159,9 @@ class ProGenAdamW:
 
     def step(self, grad_scale: float = 1.0) -> None:
         """One optimizer update from the (accumulated) flat grad buffer."""
+        if self.space.flat_grad.is_cuda:
+            from .ops.overlap import WgradQueue
+            WgradQueue.sync()  # join side-stream wgrads before the update
         self.step_count += 1
         g = self.space.flat_grad
         if dispatch.use_hip(g):

@@ -132,6 +132,18 @@ class DistributedTrainer:
     def finish_backward(self) -> None:
         """Wait for outstanding bucket reductions and average. Call after
         the final (synchronizing) backward of the step."""
+        if self.space.flat_grad.is_cuda:
+            from ..ops.overlap import WgradQueue
+            WgradQueue.sync()  # side-stream wgrads must land before reduce
+        if self.world > 1:
+            # buckets holding side-stream-wgrad params never fire their
+            # hooks (no AccumulateGrad) — reduce them now
+            for b, pending in enumerate(self._pending):
+                if pending > 0:
+                    start, end, _ = self.buckets[b]
+                    self._works.append(
+                        dist.all_reduce(self.space.flat_grad[start:end],
+                                        op=dist.ReduceOp.SUM, async_op=True))
         for w in self._works:
             w.wait()
         self._works.clear()

@@ -84,6 +84,8 @@ class GraphedTrainStep:
         self.optim.space.flat_grad.zero_()
         loss = self.loss_fn(self.module, self.static_data)
         loss.backward()
+        from .ops.overlap import WgradQueue
+        WgradQueue.sync()  # join side-stream wgrads inside the graph
         return loss.detach()
 
     def _inner(self) -> torch.Tensor:
