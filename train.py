@@ -33,7 +33,7 @@ from progen_amd.data import decode_tokens, iterator_from_tfrecords_folder
 from progen_amd.optim import ProGenAdamW
 from progen_amd.parallel import DistributedTrainer, init_distributed, is_distributed
 from progen_amd.utils import (compute_loss, confirm, exists, load_dotenv,
-                              sample, set_hardware_rng_)
+                              sample_fast, set_hardware_rng_)
 
 load_dotenv()        # reference: train.py:1-2
 set_hardware_rng_()  # reference: train.py:32 (no-op on ROCm, see utils)
@@ -274,7 +274,9 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
                 with torch.no_grad():
                     return module(seq.to(device))[0].float().cpu()
 
-            sampled = sample(fwd, prime.cpu(), seq_len, top_k=25)
+            # identical tokens to the reference sampler, O(prefix) forwards
+            sampled = sample_fast(fwd, prime.cpu(), seq_len, top_k=25,
+                                  window_size=cfg.window_size)
             sampled_str = decode_tokens(sampled[prime_length:].numpy())
             print(prime_str, '\n', '*' * 40, '\n', sampled_str)
             if wandb is not None:
