@@ -6,15 +6,17 @@ fully static, so the training step is captured once into a hipGraph and
 replayed per step — the graph is the compiled program, eager PyTorch is
 only the tracer ("HIP streams and graphs instead of a tracing compiler").
 
-Two capture modes:
-  - world_size == 1: zero-grad + forward + backward + grad-clip + fused
-    AdamW all inside ONE graph (the Adam step counter lives on device,
-    ops/hip/adamw.hip, so replays keep bias correction advancing);
-  - world_size > 1 (DP): the graph captures zero-grad + forward +
-    backward; the RCCL all-reduce of the flat grad buffer and the fused
-    optimizer run eagerly after each replay. This trades the
-    backward/communication overlap (~15 ms at 1.2B/8 GPUs) for removing
-    the ~200 ms eager dispatch gap, and avoids capturing RCCL kernels.
+The captured step is PURE-REPLAY ONLY and single-GPU only: the whole
+iteration (zero-grad + forward + backward + grad-clip + fused AdamW,
+with the Adam step counter on device) is one graph, and NOTHING else may
+launch kernels between replays. Empirically on this ROCm 7.0 stack,
+interleaving ANY eager kernel with replays — even a plain index_select —
+poisons subsequent replays (NaN gradients or HSA aperture faults; see
+profiles/r01_graph_interleave_bug.md for the isolation). Training loops
+that validate/sample/checkpoint between steps must therefore run eager
+(train.py defaults to --no-graph); bench.py replays exclusively and is
+safe. DP (world>1) raises — its all-reduce would be an eager kernel
+between replays.
 
 Warmup runs on a side stream; optimizer/param state perturbed by warmup
 and capture is snapshotted and restored.
@@ -47,6 +49,11 @@ class GraphedTrainStep:
         self.ddp = ddp
         self.loss_fn = loss_fn
         self.world = ddp.world if ddp is not None else 1
+        if self.world > 1:
+            raise RuntimeError(
+                "GraphedTrainStep is single-GPU pure-replay only: the DP "
+                "all-reduce between replays is an eager kernel, which "
+                "corrupts replay state on this ROCm stack")
         self.static_data = torch.zeros(batch, seq_len + 1, dtype=torch.long,
                                        device=device)
 
@@ -89,28 +96,14 @@ class GraphedTrainStep:
         return loss.detach()
 
     def _inner(self) -> torch.Tensor:
-        if self.world > 1:
-            # communication stays OUTSIDE the graph
-            assert self.ddp is not None
-            with self.ddp.no_sync():
-                loss = self._fwd_bwd()
-            if not torch.cuda.is_current_stream_capturing():
-                self._comm_and_step()
-            return loss
         loss = self._fwd_bwd()
         self.optim.step()
         return loss
 
-    def _comm_and_step(self) -> None:
-        dist.all_reduce(self.optim.space.flat_grad, op=dist.ReduceOp.SUM)
-        self.optim.space.flat_grad.div_(self.world)
-        self.optim.step()
-
     def run(self, data: torch.Tensor) -> torch.Tensor:
         """Replay one training step; returns the (device) loss tensor —
-        do not .item() it inside a timed region."""
+        do not .item() it inside a timed region, and do not launch other
+        GPU kernels between replays (see module docstring)."""
         self.static_data.copy_(data, non_blocking=True)
         self.graph.replay()
-        if self.world > 1:
-            self._comm_and_step()
         return self.static_loss

@@ -74,8 +74,10 @@ def _wandb(wandb_off):
 @click.option('--wandb_project_name', default='progen-training')
 @click.option('--new', default=False, is_flag=True)
 @click.option('--max_steps', default=0, help='stop after N effective batches (0 = full epoch)')
-@click.option('--graph/--no-graph', default=True,
-              help='hipGraph-capture the training step (GPU, grad_accum_every=1)')
+@click.option('--graph/--no-graph', default=False,
+              help='hipGraph-capture the training step (single-GPU, pure replay '
+                   'only: disables in-loop validation/sampling — eager kernels '
+                   'between replays corrupt replay state on this ROCm stack)')
 @click.option('--yes', default=False, is_flag=True, help='skip the --new confirmation prompt')
 def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
          data_parallel, max_grad_norm, validate_every, sample_every,
@@ -174,7 +176,7 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
     # fwd+bwd(+optimizer) sequence as one graph when shapes are static
     graphed = None
     if graph and device.type == 'cuda' and grad_accum_every == 1 \
-            and grad_accum_mode == 'sum':
+            and grad_accum_mode == 'sum' and world == 1:
         from progen_amd.runtime import GraphedTrainStep
         try:
             graphed = GraphedTrainStep(module, optim, ddp, batch_size,
@@ -242,7 +244,7 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
             if wandb is not None:
                 wandb.log({'loss': loss_val, 'tokens_per_sec': toks_per_sec})
 
-        if i % checkpoint_every == 0 and is_main:
+        if graphed is None and i % checkpoint_every == 0 and is_main:
             package = {
                 'next_seq_index': seq_index + effective_batch_size,
                 'params': tensors_to_numpy(
@@ -255,7 +257,7 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
             print(f"checkpoint to start at sequence index of "
                   f"{package['next_seq_index']}")
 
-        if i % validate_every == 0:
+        if graphed is None and i % validate_every == 0:
             valid_data = my_shard(next(valid_dataset))
             with torch.no_grad():
                 vloss = compute_loss(module, valid_data)
@@ -265,7 +267,7 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
                 if wandb is not None:
                     wandb.log({'valid_loss': vloss_val})
 
-        if i % sample_every == 0 and is_main:
+        if graphed is None and i % sample_every == 0 and is_main:
             valid_data = my_shard(next(valid_dataset))[0]
             prime = valid_data[:prime_length]
             prime_str = decode_tokens(prime.cpu().numpy())
