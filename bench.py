@@ -82,7 +82,10 @@ def main():
     data[:, -8:] = 0  # pad tail so the EOS-mask path is exercised
 
     graphed = None
-    if on_gpu and os.environ.get("PROGEN_NO_GRAPH") != "1":
+    # hipGraph step: single-GPU pure-replay only (the DP all-reduce between
+    # replays is an eager kernel, which corrupts replay state on this ROCm
+    # stack — profiles/r01_graph_interleave_bug.md)
+    if on_gpu and world == 1 and os.environ.get("PROGEN_NO_GRAPH") != "1":
         from progen_amd.runtime import GraphedTrainStep
         try:
             graphed = GraphedTrainStep(module, optim, ddp, B, N, device)
