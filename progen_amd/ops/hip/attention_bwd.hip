@@ -15,27 +15,28 @@
 // casts the fp32 accumulator to bf16 dqkv. Input qkv is PRE-ROTATED
 // (ops/hip/rope_qkv.hip).
 //
-// Geometry (occupancy-first): block = 4 waves x 32-row q-chunks = 128
-// rows of one window (sub-blocks when wsz > 128); ~240 regs and 73 KiB
-// LDS -> 2 waves/SIMD, 2 blocks/CU. Per 64-key tile:
-//   - per-row phases (S, P, dS, dQ) are wave-local over the wave's 32
-//     rows; dQ rows are exclusively owned -> plain fp32 stores;
-//   - per-key phases (dV, dK) are per-wave 16-key OUTPUT slices whose
-//     MFMA K-dim spans all 4 chunks' LDS regions, atomicAdd'ed into the
-//     fp32 accumulator (keys are shared between the q-row sub-blocks AND
-//     between adjacent windows through the lookback, progen.py:90-91);
-//   - window 0's lookback keys are the zero pad; grads discarded.
+// Geometry: block = one (batch, head, window), 4 waves, each wave owns a
+// 64-row q-chunk. Per 64-key tile the per-row phases (S, P, dS, dQ) are
+// wave-local; the per-KEY phases (dV, dK) are computed as per-wave
+// 16-key OUTPUT SLICES whose MFMA K-dim spans ALL chunks' P/dS/dO^T/Q^T
+// LDS regions — so every dV/dK element is produced by exactly ONE wave
+// and hits global memory with ONE atomicAdd per overlapping window
+// (adjacent windows share keys through the lookback, progen.py:90-91;
+// the naive per-wave accumulation was 4x more atomic traffic and was the
+// top kernel in the step profile). dQ rows are exclusively owned ->
+// plain fp32 stores. Window 0's lookback keys are the zero pad; their
+// gradients are discarded.
 //
-// LDS images (XOR-swizzled; 128-B-stride images use byte^=(row&7)<<4,
-// 64-B-stride images use byte^=((row>>2)&3)<<4):
-//   k_lds  [key64][dh64]   k'          (S B-fragments)          8 KiB
-//   kt_lds [dh64][key64]   scaled k'   (dQ B-fragments)         8 KiB
-//   v_lds  [key64][dh64]   v'          (dP B-fragments)         8 KiB
-//   qt_c   [dh64][row32]   scaled q'   (dK B-frags, per chunk)  4x4 KiB
-//   dot_c  [dh64][row32]   dO          (dV B-frags, per chunk)  4x4 KiB
-//   pdsr_c [key64][row32]  P then dS^T, then [row32][key64] dS
-//          (phase-shared: P_kr -> dV reads -> DS_kr -> dK reads ->
-//           DS_rl -> dQ reads; b64 writes for the _kr forms)     4x4 KiB
+// MFMA operand LDS images (XOR-swizzled, byte ^= (row&7)<<4):
+//   k_lds  [key][dh]   k'           (S B-fragments)
+//   kt_lds [dh][key]   scaled k'    (dQ B-fragments)
+//   v_lds  [key][dh]   v'           (dP B-fragments)
+//   qt_lds [dh][row]   scaled q'    (dK B-fragments, per chunk)
+//   dot_lds[dh][row]   dO           (dV B-fragments, per chunk)
+//   pds_lds [key][row] P then dS^T  (b64-written from the MFMA C-layout:
+//                                    4 consecutive rows at a fixed key =
+//                                    one 8-B write; per chunk)
+//   dsrl_lds [row][key] dS          (scattered b16 writes; per chunk)
 
 #include "common.h"
 
@@ -43,16 +44,10 @@
 #define KT 64
 #define ATTN_WAVES 4
 #define ATTN_BLOCK (ATTN_WAVES * WAVE)
-#define MF 2
-#define QB (MF * 16)  // 32 q rows per wave
 #define NEG_INF (-1e30f)
 
 __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return (byte_in_row ^ ((row & 7) << 4));
-}
-// swizzle for 64-byte-stride images ([x][row32] with 2-B elements)
-__device__ __forceinline__ int swz64(int row, int byte_in_row) {
-  return (byte_in_row ^ (((row >> 2) & 3) << 4));
 }
 
 __device__ __forceinline__ void load_rope(const float* rsin,
@@ -64,16 +59,15 @@ __device__ __forceinline__ void load_rope(const float* rsin,
   *(f32x4*)(cv + 4) = *(const f32x4*)(rcos + pos * DH + d0 + 4);
 }
 
-__global__ __launch_bounds__(ATTN_BLOCK, 2) void attn_bwd_kernel(
+__global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
     const short* __restrict__ dout,  // (B, N, H*DH) bf16
     const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16, PRE-ROTATED
     const short* __restrict__ out,   // (B, N, H*DH) bf16 (fwd output)
     const float* __restrict__ lse,   // (B, H, N)
-    float* __restrict__ dacc,        // (B, N, 3*H*DH) fp32, zero-init
+    float* __restrict__ dacc,        // (B, N, 3*H*DH) fp32 (own + dQ)
+    float* __restrict__ dlook,       // (B, N, 2*H*DH) fp32 (lookback k/v)
     int B, int N, int H, int wsz) {
-  const int sub_per_win = (wsz + ATTN_WAVES * QB - 1) / (ATTN_WAVES * QB);
-  const int window = blockIdx.x / sub_per_win;
-  const int sub = blockIdx.x % sub_per_win;
+  const int window = blockIdx.x;
   const int head = blockIdx.y;
   const int batch = blockIdx.z;
 
@@ -89,364 +83,371 @@ __global__ __launch_bounds__(ATTN_BLOCK, 2) void attn_bwd_kernel(
   const int q_off = head * DH;
   const int k_off = H * DH + head * DH;
   const int v_off = 2 * H * DH + head * DH;
+  const long long look_bn = (long long)batch * N * (2LL * H * DH);
+  const int lk_off = head * DH;            // k slot in dlook
+  const int lv_off = H * DH + head * DH;   // v slot in dlook
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_lds = smem;                       // 8 KiB
-  char* kt_lds = smem + 8192;               // 8 KiB
-  char* v_lds = smem + 16384;               // 8 KiB
-  char* qt_base = smem + 24576;             // 16 KiB (4 x 4 KiB)
-  char* dot_base = smem + 24576 + 16384;    // 16 KiB
-  char* pdsr_base = smem + 24576 + 32768;   // 16 KiB
-  float* d_lds = (float*)(smem + 24576 + 49152 + wid * 128);       // 512 B
-  float* lse_lds = (float*)(smem + 24576 + 49152 + 512 + wid * 128);
+  char* k_lds = smem;                                   // 8 KiB
+  char* kt_lds = smem + 8192;                           // 8 KiB
+  char* v_lds = smem + 16384;                           // 8 KiB
+  char* qt_base = smem + 24576;                         // 32 KiB (4 chunks)
+  char* dot_base = smem + 24576 + 32768;                // 32 KiB
+  char* pds_base = smem + 24576 + 65536;                // 32 KiB
+  char* dsrl_base = smem + 24576 + 98304;               // 32 KiB
+  float* d_lds = (float*)(smem + 24576 + 131072 + wid * 256);      // 1 KiB
+  float* lse_lds = (float*)(smem + 24576 + 131072 + 1024 + wid * 256);
 
-  char* qt_lds = qt_base + wid * 4096;
-  char* dot_lds = dot_base + wid * 4096;
-  char* pdsr_lds = pdsr_base + wid * 4096;
+  char* qt_lds = qt_base + wid * 8192;
+  char* dot_lds = dot_base + wid * 8192;
+  char* pds_lds = pds_base + wid * 8192;
+  char* dsrl_lds = dsrl_base + wid * 8192;
 
   const float scale = rsqrtf((float)DH);
   const int tiles = 2 * wsz / KT;
-  const int chunks = min(ATTN_WAVES, wsz / QB - sub * ATTN_WAVES);
+  const int chunks = wsz / 64;
+  const int rounds = (chunks + ATTN_WAVES - 1) / ATTN_WAVES;
 
+  // T14 staging registers (pure copies of pre-rotated k/v)
   const int su_key[2] = {(int)threadIdx.x >> 3,
                          (int)(threadIdx.x + ATTN_BLOCK) >> 3};
   const int su_d0[2] = {((int)threadIdx.x & 7) * 8,
                         (((int)threadIdx.x + ATTN_BLOCK) & 7) * 8};
 
-  const bool active = wid < chunks;
-  const int chunk_off = (sub * ATTN_WAVES + wid) * QB;  // within window
-  const int q0 = window * wsz + chunk_off;
+  for (int round = 0; round < rounds; ++round) {
+    const int chunk = round * ATTN_WAVES + wid;
+    const bool active = chunk < chunks;
+    const int nactive = min(ATTN_WAVES, chunks - round * ATTN_WAVES);
+    const int chunk_off = chunk * 64;
+    const int q0 = window * wsz + chunk_off;
 
-  bf16x8 qfrag[MF][2];  // scaled pre-rotated q fragments
-  f32x4 dqacc[MF][4];
+    bf16x8 qfrag[4][2];  // scaled pre-rotated q fragments
+    f32x4 dqacc[4][4];
 #pragma unroll
-  for (int m = 0; m < MF; ++m)
+    for (int m = 0; m < 4; ++m)
 #pragma unroll
-    for (int d = 0; d < 4; ++d) dqacc[m][d] = (f32x4){0.f, 0.f, 0.f, 0.f};
+      for (int d = 0; d < 4; ++d) dqacc[m][d] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
-  if (active) {
+    if (active) {
 #pragma unroll
-    for (int m = 0; m < MF; ++m) {
-      const int row = q0 + m * 16 + l15;
+      for (int m = 0; m < 4; ++m) {
+        const int row = q0 + m * 16 + l15;
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks) {
-        const int d0 = ks * 32 + 8 * l4;
-        bf16x8 v = *(const bf16x8*)(qkv + qkv_bn + (long long)row * HD3 +
-                                    q_off + d0);
-        bf16x8 o;
+        for (int ks = 0; ks < 2; ++ks) {
+          const int d0 = ks * 32 + 8 * l4;
+          bf16x8 v = *(const bf16x8*)(qkv + qkv_bn + (long long)row * HD3 +
+                                      q_off + d0);
+          bf16x8 o;
 #pragma unroll
-        for (int j = 0; j < 8; ++j)
-          ((short*)&o)[j] = f2bf(bf2f(((short*)&v)[j]) * scale);
-        qfrag[m][ks] = o;
-      }
-    }
-
-    // chunk staging: Q^T (scaled) and dO^T into the 64-B-stride images,
-    // plus per-row D and lse. Two lanes per row (32 rows, 64 lanes).
-    {
-      const int row = lane >> 1;              // 0..31
-      const int dhalf = (lane & 1) * 32;      // dh 0..31 / 32..63
-      const long long gq = qkv_bn + (long long)(q0 + row) * HD3 + q_off;
-      const long long go = o_bn + (long long)(q0 + row) * HD + head * DH;
-      float dsum = 0.f;
-#pragma unroll
-      for (int g = 0; g < 4; ++g) {
-        const int d0 = dhalf + g * 8;
-        bf16x8 qv = *(const bf16x8*)(qkv + gq + d0);
-        bf16x8 ov = *(const bf16x8*)(out + go + d0);
-        bf16x8 dov = *(const bf16x8*)(dout + go + d0);
-#pragma unroll
-        for (int jj = 0; jj < 8; ++jj) {  // staggered bank-spread writes
-          const int j = (jj + (int)threadIdx.x) & 7;
-          const int d = d0 + j;
-          *(short*)(qt_lds + d * 64 + swz64(d, row * 2)) =
-              f2bf(bf2f(((short*)&qv)[j]) * scale);
-          *(short*)(dot_lds + d * 64 + swz64(d, row * 2)) = ((short*)&dov)[j];
-          dsum += bf2f(((short*)&ov)[j]) * bf2f(((short*)&dov)[j]);
+          for (int j = 0; j < 8; ++j)
+            ((short*)&o)[j] = f2bf(bf2f(((short*)&v)[j]) * scale);
+          qfrag[m][ks] = o;
         }
       }
-      dsum += __shfl_xor(dsum, 1, 64);  // combine the two half-rows
-      if ((lane & 1) == 0) d_lds[row] = dsum;
-      if ((lane & 1) == 1)
-        lse_lds[row] = lse[((long long)batch * H + head) * N + q0 + row];
-    }
-  }
 
-  // ---- staging prologue (T14): tile 0 ----
-  bf16x8 kreg[2], vreg[2];
-  auto issue_loads = [&](int t) {
+      // per-round chunk staging: Q^T (scaled), dO^T, D, lse; one lane/row
+      {
+        const int row = lane;
+        const long long gq = qkv_bn + (long long)(q0 + row) * HD3 + q_off;
+        const long long go = o_bn + (long long)(q0 + row) * HD + head * DH;
+        float dsum = 0.f;
 #pragma unroll
-    for (int u = 0; u < 2; ++u) {
-      const int kpos = (window - 1) * wsz + t * KT + su_key[u];
-      if (kpos >= 0) {
-        const long long base = qkv_bn + (long long)kpos * HD3;
-        kreg[u] = *(const bf16x8*)(qkv + base + k_off + su_d0[u]);
-        vreg[u] = *(const bf16x8*)(qkv + base + v_off + su_d0[u]);
-      } else {
+        for (int g = 0; g < 8; ++g) {
+          const int d0 = g * 8;
+          bf16x8 qv = *(const bf16x8*)(qkv + gq + d0);
+          bf16x8 ov = *(const bf16x8*)(out + go + d0);
+          bf16x8 dov = *(const bf16x8*)(dout + go + d0);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int d = d0 + j;
+            *(short*)(qt_lds + d * 128 + swz(d, row * 2)) =
+                f2bf(bf2f(((short*)&qv)[j]) * scale);
+            *(short*)(dot_lds + d * 128 + swz(d, row * 2)) = ((short*)&dov)[j];
+            dsum += bf2f(((short*)&ov)[j]) * bf2f(((short*)&dov)[j]);
+          }
+        }
+        d_lds[row] = dsum;
+        lse_lds[row] = lse[((long long)batch * H + head) * N + q0 + row];
+      }
+    }
+    __syncthreads();  // qt/dot/pds regions ready & previous round done
+
+    const int max_tile = active ? ((chunk_off + 63 + wsz) / KT) : -1;
+
+    // ---- staging prologue (T14): tile 0 ----
+    bf16x8 kreg[2], vreg[2];
+    auto issue_loads = [&](int t) {
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const int kpos = (window - 1) * wsz + t * KT + su_key[u];
+        if (kpos >= 0) {
+          const long long base = qkv_bn + (long long)kpos * HD3;
+          kreg[u] = *(const bf16x8*)(qkv + base + k_off + su_d0[u]);
+          vreg[u] = *(const bf16x8*)(qkv + base + v_off + su_d0[u]);
+        } else {
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            ((short*)&kreg[u])[j] = 0;
+            ((short*)&vreg[u])[j] = 0;
+          }
+        }
+      }
+    };
+    auto write_lds = [&]() {
+#pragma unroll
+      for (int u = 0; u < 2; ++u) {
+        const int key = su_key[u];
+        const int d0 = su_d0[u];
+        *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
+        *(bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2)) = vreg[u];
 #pragma unroll
         for (int j = 0; j < 8; ++j) {
-          ((short*)&kreg[u])[j] = 0;
-          ((short*)&vreg[u])[j] = 0;
+          const int d = d0 + j;
+          *(short*)(kt_lds + d * 128 + swz(d, key * 2)) =
+              f2bf(bf2f(((short*)&kreg[u])[j]) * scale);
         }
       }
-    }
-  };
-  auto write_lds = [&]() {
-#pragma unroll
-    for (int u = 0; u < 2; ++u) {
-      const int key = su_key[u];
-      const int d0 = su_d0[u];
-      *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
-      *(bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2)) = vreg[u];
-#pragma unroll
-      for (int jj = 0; jj < 8; ++jj) {  // staggered bank-spread writes
-        const int j = (jj + (int)threadIdx.x) & 7;
-        const int d = d0 + j;
-        *(short*)(kt_lds + d * 128 + swz(d, key * 2)) =
-            f2bf(bf2f(((short*)&kreg[u])[j]) * scale);
-      }
-    }
-  };
+    };
 
-  issue_loads(0);
-  write_lds();
-  __syncthreads();
+    issue_loads(0);
+    write_lds();
+    __syncthreads();
 
-  const int max_tile = active ? ((chunk_off + QB - 1 + wsz) / KT) : -1;
+    for (int t = 0; t < tiles; ++t) {
+      if (t + 1 < tiles) issue_loads(t + 1);
+      const int kb = t * KT;
+      // chunks whose causal range covers this tile: chunk >= c_min
+      const int c_min = max(0, (t * KT - wsz - 63 + 63) / 64 - round * ATTN_WAVES);
+      const bool i_compute = active && t <= max_tile;
 
-  for (int t = 0; t < tiles; ++t) {
-    if (t + 1 < tiles) issue_loads(t + 1);
-    const int kb = t * KT;
-    // chunks (wave indices) whose causal range covers this tile
-    const int c_min_g = (t * KT - wsz) > 0 ? (t * KT - wsz) / QB : 0;
-    const int c_min = max(0, c_min_g - sub * ATTN_WAVES);
-    const int c_end = min(chunks, ATTN_WAVES);
-    const bool i_compute = active && t <= max_tile;
-
-    f32x4 s[MF][4];  // S -> P for this wave's rows
-    if (i_compute) {
+      f32x4 s[4][4];  // S -> P for this wave's rows
+      if (i_compute) {
 #pragma unroll
-      for (int m = 0; m < MF; ++m)
+        for (int m = 0; m < 4; ++m)
 #pragma unroll
-        for (int n = 0; n < 4; ++n) s[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+          for (int n = 0; n < 4; ++n) s[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
-#pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          const int key = n * 16 + l15;
-          bf16x8 kf = *(const bf16x8*)(k_lds + key * 128 +
-                                       swz(key, (ks * 32 + 8 * l4) * 2));
-#pragma unroll
-          for (int m = 0; m < MF; ++m)
-            s[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                qfrag[m][ks], kf, s[m][n], 0, 0, 0);
-        }
-
-      // P = exp(S - lse) masked; b64-write P^T into own pdsr [key][row32]
-#pragma unroll
-      for (int m = 0; m < MF; ++m)
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int rowiw = chunk_off + m * 16 + l4 * 4 + r;
-          const float l = lse_lds[m * 16 + l4 * 4 + r];
-#pragma unroll
-          for (int n = 0; n < 4; ++n) {
-            const int kpos_band = kb + n * 16 + l15;
-            float v = ((float*)&s[m][n])[r];
-            v = (kpos_band > rowiw + wsz) ? 0.f : __expf(v - l);
-            ((float*)&s[m][n])[r] = v;
-          }
-        }
-#pragma unroll
-      for (int m = 0; m < MF; ++m)
-#pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          const int key = n * 16 + l15;
-          const int row0 = m * 16 + l4 * 4;
-          short pk[4];
-#pragma unroll
-          for (int r = 0; r < 4; ++r) pk[r] = f2bf(((float*)&s[m][n])[r]);
-          *(unsigned long long*)(pdsr_lds + key * 64 + swz64(key, row0 * 2)) =
-              *(unsigned long long*)pk;
-        }
-    }
-    __syncthreads();  // all P regions ready
-
-    // ---- dV slice: keys [wid*16, wid*16+16); K spans chunks' rows ----
-    {
-      f32x4 dv[4];
-#pragma unroll
-      for (int n = 0; n < 4; ++n) dv[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
-      for (int c = c_min; c < c_end; ++c) {
-        char* pds_c = pdsr_base + c * 4096;
-        char* dot_c = dot_base + c * 4096;
-        const int key = wid * 16 + l15;
-        const int r0 = 8 * l4;  // rows 0..31 of the chunk (one K=32 step)
-        bf16x8 pf = *(const bf16x8*)(pds_c + key * 64 + swz64(key, r0 * 2));
-#pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          const int d = n * 16 + l15;
-          bf16x8 dof = *(const bf16x8*)(dot_c + d * 64 + swz64(d, r0 * 2));
-          dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
-                                                          0, 0, 0);
-        }
-      }
-      if (c_min < c_end) {
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
-          if (kpos >= 0) {
-#pragma unroll
-            for (int n = 0; n < 4; ++n)
-              atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + v_off +
-                            n * 16 + l15,
-                        ((float*)&dv[n])[r]);
-          }
-        }
-      }
-    }
-
-    // ---- dP = dO V'^T ; dS = P o (dP - D) ----
-    f32x4 dp[MF][4];
-    if (i_compute) {
-#pragma unroll
-      for (int m = 0; m < MF; ++m)
-#pragma unroll
-        for (int n = 0; n < 4; ++n) dp[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
-#pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
-#pragma unroll
-        for (int m = 0; m < MF; ++m) {
-          const int row = q0 + m * 16 + l15;
-          const int d0 = ks * 32 + 8 * l4;
-          bf16x8 dof = *(const bf16x8*)(dout + o_bn + (long long)row * HD +
-                                        head * DH + d0);
+        for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
           for (int n = 0; n < 4; ++n) {
             const int key = n * 16 + l15;
-            bf16x8 vf = *(const bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2));
-            dp[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                dof, vf, dp[m][n], 0, 0, 0);
+            bf16x8 kf = *(const bf16x8*)(k_lds + key * 128 +
+                                         swz(key, (ks * 32 + 8 * l4) * 2));
+#pragma unroll
+            for (int m = 0; m < 4; ++m)
+              s[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  qfrag[m][ks], kf, s[m][n], 0, 0, 0);
           }
-        }
+
+        // P = exp(S - lse) masked; b64-write P^T into own pds region
 #pragma unroll
-      for (int m = 0; m < MF; ++m)
+        for (int m = 0; m < 4; ++m)
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const float dval = d_lds[m * 16 + l4 * 4 + r];
+          for (int r = 0; r < 4; ++r) {
+            const int rowiw = chunk_off + m * 16 + l4 * 4 + r;
+            const float l = lse_lds[m * 16 + l4 * 4 + r];
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int kpos_band = kb + n * 16 + l15;
+              float v = ((float*)&s[m][n])[r];
+              v = (kpos_band > rowiw + wsz) ? 0.f : __expf(v - l);
+              ((float*)&s[m][n])[r] = v;
+            }
+          }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n) {
-            float p = ((float*)&s[m][n])[r];
-            float d = ((float*)&dp[m][n])[r];
-            ((float*)&dp[m][n])[r] = p * (d - dval);  // now dS
+            const int key = n * 16 + l15;
+            const int row0 = m * 16 + l4 * 4;
+            short pk[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) pk[r] = f2bf(((float*)&s[m][n])[r]);
+            *(unsigned long long*)(pds_lds + key * 128 + swz(key, row0 * 2)) =
+                *(unsigned long long*)pk;
+          }
+      }
+      __syncthreads();  // all P regions ready
+
+      // ---- dV slice: this wave owns keys [wid*16, wid*16+16) of the
+      // tile; K-dim spans contributing chunks' rows ----
+      {
+        f32x4 dv[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) dv[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        for (int c = c_min; c < nactive; ++c) {
+          char* pds_c = pds_base + c * 8192;
+          char* dot_c = dot_base + c * 8192;
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks) {
+            const int key = wid * 16 + l15;
+            const int r0 = ks * 32 + 8 * l4;
+            bf16x8 pf = *(const bf16x8*)(pds_c + key * 128 + swz(key, r0 * 2));
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int d = n * 16 + l15;
+              bf16x8 dof = *(const bf16x8*)(dot_c + d * 128 + swz(d, r0 * 2));
+              dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
+                                                              0, 0, 0);
+            }
           }
         }
-    }
-    __syncthreads();  // dV reads of every pdsr region complete
-
-    // write dS^T into pdsr (overwrites P)
-    if (i_compute) {
-#pragma unroll
-      for (int m = 0; m < MF; ++m)
-#pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          const int key = n * 16 + l15;
-          const int row0 = m * 16 + l4 * 4;
-          short dk4[4];
-#pragma unroll
-          for (int r = 0; r < 4; ++r) dk4[r] = f2bf(((float*)&dp[m][n])[r]);
-          *(unsigned long long*)(pdsr_lds + key * 64 + swz64(key, row0 * 2)) =
-              *(unsigned long long*)dk4;
-        }
-    }
-    __syncthreads();  // all dS^T regions ready
-
-    // ---- dK slice (keys [wid*16, wid*16+16)): K spans chunks ----
-    {
-      f32x4 dk[4];
-#pragma unroll
-      for (int n = 0; n < 4; ++n) dk[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
-      for (int c = c_min; c < c_end; ++c) {
-        char* pds_c = pdsr_base + c * 4096;
-        char* qt_c = qt_base + c * 4096;
-        const int key = wid * 16 + l15;
-        const int r0 = 8 * l4;
-        bf16x8 dsf = *(const bf16x8*)(pds_c + key * 64 + swz64(key, r0 * 2));
-#pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          const int d = n * 16 + l15;
-          bf16x8 qf = *(const bf16x8*)(qt_c + d * 64 + swz64(d, r0 * 2));
-          dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, dk[n],
-                                                          0, 0, 0);
-        }
-      }
-      if (c_min < c_end) {
+        const bool lookback = kb < wsz;  // tile-uniform half of the band
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
           if (kpos >= 0) {
+            float* dst = lookback
+                ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lv_off
+                : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
 #pragma unroll
-            for (int n = 0; n < 4; ++n)
-              atomicAdd(dacc + qkv_bn + (long long)kpos * HD3 + k_off +
-                            n * 16 + l15,
-                        ((float*)&dk[n])[r]);
+            for (int n = 0; n < 4; ++n) {
+              float v = ((float*)&dv[n])[r];
+              if (round > 0) v += dst[n * 16 + l15];  // later chunk rounds
+              dst[n * 16 + l15] = v;
+            }
           }
         }
       }
-    }
-    __syncthreads();  // dK reads of every pdsr region complete
 
-    // ---- overwrite own pdsr with dS [row32][key64]; dQ (wave-local) ----
-    if (i_compute) {
+      __syncthreads();  // dV reads of every pds region complete before
+                        // any wave overwrites its own with dS
+
+      // ---- dP = dO V'^T ; dS = P o (dP - D); write dS^T + dS ----
+      if (i_compute) {
+        f32x4 dp[4][4];
 #pragma unroll
-      for (int m = 0; m < MF; ++m)
+        for (int m = 0; m < 4; ++m)
 #pragma unroll
-        for (int n = 0; n < 4; ++n) {
-          const int key = n * 16 + l15;
-          const int row0 = m * 16 + l4 * 4;
+          for (int n = 0; n < 4; ++n) dp[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
 #pragma unroll
-          for (int r = 0; r < 4; ++r)
-            *(short*)(pdsr_lds + (row0 + r) * 128 + swz(row0 + r, key * 2)) =
-                f2bf(((float*)&dp[m][n])[r]);
-        }
-      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+        for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
-      for (int ks = 0; ks < 2; ++ks)
+          for (int m = 0; m < 4; ++m) {
+            const int row = q0 + m * 16 + l15;
+            const int d0 = ks * 32 + 8 * l4;
+            bf16x8 dof = *(const bf16x8*)(dout + o_bn + (long long)row * HD +
+                                          head * DH + d0);
 #pragma unroll
-        for (int m = 0; m < MF; ++m) {
-          const int row = m * 16 + l15;
-          bf16x8 dsf = *(const bf16x8*)(pdsr_lds + row * 128 +
-                                        swz(row, (ks * 32 + 8 * l4) * 2));
+            for (int n = 0; n < 4; ++n) {
+              const int key = n * 16 + l15;
+              bf16x8 vf = *(const bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2));
+              dp[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  dof, vf, dp[m][n], 0, 0, 0);
+            }
+          }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const float dval = d_lds[m * 16 + l4 * 4 + r];
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              float p = ((float*)&s[m][n])[r];
+              float d = ((float*)&dp[m][n])[r];
+              ((float*)&dp[m][n])[r] = p * (d - dval);  // now dS
+            }
+          }
+#pragma unroll
+        for (int m = 0; m < 4; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n) {
-            const int d = n * 16 + l15;
-            bf16x8 kf = *(const bf16x8*)(kt_lds + d * 128 +
-                                         swz(d, (ks * 32 + 8 * l4) * 2));
-            dqacc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                dsf, kf, dqacc[m][n], 0, 0, 0);
+            const int key = n * 16 + l15;
+            const int row0 = m * 16 + l4 * 4;
+            short dk4[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) dk4[r] = f2bf(((float*)&dp[m][n])[r]);
+            *(unsigned long long*)(pds_lds + key * 128 + swz(key, row0 * 2)) =
+                *(unsigned long long*)dk4;
+#pragma unroll
+            for (int r = 0; r < 4; ++r)
+              *(short*)(dsrl_lds + (row0 + r) * 128 + swz(row0 + r, key * 2)) =
+                  dk4[r];
+          }
+      }
+      __syncthreads();  // all dS regions ready
+
+      // ---- dK slice (keys [wid*16, wid*16+16)): K spans chunks ----
+      {
+        f32x4 dk[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) dk[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        for (int c = c_min; c < nactive; ++c) {
+          char* pds_c = pds_base + c * 8192;
+          char* qt_c = qt_base + c * 8192;
+#pragma unroll
+          for (int ks = 0; ks < 2; ++ks) {
+            const int key = wid * 16 + l15;
+            const int r0 = ks * 32 + 8 * l4;
+            bf16x8 dsf = *(const bf16x8*)(pds_c + key * 128 + swz(key, r0 * 2));
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int d = n * 16 + l15;
+              bf16x8 qf = *(const bf16x8*)(qt_c + d * 128 + swz(d, r0 * 2));
+              dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, dk[n],
+                                                              0, 0, 0);
+            }
           }
         }
-    }
-
-    __syncthreads();  // done with k/v/kt and all pdsr for tile t
-    if (t + 1 < tiles) {
-      write_lds();
-      __syncthreads();
-    }
-  }
-
-  // ---- store dQ (rows exclusively owned -> plain fp32 stores) ----
-  if (active) {
+        const bool lookback = kb < wsz;
 #pragma unroll
-    for (int m = 0; m < MF; ++m)
+        for (int r = 0; r < 4; ++r) {
+          const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
+          if (kpos >= 0) {
+            float* dst = lookback
+                ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lk_off
+                : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
 #pragma unroll
-      for (int r = 0; r < 4; ++r) {
-        const int row = q0 + m * 16 + l4 * 4 + r;
-#pragma unroll
-        for (int n = 0; n < 4; ++n)
-          dacc[qkv_bn + (long long)row * HD3 + q_off + n * 16 + l15] =
-              ((float*)&dqacc[m][n])[r];
+            for (int n = 0; n < 4; ++n) {
+              float v = ((float*)&dk[n])[r];
+              if (round > 0) v += dst[n * 16 + l15];
+              dst[n * 16 + l15] = v;
+            }
+          }
+        }
       }
+
+      // ---- dQ += dS k_s (own rows; accumulates across tiles) ----
+      if (i_compute) {
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int m = 0; m < 4; ++m) {
+            const int row = m * 16 + l15;
+            bf16x8 dsf = *(const bf16x8*)(dsrl_lds + row * 128 +
+                                          swz(row, (ks * 32 + 8 * l4) * 2));
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int d = n * 16 + l15;
+              bf16x8 kf = *(const bf16x8*)(kt_lds + d * 128 +
+                                           swz(d, (ks * 32 + 8 * l4) * 2));
+              dqacc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  dsf, kf, dqacc[m][n], 0, 0, 0);
+            }
+          }
+      }
+
+      __syncthreads();  // done reading k/v/kt LDS for tile t
+      if (t + 1 < tiles) {
+        write_lds();
+        __syncthreads();
+      }
+    }
+
+    // ---- store dQ (rows exclusively owned -> plain fp32 stores) ----
+    if (active) {
+#pragma unroll
+      for (int m = 0; m < 4; ++m)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = q0 + m * 16 + l4 * 4 + r;
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+            dacc[qkv_bn + (long long)row * HD3 + q_off + n * 16 + l15] =
+                ((float*)&dqacc[m][n])[r];
+        }
+    }
+    __syncthreads();
   }
 }
 
@@ -455,10 +456,11 @@ __global__ __launch_bounds__(ATTN_BLOCK, 2) void attn_bwd_kernel(
 // ---------------------------------------------------------------------------
 
 __global__ __launch_bounds__(256) void attn_bwd_finalize_kernel(
-    const float* __restrict__ dacc, const float* __restrict__ rsin,
-    const float* __restrict__ rcos, short* __restrict__ dqkv, int B, int N,
-    int H) {
+    const float* __restrict__ dacc, const float* __restrict__ dlook,
+    const float* __restrict__ rsin, const float* __restrict__ rcos,
+    short* __restrict__ dqkv, int B, int N, int H, int wsz) {
   const long long HD3 = 3LL * H * DH;
+  const long long HD2 = 2LL * H * DH;
   const long long total = (long long)B * N * 3 * H * (DH / 8);
   for (long long idx = blockIdx.x * 256LL + threadIdx.x; idx < total;
        idx += (long long)gridDim.x * 256) {
@@ -468,13 +470,21 @@ __global__ __launch_bounds__(256) void attn_bwd_finalize_kernel(
     const long long bn = rest / (3 * H);
     const int n = bn % N;
     const int d0 = g * 8;
-    (void)hslot;
 
-    const long long off = bn * HD3 + (long long)(rest % (3 * H)) * DH + d0;
+    const long long off = bn * HD3 + (long long)hslot * DH + d0;
     float x[8], sv[8], cv[8];
 #pragma unroll
     for (int j = 0; j < 8; ++j) x[j] = dacc[off + j];
+    // k/v slots: add the lookback contribution (exists unless this is
+    // the last window — its keys are nobody's lookback)
+    if (hslot >= H && (n / wsz) < (N / wsz) - 1) {
+      const long long loff = bn * HD2 + (long long)(hslot - H) * DH + d0;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) x[j] += dlook[loff + j];
+    }
     load_rope(rsin, rcos, n, d0, sv, cv);
+    // inverse rotation: dx[2i] = dy[2i] c + dy[2i+1] s;
+    //                   dx[2i+1] = dy[2i+1] c - dy[2i] s
 #pragma unroll
     for (int p = 0; p < 4; ++p) {
       float y0 = x[2 * p], y1 = x[2 * p + 1];
@@ -495,18 +505,16 @@ void attn_bwd_launch(const void* dout, const void* qkv, const float* rsin,
                      const float* rcos, const void* out, const float* lse,
                      float* dacc, float* dlook, void* dqkv, int B, int N,
                      int H, int wsz, hipStream_t stream) {
-  (void)dlook;  // retired: dK/dV combine via atomics into zeroed dacc
-  const int sub_per_win = (wsz + ATTN_WAVES * QB - 1) / (ATTN_WAVES * QB);
-  dim3 grid((N / wsz) * sub_per_win, H, B), block(ATTN_BLOCK);
-  size_t lds = 24576 + 49152 + 1024;  // 73 KiB
+  dim3 grid(N / wsz, H, B), block(ATTN_BLOCK);
+  size_t lds = 24576 + 131072 + 2048;  // 154 KiB
   attn_bwd_kernel<<<grid, block, lds, stream>>>(
-      (const short*)dout, (const short*)qkv, (const short*)out, lse, dacc, B,
-      N, H, wsz);
+      (const short*)dout, (const short*)qkv, (const short*)out, lse, dacc,
+      dlook, B, N, H, wsz);
   long long total = (long long)B * N * 3 * H * (DH / 8);
   int fin_grid = (int)((total + 255) / 256);
   if (fin_grid > 2048) fin_grid = 2048;
-  attn_bwd_finalize_kernel<<<fin_grid, 256, 0, stream>>>(dacc, rsin, rcos,
-                                                         (short*)dqkv, B, N, H);
+  attn_bwd_finalize_kernel<<<fin_grid, 256, 0, stream>>>(
+      dacc, dlook, rsin, rcos, (short*)dqkv, B, N, H, wsz);
 }
 
 }  // extern "C"

@@ -217,11 +217,11 @@ at::Tensor attn_bwd(const at::Tensor& dout, const at::Tensor& qkv,
   TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
   const int B = qkv.size(0), N = qkv.size(1);
   const int H = (int)heads, wsz = (int)window;
-  // dq region: plain stores by row owners; k/v regions: atomicAdd from
-  // the key-slice phases (q-row sub-blocks and adjacent windows share
-  // keys) -> zero-initialized
-  auto dacc = at::zeros_like(qkv, qkv.options().dtype(at::kFloat));
-  auto dlook = at::empty({1}, qkv.options().dtype(at::kFloat));  // retired
+  // plain-store accumulators: every element is written (dq by its row
+  // owner; own/lookback k,v by their single writing block) -> no zeroing
+  auto dacc = at::empty_like(qkv, qkv.options().dtype(at::kFloat));
+  auto dlook = at::empty({(long)B, (long)N, 2L * H * 64},
+                         qkv.options().dtype(at::kFloat));
   auto dqkv = at::empty_like(qkv);
   attn_bwd_launch(dout.data_ptr(), qkv.data_ptr(), rsin.data_ptr<float>(),
                   rcos.data_ptr<float>(), out.data_ptr(),
