@@ -27,7 +27,6 @@ from __future__ import annotations
 from typing import Callable, Optional
 
 import torch
-import torch.distributed as dist
 
 from .optim import ProGenAdamW
 from .parallel.ddp import DistributedTrainer

@@ -31,7 +31,7 @@ from progen_amd import ProGenBase, ProGenConfig
 from progen_amd.checkpoint import get_checkpoint_fns, numpy_to_tensors, tensors_to_numpy
 from progen_amd.data import decode_tokens, iterator_from_tfrecords_folder
 from progen_amd.optim import ProGenAdamW
-from progen_amd.parallel import DistributedTrainer, init_distributed, is_distributed
+from progen_amd.parallel import DistributedTrainer, init_distributed
 from progen_amd.utils import (compute_loss, confirm, exists, load_dotenv,
                               sample_fast, set_hardware_rng_)
 
