@@ -187,6 +187,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         for (int km = 0; km < 4; ++km)
 #pragma unroll
           for (int n = 0; n < MF; ++n) st[km][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        __builtin_amdgcn_s_setprio(1);  // favor the MFMA cluster (T5)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
 #pragma unroll
@@ -200,6 +201,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
                   kfrag, qfrag[n][ks], st[km][n], 0, 0, 0);
           }
         }
+        __builtin_amdgcn_s_setprio(0);
 
         // ---- mask + per-row max (in-lane over 16 keys, then x-lane) ----
         float tile_max[MF];
@@ -266,6 +268,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
               ((float*)&oacc[m][d])[r] *= alpha;
           }
         }
+        __builtin_amdgcn_s_setprio(1);  // favor the MFMA cluster (T5)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
 #pragma unroll
@@ -282,6 +285,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
             }
           }
         }
+        __builtin_amdgcn_s_setprio(0);
       }
 
       __syncthreads();
