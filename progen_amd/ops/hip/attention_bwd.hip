@@ -226,6 +226,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
         for (int m = 0; m < 4; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n) s[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        __builtin_amdgcn_s_setprio(1);  // T5: favor MFMA clusters
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
@@ -238,6 +239,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
               s[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                   qfrag[m][ks], kf, s[m][n], 0, 0, 0);
           }
+        __builtin_amdgcn_s_setprio(0);
 
         // P = exp(S - lse) masked; b64-write P^T into own pds region
 #pragma unroll
@@ -275,6 +277,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
         f32x4 dv[4];
 #pragma unroll
         for (int n = 0; n < 4; ++n) dv[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        __builtin_amdgcn_s_setprio(1);
         for (int c = c_min; c < nactive; ++c) {
           char* pds_c = pds_base + c * 8192;
           char* dot_c = dot_base + c * 8192;
@@ -292,6 +295,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             }
           }
         }
+        __builtin_amdgcn_s_setprio(0);
         const bool lookback = kb < wsz;  // tile-uniform half of the band
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
@@ -320,6 +324,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
         for (int m = 0; m < 4; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n) dp[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
@@ -336,6 +341,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
                   dof, vf, dp[m][n], 0, 0, 0);
             }
           }
+        __builtin_amdgcn_s_setprio(0);
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -372,6 +378,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
         f32x4 dk[4];
 #pragma unroll
         for (int n = 0; n < 4; ++n) dk[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        __builtin_amdgcn_s_setprio(1);
         for (int c = c_min; c < nactive; ++c) {
           char* pds_c = pds_base + c * 8192;
           char* qt_c = qt_base + c * 8192;
@@ -389,6 +396,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             }
           }
         }
+        __builtin_amdgcn_s_setprio(0);
         const bool lookback = kb < wsz;
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
