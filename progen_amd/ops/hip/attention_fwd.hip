@@ -72,10 +72,11 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
   const int v_off = 2 * H * DH + head * DH;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_lds = smem;                          // 8 KiB
-  char* v_lds = smem + KT * DH * 2;            // 8 KiB (V^T image)
-  char* p_lds = smem + 2 * KT * DH * 2 + wid * QB * KT * 2;  // 4 KiB/wave
-  float* bc_lds = (float*)(smem + 2 * KT * DH * 2 + ATTN_WAVES * QB * KT * 2 +
+  // double-buffered K and V^T tiles: stage tile t+1 into the other buffer
+  // while computing tile t -> ONE barrier per tile instead of two
+  char* kv_base = smem;                        // 2 x (8 + 8) KiB
+  char* p_lds = smem + 4 * KT * DH * 2 + wid * QB * KT * 2;  // 4 KiB/wave
+  float* bc_lds = (float*)(smem + 4 * KT * DH * 2 + ATTN_WAVES * QB * KT * 2 +
                            wid * 2 * QB * 4);  // [QB alpha | QB inv_l]/wave
 
   const float scale = rsqrtf((float)DH);
@@ -151,7 +152,9 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         }
       }
     };
-    auto write_lds = [&]() {
+    auto write_lds = [&](int buf) {
+      char* k_lds = kv_base + buf * (2 * KT * DH * 2);
+      char* v_lds = k_lds + KT * DH * 2;
 #pragma unroll
       for (int u = 0; u < 2; ++u) {
         const int key = su_key[u];
@@ -170,10 +173,12 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
     };
 
     issue_loads(0);
-    write_lds();
+    write_lds(0);
     __syncthreads();
 
     for (int t = 0; t < tiles; ++t) {
+      char* k_lds = kv_base + (t & 1) * (2 * KT * DH * 2);
+      char* v_lds = k_lds + KT * DH * 2;
       if (t + 1 < tiles) issue_loads(t + 1);
 
       if (active && t <= max_tile) {
@@ -288,11 +293,10 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         __builtin_amdgcn_s_setprio(0);
       }
 
+      // stage tile t+1 into the OTHER buffer: no reader conflict (its
+      // last readers finished before the barrier that ended tile t-1)
+      if (t + 1 < tiles) write_lds((t + 1) & 1);
       __syncthreads();
-      if (t + 1 < tiles) {
-        write_lds();
-        __syncthreads();
-      }
     }
 
     // ---- epilogue: O /= l, store out + lse ----
@@ -334,7 +338,7 @@ void attn_fwd_launch(const void* qkv_rot, void* out, float* lse, int B, int N,
                      int H, int wsz, hipStream_t stream) {
   const int sub_per_win = (wsz + ATTN_WAVES * QB - 1) / (ATTN_WAVES * QB);
   dim3 grid((N / wsz) * sub_per_win, H, B), block(ATTN_BLOCK);
-  size_t lds = (size_t)(2 * KT * DH * 2) + (size_t)ATTN_WAVES * QB * KT * 2 +
+  size_t lds = (size_t)(4 * KT * DH * 2) + (size_t)ATTN_WAVES * QB * KT * 2 +
                ATTN_WAVES * 2 * QB * 4;
   attn_fwd_kernel<<<grid, block, lds, stream>>>(
       (const short*)qkv_rot, (short*)out, lse, B, N, H, wsz);
