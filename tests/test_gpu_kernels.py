@@ -347,3 +347,27 @@ def test_graphed_step_matches_eager():
     assert err < 2e-3, err
     # device step counters advanced identically
     assert int(o1.step_dev.item()) == 5
+
+
+def test_default_toml_config_train_step():
+    """The reference default.toml config (window_size=512 > the 256-row
+    block: chunk-round path) must train end-to-end on the kernels."""
+    from progen_amd import ProGenBase, ProGenConfig
+    from progen_amd.optim import ProGenAdamW
+    from progen_amd.utils import compute_loss
+
+    cfg = ProGenConfig(num_tokens=256, dim=512, depth=6, heads=8, dim_head=64,
+                       window_size=512, seq_len=1024)
+    torch.manual_seed(31)
+    m = ProGenBase(cfg).to(device=dev(), dtype=torch.bfloat16)
+    opt = ProGenAdamW(m, lr=3e-4, max_grad_norm=0.5)
+    data = torch.randint(1, 256, (4, 1025), device=dev())
+    data[:, 0] = 0
+    losses = []
+    for _ in range(8):
+        loss = compute_loss(m, data)
+        loss.backward()
+        opt.micro_step()
+        losses.append(loss.item())
+    assert all(np.isfinite(losses))
+    assert losses[-1] < losses[0]
