@@ -1,0 +1,407 @@
+// Standalone ablation harness for the local-attention forward kernel.
+// Dev tool (not part of the extension): duplicates the production kernel
+// from progen_amd/ops/hip/attention_fwd.hip with #if VARIANT switches so
+// one gpurun call can compile and A/B several structural variants
+// in-process (guide §5.4 rules 9/24: within-probe interleaved rounds).
+//
+//   VARIANT 0: production structure (dbuf + T14 + setprio + stagger)
+//   VARIANT 1: no setprio
+//   VARIANT 2: single-buffered K/V (two barriers per tile)
+//   VARIANT 3: no T14 (loads issued at write time, not early)
+//   VARIANT 4: no bank-stagger on V^T scatter writes
+//
+// Build/run (on a GPU box):
+//   hipcc --offload-arch=gfx950 -O3 -std=c++17 -DVARIANT=0 \
+//       tools/ablate_attn.hip -o /tmp/ab0 && /tmp/ab0 32 24 1024 256
+
+#include <hip/hip_runtime.h>
+#include <cstdio>
+#include <cstdlib>
+#include <cmath>
+#include <vector>
+#include <cstring>
+#include "../progen_amd/ops/hip/common.h"
+
+#ifndef VARIANT
+#define VARIANT 0
+#endif
+
+#define DH 64
+#define KT 64
+#define ATTN_WAVES 4
+#define ATTN_BLOCK (ATTN_WAVES * WAVE)
+#define MF 2
+#define QB (MF * 16)
+#define NEG_INF (-1e30f)
+
+__device__ __forceinline__ int swz(int row, int byte_in_row) {
+  return (byte_in_row ^ ((row & 7) << 4));
+}
+
+__global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
+    const short* __restrict__ qkv, short* __restrict__ out,
+    float* __restrict__ lse_out, int B, int N, int H, int wsz) {
+  const int sub_per_win = (wsz + ATTN_WAVES * QB - 1) / (ATTN_WAVES * QB);
+  const int window = blockIdx.x / sub_per_win;
+  const int sub = blockIdx.x % sub_per_win;
+  const int head = blockIdx.y;
+  const int batch = blockIdx.z;
+
+  const int lane = threadIdx.x % WAVE;
+  const int wid = threadIdx.x / WAVE;
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+
+  const long long HD3 = 3LL * H * DH;
+  const long long qkv_bn = (long long)batch * N * HD3;
+  const int q_off = head * DH;
+  const int k_off = H * DH + head * DH;
+  const int v_off = 2 * H * DH + head * DH;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+#if VARIANT == 2
+  char* kv_base = smem;
+  char* p_lds = smem + 2 * KT * DH * 2 + wid * QB * KT * 2;
+  float* bc_lds = (float*)(smem + 2 * KT * DH * 2 + ATTN_WAVES * QB * KT * 2 +
+                           wid * 2 * QB * 4);
+#else
+  char* kv_base = smem;
+  char* p_lds = smem + 4 * KT * DH * 2 + wid * QB * KT * 2;
+  float* bc_lds = (float*)(smem + 4 * KT * DH * 2 + ATTN_WAVES * QB * KT * 2 +
+                           wid * 2 * QB * 4);
+#endif
+
+  const float scale = rsqrtf((float)DH);
+  const int tiles = 2 * wsz / KT;
+  const int chunks_all = wsz / QB;
+  const int chunks = min(ATTN_WAVES, chunks_all - sub * ATTN_WAVES);
+
+  const int su_key[2] = {(int)threadIdx.x >> 3,
+                         (int)(threadIdx.x + ATTN_BLOCK) >> 3};
+  const int su_d0[2] = {((int)threadIdx.x & 7) * 8,
+                        (((int)threadIdx.x + ATTN_BLOCK) & 7) * 8};
+
+  const bool active = wid < chunks;
+  const int chunk_off = (sub * ATTN_WAVES + wid) * QB;
+  const int q0 = window * wsz + chunk_off;
+
+  bf16x8 qfrag[MF][2];
+  if (active) {
+#pragma unroll
+    for (int m = 0; m < MF; ++m) {
+      const int row = q0 + m * 16 + l15;
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+        const int d0 = ks * 32 + 8 * l4;
+        bf16x8 v = *(const bf16x8*)(qkv + qkv_bn + (long long)row * HD3 +
+                                    q_off + d0);
+        bf16x8 o;
+#pragma unroll
+        for (int j = 0; j < 8; ++j)
+          ((short*)&o)[j] = f2bf(bf2f(((short*)&v)[j]) * scale);
+        qfrag[m][ks] = o;
+      }
+    }
+  }
+
+  float m_run[MF], l_run[MF];
+#pragma unroll
+  for (int n = 0; n < MF; ++n) {
+    m_run[n] = NEG_INF;
+    l_run[n] = 0.f;
+  }
+  f32x4 oacc[MF][4];
+#pragma unroll
+  for (int m = 0; m < MF; ++m)
+#pragma unroll
+    for (int d = 0; d < 4; ++d) oacc[m][d] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  const int max_tile = active ? ((chunk_off + QB - 1 + wsz) / KT) : -1;
+
+  bf16x8 kreg[2], vreg[2];
+  auto issue_loads = [&](int t) {
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      const int kpos = (window - 1) * wsz + t * KT + su_key[u];
+      if (kpos >= 0) {
+        const long long base = qkv_bn + (long long)kpos * HD3;
+        kreg[u] = *(const bf16x8*)(qkv + base + k_off + su_d0[u]);
+        vreg[u] = *(const bf16x8*)(qkv + base + v_off + su_d0[u]);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          ((short*)&kreg[u])[j] = 0;
+          ((short*)&vreg[u])[j] = 0;
+        }
+      }
+    }
+  };
+  auto write_lds = [&](int buf) {
+#if VARIANT == 2
+    char* k_lds = kv_base;
+    char* v_lds = k_lds + KT * DH * 2;
+    (void)buf;
+#else
+    char* k_lds = kv_base + buf * (2 * KT * DH * 2);
+    char* v_lds = k_lds + KT * DH * 2;
+#endif
+#pragma unroll
+    for (int u = 0; u < 2; ++u) {
+      const int key = su_key[u];
+      const int d0 = su_d0[u];
+      *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
+#pragma unroll
+      for (int jj = 0; jj < 8; ++jj) {
+#if VARIANT == 4
+        const int j = jj;
+#else
+        const int j = (jj + (int)threadIdx.x) & 7;
+#endif
+        const int d = d0 + j;
+        *(short*)(v_lds + d * 128 + swz(d, key * 2)) = ((short*)&vreg[u])[j];
+      }
+    }
+  };
+
+  issue_loads(0);
+  write_lds(0);
+  __syncthreads();
+
+  for (int t = 0; t < tiles; ++t) {
+#if VARIANT == 2
+    char* k_lds = kv_base;
+    char* v_lds = k_lds + KT * DH * 2;
+#else
+    char* k_lds = kv_base + (t & 1) * (2 * KT * DH * 2);
+    char* v_lds = k_lds + KT * DH * 2;
+#endif
+#if VARIANT != 3
+    if (t + 1 < tiles) issue_loads(t + 1);
+#endif
+
+    if (active && t <= max_tile) {
+      const int kb = t * KT;
+      const bool tile_full = (kb + KT - 1) <= chunk_off + wsz;
+
+      f32x4 st[4][MF];
+#pragma unroll
+      for (int km = 0; km < 4; ++km)
+#pragma unroll
+        for (int n = 0; n < MF; ++n) st[km][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+#if VARIANT != 1
+      __builtin_amdgcn_s_setprio(1);
+#endif
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+#pragma unroll
+        for (int km = 0; km < 4; ++km) {
+          const int key = km * 16 + l15;
+          const int d0 = ks * 32 + 8 * l4;
+          bf16x8 kfrag = *(const bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2));
+#pragma unroll
+          for (int n = 0; n < MF; ++n)
+            st[km][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                kfrag, qfrag[n][ks], st[km][n], 0, 0, 0);
+        }
+      }
+#if VARIANT != 1
+      __builtin_amdgcn_s_setprio(0);
+#endif
+
+      float tile_max[MF];
+#pragma unroll
+      for (int n = 0; n < MF; ++n) {
+        const int rowiw = chunk_off + n * 16 + l15;
+        float mx = NEG_INF;
+#pragma unroll
+        for (int km = 0; km < 4; ++km)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float v = ((float*)&st[km][n])[r];
+            if (!tile_full) {
+              const int kpos_band = kb + km * 16 + l4 * 4 + r;
+              if (kpos_band > rowiw + wsz) v = NEG_INF;
+              ((float*)&st[km][n])[r] = v;
+            }
+            mx = fmaxf(mx, v);
+          }
+        mx = fmaxf(mx, __shfl_xor(mx, 16, 64));
+        mx = fmaxf(mx, __shfl_xor(mx, 32, 64));
+        tile_max[n] = mx;
+      }
+
+#pragma unroll
+      for (int n = 0; n < MF; ++n) {
+        const float mnew = fmaxf(m_run[n], tile_max[n]);
+        const float alpha =
+            (m_run[n] == NEG_INF) ? 0.f : __expf(m_run[n] - mnew);
+        float psum = 0.f;
+        const int row = n * 16 + l15;
+#pragma unroll
+        for (int km = 0; km < 4; ++km) {
+          short pk[4];
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            float v = ((float*)&st[km][n])[r];
+            float p = (v <= NEG_INF) ? 0.f : __expf(v - mnew);
+            psum += p;
+            pk[r] = f2bf(p);
+          }
+          *(unsigned long long*)(p_lds + row * 128 +
+                                 swz(row, (km * 16 + l4 * 4) * 2)) =
+              *(unsigned long long*)pk;
+        }
+        psum += __shfl_xor(psum, 16, 64);
+        psum += __shfl_xor(psum, 32, 64);
+        l_run[n] = l_run[n] * alpha + psum;
+        m_run[n] = mnew;
+        if (l4 == 0) bc_lds[row] = alpha;
+      }
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+
+#pragma unroll
+      for (int m = 0; m < MF; ++m) {
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const float alpha = bc_lds[m * 16 + l4 * 4 + r];
+#pragma unroll
+          for (int d = 0; d < 4; ++d)
+            ((float*)&oacc[m][d])[r] *= alpha;
+        }
+      }
+#if VARIANT != 1
+      __builtin_amdgcn_s_setprio(1);
+#endif
+#pragma unroll
+      for (int ks = 0; ks < 2; ++ks) {
+#pragma unroll
+        for (int m = 0; m < MF; ++m) {
+          const int row = m * 16 + l15;
+          const int kk0 = ks * 32 + 8 * l4;
+          bf16x8 pfrag = *(const bf16x8*)(p_lds + row * 128 + swz(row, kk0 * 2));
+#pragma unroll
+          for (int d = 0; d < 4; ++d) {
+            const int dcol = d * 16 + l15;
+            bf16x8 vfrag = *(const bf16x8*)(v_lds + dcol * 128 + swz(dcol, kk0 * 2));
+            oacc[m][d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                pfrag, vfrag, oacc[m][d], 0, 0, 0);
+          }
+        }
+      }
+#if VARIANT != 1
+      __builtin_amdgcn_s_setprio(0);
+#endif
+    }
+
+#if VARIANT == 2
+    __syncthreads();
+    if (t + 1 < tiles) {
+#if VARIANT == 3
+      issue_loads(t + 1);
+#endif
+      write_lds(0);
+      __syncthreads();
+    }
+#else
+    if (t + 1 < tiles) {
+#if VARIANT == 3
+      issue_loads(t + 1);
+#endif
+      write_lds((t + 1) & 1);
+    }
+    __syncthreads();
+#endif
+  }
+
+  if (active) {
+#pragma unroll
+    for (int n = 0; n < MF; ++n) {
+      const int row = n * 16 + l15;
+      if (l4 == 0) {
+        bc_lds[QB + row] = 1.0f / l_run[n];
+        lse_out[((long long)batch * H + head) * N + q0 + row] =
+            m_run[n] + logf(l_run[n]);
+      }
+    }
+    asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+    const long long out_bn = ((long long)batch * N) * (long long)(H * DH);
+#pragma unroll
+    for (int m = 0; m < MF; ++m) {
+#pragma unroll
+      for (int r = 0; r < 4; ++r) {
+        const int row = q0 + m * 16 + l4 * 4 + r;
+        const float inv_l = bc_lds[QB + m * 16 + l4 * 4 + r];
+#pragma unroll
+        for (int d = 0; d < 4; ++d) {
+          const int dcol = d * 16 + l15;
+          out[out_bn + (long long)row * (H * DH) + head * DH + dcol] =
+              f2bf(((float*)&oacc[m][d])[r] * inv_l);
+        }
+      }
+    }
+  }
+}
+
+int main(int argc, char** argv) {
+  int B = argc > 1 ? atoi(argv[1]) : 32;
+  int H = argc > 2 ? atoi(argv[2]) : 24;
+  int N = argc > 3 ? atoi(argv[3]) : 1024;
+  int wsz = argc > 4 ? atoi(argv[4]) : 256;
+  int iters = argc > 5 ? atoi(argv[5]) : 200;
+
+  long long qn = (long long)B * N * 3 * H * DH;
+  long long on = (long long)B * N * H * DH;
+  std::vector<short> hq(qn);
+  srand(7);
+  for (long long i = 0; i < qn; ++i) {
+    float f = ((float)rand() / RAND_MAX - 0.5f) * 0.25f;  // random data
+    union { float f; unsigned u; } c;
+    c.f = f;
+    hq[i] = (short)(c.u >> 16);
+  }
+  short *dq, *dout_;
+  float* dlse;
+  hipMalloc(&dq, qn * 2);
+  hipMalloc(&dout_, on * 2);
+  hipMalloc(&dlse, (long long)B * H * N * 4);
+  hipMemcpy(dq, hq.data(), qn * 2, hipMemcpyHostToDevice);
+
+  const int sub_per_win = (wsz + ATTN_WAVES * QB - 1) / (ATTN_WAVES * QB);
+  dim3 grid((N / wsz) * sub_per_win, H, B), block(ATTN_BLOCK);
+#if VARIANT == 2
+  size_t lds = (size_t)(2 * KT * DH * 2) + (size_t)ATTN_WAVES * QB * KT * 2 +
+               ATTN_WAVES * 2 * QB * 4;
+#else
+  size_t lds = (size_t)(4 * KT * DH * 2) + (size_t)ATTN_WAVES * QB * KT * 2 +
+               ATTN_WAVES * 2 * QB * 4;
+#endif
+
+  // warmup
+  for (int i = 0; i < 20; ++i)
+    attn_fwd_kernel<<<grid, block, lds>>>(dq, dout_, dlse, B, N, H, wsz);
+  hipDeviceSynchronize();
+
+  hipEvent_t e0, e1;
+  hipEventCreate(&e0);
+  hipEventCreate(&e1);
+  hipEventRecord(e0);
+  for (int i = 0; i < iters; ++i)
+    attn_fwd_kernel<<<grid, block, lds>>>(dq, dout_, dlse, B, N, H, wsz);
+  hipEventRecord(e1);
+  hipEventSynchronize(e1);
+  float ms;
+  hipEventElapsedTime(&ms, e0, e1);
+  double us = ms * 1000.0 / iters;
+  // useful flops: QK^T + PV over the 2*wsz band (upper bound; masked
+  // tiles counted — consistent across variants)
+  double fl = 2.0 * B * H * (double)N * (2.0 * wsz) * DH * 2.0;
+  printf("VARIANT %d: %.1f us/call  %.1f TF/s  (B=%d H=%d N=%d wsz=%d)\n",
+         VARIANT, us, fl / (us * 1e-6) / 1e12, B, H, N, wsz);
+  // checksum for cross-variant comparison
+  std::vector<short> ho(on);
+  hipMemcpy(ho.data(), dout_, on * 2, hipMemcpyDeviceToHost);
+  double cs = 0;
+  for (long long i = 0; i < on; i += 97) cs += bf2f(ho[i]);
+  printf("checksum %.6f\n", cs);
+  return 0;
+}
