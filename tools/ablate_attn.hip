@@ -401,7 +401,11 @@ int main(int argc, char** argv) {
   std::vector<short> ho(on);
   hipMemcpy(ho.data(), dout_, on * 2, hipMemcpyDeviceToHost);
   double cs = 0;
-  for (long long i = 0; i < on; i += 97) cs += bf2f(ho[i]);
+  for (long long i = 0; i < on; i += 97) {
+    union { unsigned u; float f; } c;
+    c.u = ((unsigned)(unsigned short)ho[i]) << 16;
+    cs += c.f;
+  }
   printf("checksum %.6f\n", cs);
   return 0;
 }
