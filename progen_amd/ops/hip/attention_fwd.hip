@@ -160,12 +160,11 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         const int key = su_key[u];
         const int d0 = su_d0[u];
         *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
-        // stagger j by lane: concurrent lanes hit different (d&7) -> the
-        // XOR swizzle spreads the b16 writes over 8 bank groups instead
-        // of all landing on one (PMC: 16-way conflicts without this)
+        // NOTE: a lane-staggered write order was tried for bank spreading
+        // and measured 29% SLOWER (runtime-indexed vector extract goes to
+        // scratch; tools/ablate_attn.hip VARIANT 4) — keep static order.
 #pragma unroll
-        for (int jj = 0; jj < 8; ++jj) {
-          const int j = (jj + (int)threadIdx.x) & 7;
+        for (int j = 0; j < 8; ++j) {
           const int d = d0 + j;
           *(short*)(v_lds + d * 128 + swz(d, key * 2)) = ((short*)&vreg[u])[j];
         }

@@ -83,8 +83,7 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_fwd_kernel(
         bf16x8 v = *(const bf16x8*)(g_ln + bND + (long long)(t * 64 + k) * D +
                                     d0 + dd);
 #pragma unroll
-        for (int jj = 0; jj < 8; ++jj) {  // staggered: bank-spread writes
-          const int j = (jj + (int)threadIdx.x) & 7;
+        for (int j = 0; j < 8; ++j) {
           const int d = dd + j;
           *(short*)(gt_lds + d * 128 + swz(d, k * 2)) = ((short*)&v)[j];
         }
@@ -184,8 +183,7 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_dgate_kernel(
         bf16x8 v = *(const bf16x8*)(t_in + bND + (long long)(t * 64 + m) * D +
                                     d0 + dd);
 #pragma unroll
-        for (int jj = 0; jj < 8; ++jj) {  // staggered: bank-spread writes
-          const int j = (jj + (int)threadIdx.x) & 7;
+        for (int j = 0; j < 8; ++j) {
           const int d = dd + j;
           *(short*)(tt_lds + d * 128 + swz(d, m * 2)) = ((short*)&v)[j];
         }
@@ -207,8 +205,7 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_dgate_kernel(
         bf16x8 wv = *(const bf16x8*)(w + (long long)m * N + kglob);
         char* wt_region = smem + 8192 + (k8 / 64) * 8192;
 #pragma unroll
-        for (int jj = 0; jj < 8; ++jj) {  // staggered: bank-spread writes
-          const int j = (jj + (int)threadIdx.x) & 7;
+        for (int j = 0; j < 8; ++j) {
           const int k = (k8 + j) & 63;  // k index within the wave region
           short v = (m >= kglob + j) ? ((short*)&wv)[j] : (short)0;
           *(short*)(wt_region + k * 128 + swz(k, mm * 2)) = v;
