@@ -25,12 +25,25 @@
 #ifndef VARIANT
 #define VARIANT 0
 #endif
+// VARIANT 5: V^T image with padded 144-B row stride (conflict-free b128
+// reads without an XOR swizzle)
+#if VARIANT == 5
+#define VROW(d) ((d) * 144)
+#define VSWZ(d, b) (b)
+#define VBYTES (64 * 144)
+#else
+#define VROW(d) ((d) * 128)
+#define VSWZ(d, b) swz(d, b)
+#define VBYTES (64 * 128)
+#endif
 
 #define DH 64
 #define KT 64
 #define ATTN_WAVES 4
 #define ATTN_BLOCK (ATTN_WAVES * WAVE)
+#ifndef MF
 #define MF 2
+#endif
 #define QB (MF * 16)
 #define NEG_INF (-1e30f)
 
@@ -60,14 +73,16 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
 #if VARIANT == 2
+  const int KVSTRIDE = KT * DH * 2 + VBYTES;
   char* kv_base = smem;
-  char* p_lds = smem + 2 * KT * DH * 2 + wid * QB * KT * 2;
-  float* bc_lds = (float*)(smem + 2 * KT * DH * 2 + ATTN_WAVES * QB * KT * 2 +
+  char* p_lds = smem + KVSTRIDE + wid * QB * KT * 2;
+  float* bc_lds = (float*)(smem + KVSTRIDE + ATTN_WAVES * QB * KT * 2 +
                            wid * 2 * QB * 4);
 #else
+  const int KVSTRIDE = KT * DH * 2 + VBYTES;
   char* kv_base = smem;
-  char* p_lds = smem + 4 * KT * DH * 2 + wid * QB * KT * 2;
-  float* bc_lds = (float*)(smem + 4 * KT * DH * 2 + ATTN_WAVES * QB * KT * 2 +
+  char* p_lds = smem + 2 * KVSTRIDE + wid * QB * KT * 2;
+  float* bc_lds = (float*)(smem + 2 * KVSTRIDE + ATTN_WAVES * QB * KT * 2 +
                            wid * 2 * QB * 4);
 #endif
 
@@ -142,7 +157,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
     char* v_lds = k_lds + KT * DH * 2;
     (void)buf;
 #else
-    char* k_lds = kv_base + buf * (2 * KT * DH * 2);
+    char* k_lds = kv_base + buf * KVSTRIDE;
     char* v_lds = k_lds + KT * DH * 2;
 #endif
 #pragma unroll
@@ -151,14 +166,9 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
       const int d0 = su_d0[u];
       *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
 #pragma unroll
-      for (int jj = 0; jj < 8; ++jj) {
-#if VARIANT == 4
-        const int j = jj;
-#else
-        const int j = (jj + (int)threadIdx.x) & 7;
-#endif
+      for (int j = 0; j < 8; ++j) {
         const int d = d0 + j;
-        *(short*)(v_lds + d * 128 + swz(d, key * 2)) = ((short*)&vreg[u])[j];
+        *(short*)(v_lds + VROW(d) + VSWZ(d, key * 2)) = ((short*)&vreg[u])[j];
       }
     }
   };
@@ -172,7 +182,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
     char* k_lds = kv_base;
     char* v_lds = k_lds + KT * DH * 2;
 #else
-    char* k_lds = kv_base + (t & 1) * (2 * KT * DH * 2);
+    char* k_lds = kv_base + (t & 1) * KVSTRIDE;
     char* v_lds = k_lds + KT * DH * 2;
 #endif
 #if VARIANT != 3
@@ -282,7 +292,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 #pragma unroll
           for (int d = 0; d < 4; ++d) {
             const int dcol = d * 16 + l15;
-            bf16x8 vfrag = *(const bf16x8*)(v_lds + dcol * 128 + swz(dcol, kk0 * 2));
+            bf16x8 vfrag = *(const bf16x8*)(v_lds + VROW(dcol) + VSWZ(dcol, kk0 * 2));
             oacc[m][d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                 pfrag, vfrag, oacc[m][d], 0, 0, 0);
           }
@@ -369,10 +379,10 @@ int main(int argc, char** argv) {
   const int sub_per_win = (wsz + ATTN_WAVES * QB - 1) / (ATTN_WAVES * QB);
   dim3 grid((N / wsz) * sub_per_win, H, B), block(ATTN_BLOCK);
 #if VARIANT == 2
-  size_t lds = (size_t)(2 * KT * DH * 2) + (size_t)ATTN_WAVES * QB * KT * 2 +
+  size_t lds = (size_t)(KT * DH * 2 + VBYTES) + (size_t)ATTN_WAVES * QB * KT * 2 +
                ATTN_WAVES * 2 * QB * 4;
 #else
-  size_t lds = (size_t)(4 * KT * DH * 2) + (size_t)ATTN_WAVES * QB * KT * 2 +
+  size_t lds = 2 * (size_t)(KT * DH * 2 + VBYTES) + (size_t)ATTN_WAVES * QB * KT * 2 +
                ATTN_WAVES * 2 * QB * 4;
 #endif
 
