@@ -41,8 +41,8 @@ CONFIGS = {
 def main():
     p = argparse.ArgumentParser()
     p.add_argument("--gpus", type=int, default=1)
-    p.add_argument("--steps", type=int, default=8)
-    p.add_argument("--warmup", type=int, default=3)
+    p.add_argument("--steps", type=int, default=20)
+    p.add_argument("--warmup", type=int, default=8)
     p.add_argument("--model", default="progen-1.2b", choices=list(CONFIGS))
     p.add_argument("--batch", type=int, default=64, help="per-GPU batch size")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
