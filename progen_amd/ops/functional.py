@@ -12,8 +12,6 @@ also the numerics oracle for the kernels.
 
 from __future__ import annotations
 
-from typing import Optional, Tuple
-
 import torch
 
 from . import dispatch, reference

@@ -22,7 +22,6 @@ unbatched (N, ...) traced-through-vmap layout.
 
 from __future__ import annotations
 
-import math
 from typing import Optional, Tuple
 
 import torch

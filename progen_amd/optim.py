@@ -34,7 +34,6 @@ Math matches optax:
 
 from __future__ import annotations
 
-import math
 from typing import Dict, List, Optional, Tuple
 
 import torch

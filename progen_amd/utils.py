@@ -10,7 +10,7 @@ from __future__ import annotations
 
 import os
 import shutil
-from typing import Callable, Optional, Tuple
+from typing import Callable, Optional
 
 import torch
 

@@ -6,7 +6,6 @@ given string and emits a gumbel-max top-k sample.
 """
 
 import click
-import numpy as np
 import torch
 
 from progen_amd import ProGenBase, ProGenConfig
