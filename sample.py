@@ -11,7 +11,9 @@ import torch
 from progen_amd import ProGenBase, ProGenConfig
 from progen_amd.checkpoint import get_checkpoint_fns, numpy_to_tensors
 from progen_amd.data import decode_tokens, encode_tokens
-from progen_amd.utils import sample, sample_fast
+from progen_amd.utils import load_dotenv, sample, sample_fast
+
+load_dotenv()  # reference: sample.py:1-2
 
 
 @click.command()
