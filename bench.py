@@ -82,10 +82,11 @@ def main():
     data[:, -8:] = 0  # pad tail so the EOS-mask path is exercised
 
     graphed = None
-    # hipGraph step: single-GPU pure-replay only (the DP all-reduce between
-    # replays is an eager kernel, which corrupts replay state on this ROCm
-    # stack — profiles/r01_graph_interleave_bug.md)
-    if on_gpu and world == 1 and os.environ.get("PROGEN_NO_GRAPH") != "1":
+    # hipGraph step: pure-replay only (profiles/r01_graph_interleave_bug.md).
+    # Single-GPU by default; PROGEN_GRAPH_DP=1 opts the DP step (incl. the
+    # RCCL all-reduce) into the capture.
+    want_graph = world == 1 or os.environ.get("PROGEN_GRAPH_DP") == "1"
+    if on_gpu and want_graph and os.environ.get("PROGEN_NO_GRAPH") != "1":
         from progen_amd.runtime import GraphedTrainStep
         try:
             graphed = GraphedTrainStep(module, optim, ddp, B, N, device)

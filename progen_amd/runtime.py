@@ -15,8 +15,13 @@ poisons subsequent replays (NaN gradients or HSA aperture faults; see
 profiles/r01_graph_interleave_bug.md for the isolation). Training loops
 that validate/sample/checkpoint between steps must therefore run eager
 (train.py defaults to --no-graph); bench.py replays exclusively and is
-safe. DP (world>1) raises — its all-reduce would be an eager kernel
-between replays.
+safe. DP (world>1) raises by default — its all-reduce would be an
+eager kernel between replays. With PROGEN_GRAPH_DP=1 the WHOLE DP step
+(including the RCCL all-reduce and the optimizer) is captured instead,
+keeping the pure-replay rule: tools/rccl_graph_probe.py verified that
+RCCL collectives capture and replay correctly on this stack (1-rank
+group; multi-rank is unvalidated on this pool's 1-GPU boxes, hence the
+opt-in).
 
 Warmup runs on a side stream; optimizer/param state perturbed by warmup
 and capture is snapshotted and restored.
@@ -24,9 +29,11 @@ and capture is snapshotted and restored.
 
 from __future__ import annotations
 
+import os
 from typing import Callable, Optional
 
 import torch
+import torch.distributed as dist
 
 from .optim import ProGenAdamW
 from .parallel.ddp import DistributedTrainer
@@ -48,11 +55,11 @@ class GraphedTrainStep:
         self.ddp = ddp
         self.loss_fn = loss_fn
         self.world = ddp.world if ddp is not None else 1
-        if self.world > 1:
+        if self.world > 1 and os.environ.get("PROGEN_GRAPH_DP") != "1":
             raise RuntimeError(
-                "GraphedTrainStep is single-GPU pure-replay only: the DP "
-                "all-reduce between replays is an eager kernel, which "
-                "corrupts replay state on this ROCm stack")
+                "GraphedTrainStep under DP requires PROGEN_GRAPH_DP=1 "
+                "(captures the RCCL all-reduce inside the graph; see the "
+                "module docstring) — default DP runs eager")
         self.static_data = torch.zeros(batch, seq_len + 1, dtype=torch.long,
                                        device=device)
 
@@ -95,7 +102,15 @@ class GraphedTrainStep:
         return loss.detach()
 
     def _inner(self) -> torch.Tensor:
-        loss = self._fwd_bwd()
+        if self.world > 1:
+            # whole DP step in-graph: comm is part of the replay
+            assert self.ddp is not None
+            with self.ddp.no_sync():
+                loss = self._fwd_bwd()
+            dist.all_reduce(self.optim.space.flat_grad, op=dist.ReduceOp.SUM)
+            self.optim.space.flat_grad.div_(self.world)
+        else:
+            loss = self._fwd_bwd()
         self.optim.step()
         return loss
 
