@@ -282,7 +282,12 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
             sampled_str = decode_tokens(sampled[prime_length:].numpy())
             print(prime_str, '\n', '*' * 40, '\n', sampled_str)
             if wandb is not None:
-                wandb.log({'samples': sampled_str})
+                # same markup the reference renders via jinja2
+                # (reference: train.py:28,222)
+                html = (f"<i>{prime_str}</i><br/><br/>"
+                        f'<div style="overflow-wrap: break-word;">'
+                        f"{sampled_str}</div>")
+                wandb.log({'samples': wandb.Html(html)})
 
 
 if __name__ == '__main__':
