@@ -158,6 +158,12 @@ def main():
             },
         }))
 
+    # synchronized teardown (see train.py): a fast rank exiting early can
+    # SIGABRT a slower rank still inside process-group destruction
+    if world > 1:
+        torch.distributed.barrier()
+        torch.distributed.destroy_process_group()
+
 
 if __name__ == "__main__":
     main()

@@ -1,0 +1,207 @@
+"""Cached incremental decoding (serving path).
+
+The reference's sampler (reference: progen_transformer/utils.py:106-135)
+re-runs a full-length forward per emitted token — O(L * N^2). This module
+maintains per-layer recurrent state so each new token costs one O(window)
+attention row, one O(n) SGU row and the per-token GEMV projections:
+
+  - token shift needs the previous position's LN'd row (progen.py:43-46);
+  - windowed attention needs the rotated k/v of the current + previous
+    window only (progen.py:88-96) — window 0's zero lookback keys enter
+    the softmax denominator unmasked, which the step reproduces exactly;
+  - the SGU gate row m is sum_{n<=m} W[m,n]*gate_ln[n] + b[m]
+    (progen.py:179-182), so the LN'd gate history is the cache.
+
+Emitted tokens are identical to ``utils.sample`` up to fp reduction-order
+noise (tests/test_decode.py pins exact token equality on seeded fp32
+models). This is the O(1)-per-token analog of a KV cache for this
+architecture; ``sample_cached`` mirrors the sampler quirks (strict-> top-k
+with zeros-not-inf, gumbel-max, zero-after-second-pad).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.nn.functional as F
+
+from .models.progen import ProGenBase
+from .ops import reference as R
+
+
+def _ln_row(x: torch.Tensor, weight: torch.Tensor, eps: float = 1e-5) -> torch.Tensor:
+    """Scale-only LN of a (B, D) row (progen.py:22); stats in fp32."""
+    return R.layernorm_nobias(x.unsqueeze(1), weight, eps).squeeze(1)
+
+
+def _shift_row(y: torch.Tensor, prev: Optional[torch.Tensor]) -> torch.Tensor:
+    """Token shift at one position: first ceil(D/2) channels come from the
+    PREVIOUS position's LN'd row (zeros at position 0)  (progen.py:43-46)."""
+    d = y.shape[-1]
+    split = -(-d // 2)
+    head = torch.zeros_like(y[..., :split]) if prev is None else prev[..., :split]
+    return torch.cat((head, y[..., split:]), dim=-1)
+
+
+class DecodeCache:
+    """Recurrent state for one ``ProGenBase`` forward, batch-first (B, ...).
+
+    Buffers are allocated once at ``seq_len`` capacity (288 GB HBM3E makes
+    full-length caches the right trade on MI355X — no ring-buffer
+    bookkeeping on the hot path)."""
+
+    def __init__(self, model: ProGenBase, batch: int = 1,
+                 device=None, dtype=None):
+        cfg = model.cfg
+        p = next(model.parameters())
+        device = p.device if device is None else device
+        dtype = p.dtype if dtype is None else dtype
+        self.cfg = cfg
+        self.pos = 0
+        N, H, DH = cfg.seq_len, cfg.heads, cfg.dim_head
+        self.k = []    # per attn layer: (B, H, N, DH) rotated keys
+        self.v = []
+        self.attn_prev = [None] * cfg.depth   # prev LN row (attn branch)
+        self.ff_prev = [None] * cfg.depth     # prev LN row (ff branch)
+        self.gate_hist = []  # per SGU layer: (B, N, d2) LN'd gate rows
+        for _attn, ff in model.layers:
+            self.k.append(torch.zeros(batch, H, N, DH, device=device, dtype=dtype))
+            self.v.append(torch.zeros(batch, H, N, DH, device=device, dtype=dtype))
+            if ff.sgu is not None:
+                d2 = ff.sgu.norm_weight.shape[0]
+                self.gate_hist.append(
+                    torch.zeros(batch, N, d2, device=device, dtype=dtype))
+            else:
+                self.gate_hist.append(None)
+
+
+def _attn_step(attn, x, sin_p, cos_p, cache: DecodeCache, li: int) -> torch.Tensor:
+    """One LocalAttention row at position cache.pos. x: (B, dim)."""
+    p = cache.pos
+    H, DH, wsz = attn.heads, cache.cfg.dim_head, attn.window_size
+    y = _ln_row(x, attn.norm_weight)
+    y_in = _shift_row(y, cache.attn_prev[li]) if attn.shift_tokens else y
+    cache.attn_prev[li] = y
+    qkv = F.linear(y_in, attn.to_qkv.weight)            # (B, 3*H*DH)
+    B = qkv.shape[0]
+    q, k, v = qkv.view(B, 3, H, DH).unbind(1)
+    # interleaved rotary on q, k AND v at position p (progen.py:87)
+    sc, ss = cos_p.to(q.dtype), sin_p.to(q.dtype)
+    q, k, v = (t * sc + R.rotate_every_two(t) * ss for t in (q, k, v))
+    cache.k[li][:, :, p] = k
+    cache.v[li][:, :, p] = v
+
+    win = p // wsz
+    start = (win - 1) * wsz if win > 0 else 0
+    keys = cache.k[li][:, :, start:p + 1]               # (B, H, n, DH)
+    vals = cache.v[li][:, :, start:p + 1]
+    s = torch.einsum("bhd,bhnd->bhn", q, keys) * (DH ** -0.5)
+    if win == 0:
+        # window 0's zero lookback keys are UNMASKED (progen.py:90-96):
+        # wsz extra zero logits in the softmax, zero values
+        s = F.pad(s, (wsz, 0))
+    s = s - s.amax(dim=-1, keepdim=True)
+    a = s.softmax(dim=-1)
+    if win == 0:
+        a = a[..., wsz:]
+    out = torch.einsum("bhn,bhnd->bhd", a, vals).reshape(B, H * DH)
+    return F.linear(out, attn.to_out.weight, attn.to_out.bias)
+
+
+def _ff_step(ff, x, cache: DecodeCache, li: int) -> torch.Tensor:
+    """One FeedForward row at position cache.pos. x: (B, dim)."""
+    y = _ln_row(x, ff.norm_weight)
+    y_in = _shift_row(y, cache.ff_prev[li]) if ff.shift_tokens else y
+    cache.ff_prev[li] = y
+    h = F.linear(y_in, ff.proj_in.weight, ff.proj_in.bias)
+    if ff.glu:
+        a, g = h.chunk(2, dim=-1)
+        h = a * F.gelu(g, approximate="tanh")
+    else:
+        h = F.gelu(h, approximate="tanh")
+    if ff.sgu is not None:
+        p = cache.pos
+        xa, gate = h.chunk(2, dim=-1)
+        gate_ln = _ln_row(gate, ff.sgu.norm_weight)
+        hist = cache.gate_hist[li]
+        hist[:, p] = gate_ln
+        w_row = ff.sgu.spatial_weights[p, :p + 1]       # causal row (progen.py:179)
+        gate_out = torch.einsum("n,bnd->bd", w_row.to(hist.dtype),
+                                hist[:, :p + 1]) + ff.sgu.spatial_biases[p]
+        h = xa * gate_out
+        h = F.linear(h, ff.sgu.proj_out.weight, ff.sgu.proj_out.bias)
+    return F.linear(h, ff.proj_out.weight, ff.proj_out.bias)
+
+
+@torch.no_grad()
+def forward_step(model: ProGenBase, token: torch.Tensor,
+                 cache: DecodeCache) -> torch.Tensor:
+    """Advance the cache by one token; returns (B, V) logits at this
+    position — identical (up to fp noise) to row ``cache.pos`` of a full
+    ``model(seq)`` forward over the same prefix."""
+    p = cache.pos
+    assert p < model.cfg.seq_len, "decode past seq_len"
+    if model.rotary_sin.dtype != torch.float32:
+        sin, cos = R.fixed_pos_embedding(model.cfg.seq_len, model.cfg.dim_head,
+                                         device=model.rotary_sin.device)
+        model.rotary_sin, model.rotary_cos = sin, cos
+    sin_p, cos_p = model.rotary_sin[p], model.rotary_cos[p]
+    h = model.embed(token.long().reshape(-1))           # (B, dim)
+    for li, (attn, ff) in enumerate(model.layers):
+        h = h + _attn_step(attn, h, sin_p, cos_p, cache, li)
+        h = h + _ff_step(ff, h, cache, li)
+    h = _ln_row(h, model.final_norm_weight)
+    logits = F.linear(h, model.to_logits.weight, model.to_logits.bias)
+    cache.pos = p + 1
+    return logits
+
+
+@torch.no_grad()
+def sample_cached(
+    model: ProGenBase,
+    prime: torch.Tensor,
+    length: int,
+    top_k: Optional[int] = None,
+    add_bos: bool = False,
+    generator: Optional[torch.Generator] = None,
+    device=None,
+) -> torch.Tensor:
+    """Drop-in ``utils.sample`` with O(window + n_sgu) per-token cost.
+
+    Same decoding semantics (reference: utils.py:106-135): gumbel-max with
+    strict-> top-k masking (excluded logits -> 0, not -inf), pad-to-length
+    output, everything after the second pad/EOS zeroed. Stops forwarding
+    at EOS instead of emitting to full length."""
+    dev = next(model.parameters()).device if device is None else device
+    prime = torch.as_tensor(prime, device=dev).long().flatten()
+    start_pos = prime.shape[-1]
+    pad = (0, length - start_pos) if not add_bos else (1, length - start_pos - 1)
+    seq = F.pad(prime, pad)
+    if add_bos:
+        start_pos += 1
+
+    cache = DecodeCache(model, batch=1, device=dev)
+    logits = None
+    for p in range(start_pos):                           # prefill
+        logits = forward_step(model, seq[p:p + 1], cache)
+
+    pads_seen = int((seq[:start_pos] == 0).sum())
+    for curr_pos in range(start_pos, length):
+        logits_row = logits[0].float()
+        noise = R.gumbel_noise(logits_row.shape, generator=generator,
+                               device=logits_row.device)
+        if top_k is not None:
+            mask, logits_row = R.select_top_k(logits_row, top_k)
+            noise = noise * mask
+        sampled = (logits_row + noise).argmax(dim=-1)
+        seq[curr_pos] = sampled
+        if sampled.item() == 0:
+            pads_seen += 1
+            if pads_seen >= 2:
+                break
+        if curr_pos + 1 < length:
+            logits = forward_step(model, seq[curr_pos:curr_pos + 1], cache)
+
+    remove_after_eos = (seq == 0).long().cumsum(dim=-1) > 1
+    return seq * (~remove_after_eos).long()

@@ -289,6 +289,14 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
                         f"{sampled_str}</div>")
                 wandb.log({'samples': wandb.Html(html)})
 
+    # synchronized teardown: without the barrier a fast rank can exit and
+    # close its gloo/RCCL connections while a slower rank is still inside
+    # process-group destruction, which SIGABRTs the slower rank
+    if world > 1:
+        import torch.distributed as dist
+        dist.barrier()
+        dist.destroy_process_group()
+
 
 if __name__ == '__main__':
     main()
