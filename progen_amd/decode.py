@@ -174,7 +174,10 @@ def sample_cached(
     output, everything after the second pad/EOS zeroed. Stops forwarding
     at EOS instead of emitting to full length."""
     dev = next(model.parameters()).device if device is None else device
-    prime = torch.as_tensor(prime, device=dev).long().flatten()
+    # the sequence and sampling math live on CPU (matching utils.sample's
+    # generator semantics — its fn closure returns .cpu() logits); only
+    # the forward_step runs on the model device
+    prime = torch.as_tensor(prime).long().flatten().cpu()
     start_pos = prime.shape[-1]
     pad = (0, length - start_pos) if not add_bos else (1, length - start_pos - 1)
     seq = F.pad(prime, pad)
@@ -184,11 +187,11 @@ def sample_cached(
     cache = DecodeCache(model, batch=1, device=dev)
     logits = None
     for p in range(start_pos):                           # prefill
-        logits = forward_step(model, seq[p:p + 1], cache)
+        logits = forward_step(model, seq[p:p + 1].to(dev), cache)
 
     pads_seen = int((seq[:start_pos] == 0).sum())
     for curr_pos in range(start_pos, length):
-        logits_row = logits[0].float()
+        logits_row = logits[0].float().cpu()
         noise = R.gumbel_noise(logits_row.shape, generator=generator,
                                device=logits_row.device)
         if top_k is not None:
@@ -201,7 +204,8 @@ def sample_cached(
             if pads_seen >= 2:
                 break
         if curr_pos + 1 < length:
-            logits = forward_step(model, seq[curr_pos:curr_pos + 1], cache)
+            logits = forward_step(model, seq[curr_pos:curr_pos + 1].to(dev),
+                                  cache)
 
     remove_after_eos = (seq == 0).long().cumsum(dim=-1) > 1
     return seq * (~remove_after_eos).long()

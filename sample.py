@@ -22,7 +22,10 @@ load_dotenv()  # reference: sample.py:1-2
 @click.option('--prime', default='')
 @click.option('--fast', default=False, is_flag=True,
               help='length-growing decode with EOS early-exit (identical tokens)')
-def main(seed, checkpoint_path, prime, fast):
+@click.option('--cached', default=False, is_flag=True,
+              help='incremental decode with per-layer recurrent caches '
+                   '(O(window) per token; identical semantics)')
+def main(seed, checkpoint_path, prime, fast, cached):
     _, get_last_checkpoint, _ = get_checkpoint_fns(checkpoint_path)
     last_checkpoint = get_last_checkpoint()
     if last_checkpoint is None:
@@ -60,7 +63,11 @@ def main(seed, checkpoint_path, prime, fast):
         with torch.no_grad():
             return module(seq.to(device))[0].float().cpu()
 
-    if fast:
+    if cached:
+        from progen_amd.decode import sample_cached
+        sampled = sample_cached(module, prime_tensor, seq_len, top_k=25,
+                                add_bos=True, generator=g)
+    elif fast:
         sampled = sample_fast(fwd, prime_tensor, seq_len, top_k=25,
                               add_bos=True, generator=g,
                               window_size=cfg.window_size)
