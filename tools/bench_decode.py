@@ -5,7 +5,11 @@ Usage (GPU box):  python tools/bench_decode.py [--tokens 128]
 """
 
 import argparse
+import os
+import sys
 import time
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
 
 import torch
 
