@@ -73,7 +73,12 @@ class LocalAttention(nn.Module):
 
     def forward(self, x: torch.Tensor, sin: torch.Tensor, cos: torch.Tensor) -> torch.Tensor:
         x = OF.ln_shift(x, self.norm_weight, shift=self.shift_tokens)
-        qkv = self.to_qkv(x)
+        return self.inner(x, sin, cos)
+
+    def inner(self, y: torch.Tensor, sin: torch.Tensor, cos: torch.Tensor) -> torch.Tensor:
+        """Branch body AFTER the LN+shift prologue (the prologue is fused
+        with the residual add in ProGenBase.forward)."""
+        qkv = self.to_qkv(y)
         out = OF.local_attention(qkv, sin, cos, self.heads, self.window_size)
         return self.to_out(out)
 
@@ -118,6 +123,10 @@ class FeedForward(nn.Module):
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
         x = OF.ln_shift(x, self.norm_weight, shift=self.shift_tokens)
+        return self.inner(x)
+
+    def inner(self, x: torch.Tensor) -> torch.Tensor:
+        """Branch body AFTER the LN+shift prologue (see LocalAttention.inner)."""
         x = self.proj_in(x)
         if self.glu:
             x = OF.glu_gelu(x)
@@ -168,12 +177,21 @@ class ProGenBase(nn.Module):
             self.rotary_sin, self.rotary_cos = sin, cos
         sin = self.rotary_sin[:n]
         cos = self.rotary_cos[:n]
+        # residual adds are fused into each branch's LN+shift prologue
+        # (ops/hip/ln_shift.hip RES variant): ln_shift_res(h, r) returns
+        # the LN'd branch input AND the updated residual stream h + r,
+        # so `h = h + branch(h)` never runs as a separate pass
         h = self.embed(x.long())
+        r = None
         for attn, ff in self.layers:
-            h = h + attn(h, sin, cos)
-            h = h + ff(h)
-        h = OF.ln_shift(h, self.final_norm_weight, shift=False)
-        return self.to_logits(h)
+            y, h = OF.ln_shift_res(h, r, attn.norm_weight,
+                                   shift=attn.shift_tokens)
+            a = attn.inner(y, sin, cos)
+            y, h = OF.ln_shift_res(h, a, ff.norm_weight,
+                                   shift=ff.shift_tokens)
+            r = ff.inner(y)
+        y, _ = OF.ln_shift_res(h, r, self.final_norm_weight, shift=False)
+        return self.to_logits(y)
 
     def num_params(self) -> int:
         return sum(p.numel() for p in self.parameters())

@@ -47,6 +47,47 @@ def ln_shift(x: torch.Tensor, weight: torch.Tensor, shift: bool = True,
     return reference.ln_shift(x, weight, shift, eps)
 
 
+class _LnShiftResFn(torch.autograd.Function):
+    """Residual-add-fused LN+shift: forms s = x + res in-kernel (bf16
+    rounding identical to an eager add), normalizes s, and returns
+    (y, s) — s is the new residual stream. Backward returns the SAME
+    gradient tensor for x and res (d(x+res) fans out identically); both
+    are intermediate activations here, never leaves, so sharing is safe
+    and saves the autograd accumulation pass."""
+
+    @staticmethod
+    def forward(ctx, x, res, weight, shift: bool, eps: float):
+        C = dispatch.ext()
+        y, s, mean, rstd = C.ln_shift_res_fwd(x, res, weight, bool(shift),
+                                              float(eps))
+        ctx.save_for_backward(s, weight, mean, rstd)
+        ctx.shift = bool(shift)
+        ctx.set_materialize_grads(False)
+        return y, s
+
+    @staticmethod
+    def backward(ctx, dy, ds):
+        s, weight, mean, rstd = ctx.saved_tensors
+        C = dispatch.ext()
+        ds = ds.contiguous() if ds is not None else None
+        dx, dweight = C.ln_shift_res_bwd(dy.contiguous(), ds, s, weight,
+                                         mean, rstd, ctx.shift)
+        return dx, dx, dweight, None, None
+
+
+def ln_shift_res(x: torch.Tensor, res, weight: torch.Tensor,
+                 shift: bool = True, eps: float = 1e-5):
+    """Residual add + LN + shift in one pass: returns (y, s) with
+    s = x + res (the updated residual stream) and y = ln_shift(s).
+    ``res=None`` degenerates to plain ln_shift with s = x."""
+    if dispatch.use_hip(x):
+        if res is None:
+            return _LnShiftFn.apply(x, weight, shift, eps), x
+        return _LnShiftResFn.apply(x, res.contiguous(), weight, shift, eps)
+    s = x if res is None else x + res
+    return reference.ln_shift(s, weight, shift, eps), s
+
+
 # ---------------------------------------------------------------------------
 # fused local window attention (rotary + window + softmax + AV)
 # ---------------------------------------------------------------------------

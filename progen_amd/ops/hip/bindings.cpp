@@ -36,10 +36,34 @@ std::vector<at::Tensor> ln_shift_fwd(const at::Tensor& x, const at::Tensor& g,
   auto y = at::empty_like(x);
   auto mean = at::empty({(long)B * N}, x.options().dtype(at::kFloat));
   auto rstd = at::empty_like(mean);
-  ln_shift_fwd_launch(x.data_ptr(), g.data_ptr(), y.data_ptr(),
-                      mean.data_ptr<float>(), rstd.data_ptr<float>(), B * N, N,
-                      D, (float)eps, shift, bf, cur_stream());
+  ln_shift_fwd_launch(x.data_ptr(), nullptr, g.data_ptr(), y.data_ptr(),
+                      nullptr, mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                      B * N, N, D, (float)eps, shift, bf, cur_stream());
   return {y, mean, rstd};
+}
+
+// residual-fused variant: s = x + res is formed in-kernel (bf16 rounding
+// identical to an eager add), LN runs on s, and s is returned as the new
+// residual stream — the separate elementwise add disappears.
+std::vector<at::Tensor> ln_shift_res_fwd(const at::Tensor& x,
+                                         const at::Tensor& res,
+                                         const at::Tensor& g, bool shift,
+                                         double eps) {
+  TORCH_CHECK(x.is_cuda() && x.is_contiguous() && x.dim() == 3);
+  TORCH_CHECK(res.is_contiguous() && res.sizes() == x.sizes() &&
+              res.scalar_type() == x.scalar_type());
+  bool bf = check_dtype(x);
+  const int B = x.size(0), N = x.size(1), D = x.size(2);
+  TORCH_CHECK(D % 16 == 0, "ln_shift: D must be a multiple of 16");
+  auto y = at::empty_like(x);
+  auto s = at::empty_like(x);
+  auto mean = at::empty({(long)B * N}, x.options().dtype(at::kFloat));
+  auto rstd = at::empty_like(mean);
+  ln_shift_fwd_launch(x.data_ptr(), res.data_ptr(), g.data_ptr(), y.data_ptr(),
+                      s.data_ptr(), mean.data_ptr<float>(),
+                      rstd.data_ptr<float>(), B * N, N, D, (float)eps, shift,
+                      bf, cur_stream());
+  return {y, s, mean, rstd};
 }
 
 std::vector<at::Tensor> ln_shift_bwd(const at::Tensor& dy, const at::Tensor& x,
@@ -53,11 +77,41 @@ std::vector<at::Tensor> ln_shift_bwd(const at::Tensor& dy, const at::Tensor& x,
   int nblocks = std::min(R, 1024);  // >=4 blocks/CU for latency hiding
   auto dx = at::empty_like(x);
   auto dw_part = at::empty({nblocks, D}, x.options().dtype(at::kFloat));
-  ln_shift_bwd_launch(dy.data_ptr(), x.data_ptr(), g.data_ptr(),
+  ln_shift_bwd_launch(dy.data_ptr(), nullptr, x.data_ptr(), g.data_ptr(),
                       mean.data_ptr<float>(), rstd.data_ptr<float>(),
                       dx.data_ptr(), dw_part.data_ptr<float>(), nblocks, R, N,
                       D, shift, bf, cur_stream());
   auto dw = dw_part.sum(0).to(x.scalar_type());
+  return {dx, dw};
+}
+
+// backward of the residual-fused variant: ``s_in`` is the saved summed
+// stream (stats were computed on it); ``ds`` (optional) is the gradient
+// flowing into s from its later uses. The returned dx is the gradient of
+// BOTH addends (d(x + res) fans out identically).
+std::vector<at::Tensor> ln_shift_res_bwd(const at::Tensor& dy,
+                                         const c10::optional<at::Tensor>& ds,
+                                         const at::Tensor& s_in,
+                                         const at::Tensor& g,
+                                         const at::Tensor& mean,
+                                         const at::Tensor& rstd, bool shift) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && s_in.is_contiguous());
+  bool bf = check_dtype(s_in);
+  const int B = s_in.size(0), N = s_in.size(1), D = s_in.size(2);
+  const int R = B * N;
+  int nblocks = std::min(R, 1024);
+  const void* ds_ptr = nullptr;
+  if (ds.has_value()) {
+    TORCH_CHECK(ds->is_contiguous() && ds->sizes() == s_in.sizes());
+    ds_ptr = ds->data_ptr();
+  }
+  auto dx = at::empty_like(s_in);
+  auto dw_part = at::empty({nblocks, D}, s_in.options().dtype(at::kFloat));
+  ln_shift_bwd_launch(dy.data_ptr(), ds_ptr, s_in.data_ptr(), g.data_ptr(),
+                      mean.data_ptr<float>(), rstd.data_ptr<float>(),
+                      dx.data_ptr(), dw_part.data_ptr<float>(), nblocks, R, N,
+                      D, shift, bf, cur_stream());
+  auto dw = dw_part.sum(0).to(s_in.scalar_type());
   return {dx, dw};
 }
 
@@ -275,6 +329,10 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgu_dw", &sgu_dw, "SGU backward: dW");
   m.def("ln_shift_fwd", &ln_shift_fwd, "fused LN+shift forward");
   m.def("ln_shift_bwd", &ln_shift_bwd, "fused LN+shift backward");
+  m.def("ln_shift_res_fwd", &ln_shift_res_fwd,
+        "residual-add-fused LN+shift forward");
+  m.def("ln_shift_res_bwd", &ln_shift_res_bwd,
+        "residual-add-fused LN+shift backward");
   m.def("glu_fwd", &glu_fwd, "GLU-GELU forward");
   m.def("glu_bwd", &glu_bwd, "GLU-GELU backward");
   m.def("gelu_fwd", &gelu_fwd, "GELU forward");

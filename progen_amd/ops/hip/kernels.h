@@ -4,13 +4,14 @@
 
 extern "C" {
 
-void ln_shift_fwd_launch(const void* x, const void* g, void* y, float* mean,
-                         float* rstd, int R, int N, int D, float eps,
-                         bool shift, bool is_bf16, hipStream_t stream);
-void ln_shift_bwd_launch(const void* dy, const void* x, const void* g,
-                         const float* mean, const float* rstd, void* dx,
-                         float* dw_part, int nblocks, int R, int N, int D,
-                         bool shift, bool is_bf16, hipStream_t stream);
+void ln_shift_fwd_launch(const void* x, const void* res, const void* g,
+                         void* y, void* s_out, float* mean, float* rstd,
+                         int R, int N, int D, float eps, bool shift,
+                         bool is_bf16, hipStream_t stream);
+void ln_shift_bwd_launch(const void* dy, const void* ds, const void* x,
+                         const void* g, const float* mean, const float* rstd,
+                         void* dx, float* dw_part, int nblocks, int R, int N,
+                         int D, bool shift, bool is_bf16, hipStream_t stream);
 
 void glu_fwd_launch(const void* h, void* y, long long rows, int H,
                     bool is_bf16, hipStream_t stream);

@@ -19,9 +19,14 @@
 // forward
 // ---------------------------------------------------------------------------
 
-template <typename VEC, bool SHIFT, bool IS_BF16>
+// RES: fuse the residual add s = x + res (bf16 rounding identical to an
+// eager bf16 add); stats and y are computed on s, and s is written out as
+// the new residual stream — the separate elementwise add (and its
+// backward-side grad accumulation) disappears.
+template <typename VEC, bool SHIFT, bool IS_BF16, bool RES>
 __global__ __launch_bounds__(LN_BLOCK) void ln_shift_fwd_kernel(
-    const VEC* __restrict__ x, const VEC* __restrict__ g, VEC* __restrict__ y,
+    const VEC* __restrict__ x, const VEC* __restrict__ res,
+    const VEC* __restrict__ g, VEC* __restrict__ y, VEC* __restrict__ s_out,
     float* __restrict__ mean, float* __restrict__ rstd, int N, int Dv,
     float eps) {
   // Dv = D / VLEN (vector units); each VEC is 8 bf16 or 4 f32 (16 B)
@@ -35,6 +40,20 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_fwd_kernel(
   float s = 0.f, ss = 0.f;
   for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
     VEC v = x[base + i];
+    if (RES) {
+      VEC rv = res[base + i];
+      VEC sv;
+#pragma unroll
+      for (int j = 0; j < VLEN; ++j) {
+        if (IS_BF16) {
+          ((short*)&sv)[j] = f2bf(bf2f(((short*)&v)[j]) + bf2f(((short*)&rv)[j]));
+        } else {
+          ((float*)&sv)[j] = ((float*)&v)[j] + ((float*)&rv)[j];
+        }
+      }
+      s_out[base + i] = sv;
+      v = sv;
+    }
 #pragma unroll
     for (int j = 0; j < VLEN; ++j) {
       float f = IS_BF16 ? bf2f(((short*)&v)[j]) : ((float*)&v)[j];
@@ -79,7 +98,8 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_fwd_kernel(
 
   const int halfv = (D / 2) / VLEN;  // D even, half % VLEN == 0 (checked host-side)
   for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
-    VEC v = x[base + i];
+    // with RES the summed row was just written: the re-read hits L1/L2
+    VEC v = RES ? s_out[base + i] : x[base + i];
     VEC gw = g[i];
     VEC o;
 #pragma unroll
@@ -114,9 +134,13 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_fwd_kernel(
 //   e[d >= half] = dy[row][d]
 // dx = rs * (e*g - mean(e*g) - xhat * mean(e*g*xhat));  dw += e * xhat
 
-template <typename VEC, bool SHIFT, bool IS_BF16>
+// DS: also add the gradient flowing into the summed stream s (the fused
+// variant returns s as a second output; dx then serves as the gradient of
+// BOTH addends, since d(x + res) fans out identically).
+template <typename VEC, bool SHIFT, bool IS_BF16, bool DS>
 __global__ __launch_bounds__(LN_BLOCK) void ln_shift_bwd_kernel(
-    const VEC* __restrict__ dy, const VEC* __restrict__ x,
+    const VEC* __restrict__ dy, const VEC* __restrict__ ds,
+    const VEC* __restrict__ x,
     const VEC* __restrict__ g, const float* __restrict__ mean,
     const float* __restrict__ rstd, VEC* __restrict__ dx,
     float* __restrict__ dw_part, int R, int N, int Dv) {
@@ -200,6 +224,8 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_bwd_kernel(
       }
       VEC v = x[base + i];
       VEC gw = g[i];
+      VEC dsv;
+      if (DS) dsv = ds[base + i];
       VEC o;
 #pragma unroll
       for (int j = 0; j < VLEN; ++j) {
@@ -208,6 +234,7 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_bwd_kernel(
         float gj = IS_BF16 ? bf2f(((short*)&gw)[j]) : ((float*)&gw)[j];
         float xh = (f - mu) * rs;
         float dxv = rs * (ej * gj - bsum * inv_d - xh * a * inv_d);
+        if (DS) dxv += IS_BF16 ? bf2f(((short*)&dsv)[j]) : ((float*)&dsv)[j];
         if (IS_BF16) ((short*)&o)[j] = f2bf(dxv);
         else ((float*)&o)[j] = dxv;
         dw_lds[i * VLEN + j] += ej * xh;  // thread-private index: no race
@@ -227,56 +254,47 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_bwd_kernel(
 
 extern "C" {
 
-void ln_shift_fwd_launch(const void* x, const void* g, void* y, float* mean,
-                         float* rstd, int R, int N, int D, float eps,
-                         bool shift, bool is_bf16, hipStream_t stream) {
+void ln_shift_fwd_launch(const void* x, const void* res, const void* g,
+                         void* y, void* s_out, float* mean, float* rstd,
+                         int R, int N, int D, float eps, bool shift,
+                         bool is_bf16, hipStream_t stream) {
   dim3 grid(R), block(LN_BLOCK);
+#define LNF(VEC, SH, BF, RS)                                              \
+  ln_shift_fwd_kernel<VEC, SH, BF, RS><<<grid, block, 0, stream>>>(       \
+      (const VEC*)x, (const VEC*)res, (const VEC*)g, (VEC*)y,             \
+      (VEC*)s_out, mean, rstd, N, Dv, eps)
   if (is_bf16) {
     int Dv = D / 8;
-    if (shift)
-      ln_shift_fwd_kernel<bf16x8, true, true><<<grid, block, 0, stream>>>(
-          (const bf16x8*)x, (const bf16x8*)g, (bf16x8*)y, mean, rstd, N, Dv, eps);
-    else
-      ln_shift_fwd_kernel<bf16x8, false, true><<<grid, block, 0, stream>>>(
-          (const bf16x8*)x, (const bf16x8*)g, (bf16x8*)y, mean, rstd, N, Dv, eps);
+    if (shift) { if (res) LNF(bf16x8, true, true, true); else LNF(bf16x8, true, true, false); }
+    else       { if (res) LNF(bf16x8, false, true, true); else LNF(bf16x8, false, true, false); }
   } else {
     int Dv = D / 4;
-    if (shift)
-      ln_shift_fwd_kernel<f32x4, true, false><<<grid, block, 0, stream>>>(
-          (const f32x4*)x, (const f32x4*)g, (f32x4*)y, mean, rstd, N, Dv, eps);
-    else
-      ln_shift_fwd_kernel<f32x4, false, false><<<grid, block, 0, stream>>>(
-          (const f32x4*)x, (const f32x4*)g, (f32x4*)y, mean, rstd, N, Dv, eps);
+    if (shift) { if (res) LNF(f32x4, true, false, true); else LNF(f32x4, true, false, false); }
+    else       { if (res) LNF(f32x4, false, false, true); else LNF(f32x4, false, false, false); }
   }
+#undef LNF
 }
 
-void ln_shift_bwd_launch(const void* dy, const void* x, const void* g,
-                         const float* mean, const float* rstd, void* dx,
-                         float* dw_part, int nblocks, int R, int N, int D,
-                         bool shift, bool is_bf16, hipStream_t stream) {
+void ln_shift_bwd_launch(const void* dy, const void* ds, const void* x,
+                         const void* g, const float* mean, const float* rstd,
+                         void* dx, float* dw_part, int nblocks, int R, int N,
+                         int D, bool shift, bool is_bf16, hipStream_t stream) {
   dim3 grid(nblocks), block(LN_BLOCK);
   size_t lds = (size_t)D * 4 + LN_WAVES * 4 + 16;
+#define LNB(VEC, SH, BF, WDS)                                             \
+  ln_shift_bwd_kernel<VEC, SH, BF, WDS><<<grid, block, lds, stream>>>(    \
+      (const VEC*)dy, (const VEC*)ds, (const VEC*)x, (const VEC*)g, mean, \
+      rstd, (VEC*)dx, dw_part, R, N, Dv)
   if (is_bf16) {
     int Dv = D / 8;
-    if (shift)
-      ln_shift_bwd_kernel<bf16x8, true, true><<<grid, block, lds, stream>>>(
-          (const bf16x8*)dy, (const bf16x8*)x, (const bf16x8*)g, mean, rstd,
-          (bf16x8*)dx, dw_part, R, N, Dv);
-    else
-      ln_shift_bwd_kernel<bf16x8, false, true><<<grid, block, lds, stream>>>(
-          (const bf16x8*)dy, (const bf16x8*)x, (const bf16x8*)g, mean, rstd,
-          (bf16x8*)dx, dw_part, R, N, Dv);
+    if (shift) { if (ds) LNB(bf16x8, true, true, true); else LNB(bf16x8, true, true, false); }
+    else       { if (ds) LNB(bf16x8, false, true, true); else LNB(bf16x8, false, true, false); }
   } else {
     int Dv = D / 4;
-    if (shift)
-      ln_shift_bwd_kernel<f32x4, true, false><<<grid, block, lds, stream>>>(
-          (const f32x4*)dy, (const f32x4*)x, (const f32x4*)g, mean, rstd,
-          (f32x4*)dx, dw_part, R, N, Dv);
-    else
-      ln_shift_bwd_kernel<f32x4, false, false><<<grid, block, lds, stream>>>(
-          (const f32x4*)dy, (const f32x4*)x, (const f32x4*)g, mean, rstd,
-          (f32x4*)dx, dw_part, R, N, Dv);
+    if (shift) { if (ds) LNB(f32x4, true, false, true); else LNB(f32x4, true, false, false); }
+    else       { if (ds) LNB(f32x4, false, false, true); else LNB(f32x4, false, false, false); }
   }
+#undef LNB
 }
 
 }  // extern "C"

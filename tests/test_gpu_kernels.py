@@ -51,6 +51,41 @@ def test_ln_shift_fwd_bwd(dtype, tol, shift):
     assert rel_err(g.grad, g32.grad) < tol * 3
 
 
+@pytest.mark.parametrize("dtype,tol", [(torch.bfloat16, 2e-2), (torch.float32, 1e-5)])
+@pytest.mark.parametrize("shift", [True, False])
+@pytest.mark.parametrize("use_ds", [True, False])
+def test_ln_shift_res_fwd_bwd(dtype, tol, shift, use_ds):
+    torch.manual_seed(2)
+    B, N, D = 2, 32, 128
+    x = torch.randn(B, N, D, device=dev(), dtype=dtype, requires_grad=True)
+    r = torch.randn(B, N, D, device=dev(), dtype=dtype, requires_grad=True)
+    g = torch.randn(D, device=dev(), dtype=dtype, requires_grad=True)
+
+    y, s = OF.ln_shift_res(x, r, g, shift=shift)
+    dy = torch.randn_like(y)
+    if use_ds:
+        ds = torch.randn_like(s)
+        torch.autograd.backward([y, s], [dy, ds])
+    else:
+        y.backward(dy)
+
+    x32 = x.detach().float().cpu().requires_grad_(True)
+    r32 = r.detach().float().cpu().requires_grad_(True)
+    g32 = g.detach().float().cpu().requires_grad_(True)
+    s32 = x32 + r32
+    y32 = R.ln_shift(s32, g32, shift=shift)
+    if use_ds:
+        torch.autograd.backward([y32, s32], [dy.float().cpu(), ds.float().cpu()])
+    else:
+        y32.backward(dy.float().cpu())
+
+    assert rel_err(y, y32) < tol
+    assert rel_err(s, s32) < tol
+    assert rel_err(x.grad, x32.grad) < tol * 3
+    assert rel_err(r.grad, r32.grad) < tol * 3
+    assert rel_err(g.grad, g32.grad) < tol * 3
+
+
 # ---------------------------------------------------------------------------
 # glu / gelu
 # ---------------------------------------------------------------------------
