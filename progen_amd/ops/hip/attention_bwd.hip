@@ -307,12 +307,8 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
 #pragma unroll
             for (int n = 0; n < 4; ++n) {
               float v = ((float*)&dv[n])[r];
-              // round 0 zero-covers the whole band with plain stores;
-              // later rounds use fire-and-forget fp32 atomics — the
-              // dependent load+add+store chain was 25% of kernel time
-              // (tools/ablate_attn_bwd.hip VARIANT 2)
-              if (round > 0) atomicAdd(dst + n * 16 + l15, v);
-              else dst[n * 16 + l15] = v;
+              if (round > 0) v += dst[n * 16 + l15];  // later chunk rounds
+              dst[n * 16 + l15] = v;
             }
           }
         }
@@ -412,8 +408,8 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
 #pragma unroll
             for (int n = 0; n < 4; ++n) {
               float v = ((float*)&dk[n])[r];
-              if (round > 0) atomicAdd(dst + n * 16 + l15, v);
-              else dst[n * 16 + l15] = v;
+              if (round > 0) v += dst[n * 16 + l15];
+              dst[n * 16 + l15] = v;
             }
           }
         }
