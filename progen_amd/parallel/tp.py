@@ -96,6 +96,27 @@ class _ReduceFromTP(torch.autograd.Function):
         return dx
 
 
+class _SumBothTP(torch.autograd.Function):
+    """All-reduce forward AND backward. Needed when the reduced value
+    feeds DIFFERENT computations on each rank (e.g. distributed LN
+    statistics normalizing rank-local channel shards): dL/dS then has a
+    distinct per-rank part that must itself be summed. (_ReduceFromTP's
+    identity backward is only correct when the consumers of the reduced
+    value are replicated, as in a row-parallel linear.)"""
+
+    @staticmethod
+    def forward(ctx, x):
+        x = x.contiguous().clone()
+        dist.all_reduce(x, group=_TP_GROUP)
+        return x
+
+    @staticmethod
+    def backward(ctx, dy):
+        dy = dy.contiguous().clone()
+        dist.all_reduce(dy, group=_TP_GROUP)
+        return dy
+
+
 class _GatherFromTP(torch.autograd.Function):
     """All-gather shards along the last dim; backward slices out this
     rank's piece (feature-dim gather has no cross-rank grad mixing)."""
@@ -121,6 +142,10 @@ def reduce_from_tp(x: torch.Tensor) -> torch.Tensor:
     return _ReduceFromTP.apply(x) if tp_size() > 1 else x
 
 
+def sum_both_tp(x: torch.Tensor) -> torch.Tensor:
+    return _SumBothTP.apply(x) if tp_size() > 1 else x
+
+
 def _glu_shard_rows(out_features: int, tp: int, rank: int) -> torch.Tensor:
     """Row indices for a GLU-paired column shard: matching slices of the
     value half and the gate half, so chunk(2) of the sharded output pairs
@@ -140,7 +165,8 @@ class ColumnParallelLinear(torch.nn.Module):
     nonlinearity directly)."""
 
     def __init__(self, in_features: int, out_features: int, bias: bool = True,
-                 shard_glu: bool = False, gather_output: bool = False):
+                 shard_glu: bool = False, gather_output: bool = False,
+                 rows: Optional[torch.Tensor] = None):
         super().__init__()
         tp, rank = tp_size(), tp_rank()
         assert out_features % tp == 0, (out_features, tp)
@@ -151,6 +177,11 @@ class ColumnParallelLinear(torch.nn.Module):
         self.local_out = out_features // tp
         self.shard_glu = shard_glu
         self.gather_output = gather_output
+        # custom shard rows (e.g. the QKV head shard takes this rank's
+        # head block in EACH of the q/k/v sections)
+        self.custom_rows = rows
+        if rows is not None:
+            assert rows.numel() == self.local_out, (rows.numel(), self.local_out)
         self.weight = torch.nn.Parameter(
             torch.empty(self.local_out, in_features))
         self.bias = torch.nn.Parameter(
@@ -171,6 +202,8 @@ class ColumnParallelLinear(torch.nn.Module):
 
     def _rows(self) -> torch.Tensor:
         tp, rank = tp_size(), tp_rank()
+        if self.custom_rows is not None:
+            return self.custom_rows
         if self.shard_glu:
             return _glu_shard_rows(self.out_features, tp, rank)
         per = self.out_features // tp
