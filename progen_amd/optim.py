@@ -112,6 +112,12 @@ class ProGenAdamW:
         # correct Adam bias correction without host involvement
         self.step_dev = torch.zeros(1, dtype=torch.int32, device=flat.device) \
             if flat.is_cuda else None
+        # optional hook returning the GLOBAL grad sum-of-squares (0-dim
+        # tensor) for clip-by-global-norm under model parallelism: under
+        # TP the local flat buffer holds only this rank's shards, so the
+        # true norm needs a cross-rank reduction
+        # (progen_amd/parallel/tp_model.py::tp_grad_sumsq_fn)
+        self.norm_sumsq_fn = None
 
         # per-chunk decay flags: weight decay only on ndim>1 params
         # (reference: train.py:115 exclude_norm_and_bias_params)
@@ -170,10 +176,14 @@ class ProGenAdamW:
 
     # -- implementations ----------------------------------------------------
 
-    def _clip_coef(self, g32: torch.Tensor) -> torch.Tensor:
+    def _clip_coef(self, g32: torch.Tensor,
+                   grad_scale: float = 1.0) -> torch.Tensor:
         if self.max_grad_norm is None:
             return torch.ones((), device=g32.device)
-        norm = torch.linalg.vector_norm(g32)
+        if self.norm_sumsq_fn is not None:
+            norm = self.norm_sumsq_fn().sqrt() * grad_scale
+        else:
+            norm = torch.linalg.vector_norm(g32)
         # optax clip_by_global_norm: g * max_norm / max(norm, max_norm)
         return self.max_grad_norm / torch.clamp_min(norm, self.max_grad_norm)
 
@@ -191,7 +201,7 @@ class ProGenAdamW:
 
     def _step_eager(self, grad_scale: float) -> None:
         g32 = self.space.flat_grad.float() * grad_scale
-        g32 *= self._clip_coef(g32)
+        g32 *= self._clip_coef(g32, grad_scale)
         self.master.sub_(self._adamw_update(g32))
         if self.is_low_precision:
             self.space.flat.copy_(self.master.to(self.space.flat.dtype))
@@ -218,6 +228,9 @@ class ProGenAdamW:
         C = dispatch.ext()
         if self.max_grad_norm is None:
             clip_coef = torch.ones(1, device=self.master.device)
+        elif self.norm_sumsq_fn is not None:
+            norm = self.norm_sumsq_fn().sqrt() * grad_scale
+            clip_coef = self.max_grad_norm / torch.clamp_min(norm, self.max_grad_norm)
         else:
             # norm(scale*g) = scale*norm(g): fused sumsq kernel, no fp32
             # grad copy, clip coefficient stays on device (no host sync)

@@ -143,6 +143,51 @@ def tp_shard_(model: ProGenBase) -> ProGenBase:
     return model
 
 
+def sharded_params(model: ProGenBase) -> List[torch.nn.Parameter]:
+    """Params whose flat-buffer content is rank-LOCAL (a shard of the
+    full model): column/row linear weights (and column biases) and the
+    channel-sharded SGU gate-LN scale. Everything else is replicated."""
+    out = []
+    for mod in model.modules():
+        if isinstance(mod, tp.ColumnParallelLinear):
+            out.append(mod.weight)
+            if mod.bias is not None:
+                out.append(mod.bias)
+        elif isinstance(mod, tp.RowParallelLinear):
+            out.append(mod.weight)  # bias is replicated (added post-reduce)
+        elif isinstance(mod, TPSGU):
+            out.append(mod.norm_weight)
+    return out
+
+
+def tp_grad_sumsq_fn(model: ProGenBase):
+    """Returns a callable computing the GLOBAL grad sum-of-squares for
+    clip-by-global-norm: sharded params' local sumsq is all-reduced
+    across the TP group; replicated params (identical grads on every
+    rank) are counted once. Attach as ``optim.norm_sumsq_fn``."""
+    shard_ids = {id(p) for p in sharded_params(model)}
+
+    def fn() -> torch.Tensor:
+        s_sh = None
+        s_rep = None
+        for p in model.parameters():
+            if p.grad is None:
+                continue
+            v = (p.grad.float() ** 2).sum()
+            if id(p) in shard_ids:
+                s_sh = v if s_sh is None else s_sh + v
+            else:
+                s_rep = v if s_rep is None else s_rep + v
+        dev = next(model.parameters()).device
+        s_sh = s_sh if s_sh is not None else torch.zeros((), device=dev)
+        s_rep = s_rep if s_rep is not None else torch.zeros((), device=dev)
+        s_sh = s_sh.contiguous().clone()
+        dist.all_reduce(s_sh, group=tp.tp_group())
+        return s_sh + s_rep
+
+    return fn
+
+
 def replicated_partial_grad_params(model: ProGenBase) -> List[torch.nn.Parameter]:
     """Replicated params whose per-rank gradient is PARTIAL (local
     channels only): the SGU spatial weights/biases."""
