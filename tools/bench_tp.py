@@ -33,6 +33,8 @@ def main():
     ap.add_argument("--steps", type=int, default=20)
     ap.add_argument("--warmup", type=int, default=5)
     ap.add_argument("--batch", type=int, default=8)
+    ap.add_argument("--model", default="progen_6b",
+                    help="configs/model/<name>.toml")
     args = ap.parse_args()
 
     world = int(os.environ.get("WORLD_SIZE", "1"))
@@ -45,9 +47,14 @@ def main():
         torch.cuda.set_device(local_rank)
     device = torch.device("cuda", local_rank) if on_gpu else torch.device("cpu")
 
-    cfg = ProGenConfig(num_tokens=256, dim=4096, depth=24, heads=32,
-                       dim_head=128, window_size=512, seq_len=2048,
-                       ff_glu=True, global_mlp_depth=2)
+    try:
+        import tomllib
+    except ModuleNotFoundError:
+        import tomli as tomllib
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    with open(os.path.join(repo, "configs", "model",
+                           f"{args.model}.toml"), "rb") as f:
+        cfg = ProGenConfig.from_dict(tomllib.load(f))
     torch.manual_seed(1234)  # identical full weights on every rank
     model = ProGenBase(cfg)
     if world > 1:
@@ -63,7 +70,7 @@ def main():
 
     B, N = args.batch, cfg.seq_len
     g = torch.Generator().manual_seed(7)
-    data = torch.randint(1, 256, (B, N + 1), generator=g).to(device)
+    data = torch.randint(1, cfg.num_tokens, (B, N + 1), generator=g).to(device)
     data[:, 0] = 0
 
     def step():
@@ -108,7 +115,7 @@ def main():
             "scaling": "strong",
             "dtype": "bf16" if on_gpu else "fp32",
             "data": "synthetic",
-            "config": {"model": "ProGen-6B", "global_batch": B,
+            "config": {"model": args.model, "global_batch": B,
                        "seq_len": N, "parallelism": f"tp{world}"},
         }))
     if world > 1:
