@@ -209,3 +209,63 @@ def sample_cached(
 
     remove_after_eos = (seq == 0).long().cumsum(dim=-1) > 1
     return seq * (~remove_after_eos).long()
+
+
+@torch.no_grad()
+def sample_cached_batch(
+    model: ProGenBase,
+    primes,
+    length: int,
+    top_k: Optional[int] = None,
+    generator: Optional[torch.Generator] = None,
+    device=None,
+) -> torch.Tensor:
+    """Batched incremental decode: one forward_step advances ALL rows
+    (the per-layer caches are batch-first), so serving throughput scales
+    with batch at the same per-step latency.
+
+    ``primes``: list of 1-D int tensors (ragged). Shorter primes are
+    left-padded into the batch by decoding their sampled continuations
+    only after their own prime ends — implemented by overwriting
+    positions < len(prime_i) with the prime tokens. Rows that hit their
+    second pad/EOS keep emitting into a dead tail that is zeroed at the
+    end (same zero-after-second-pad rule as ``utils.sample``). With
+    ``top_k=None`` and ``generator=None`` decoding is greedy argmax and
+    each row's output equals its single-row ``sample_cached`` run
+    (tests/test_decode.py::test_sample_cached_batch_greedy_parity).
+    """
+    dev = next(model.parameters()).device if device is None else device
+    primes = [torch.as_tensor(p).long().flatten().cpu() for p in primes]
+    B = len(primes)
+    lens = [int(p.shape[-1]) for p in primes]
+    seq = torch.zeros(B, length, dtype=torch.long)
+    for i, p in enumerate(primes):
+        seq[i, :lens[i]] = p
+
+    cache = DecodeCache(model, batch=B, device=dev)
+    greedy = top_k is None and generator is None
+    # per-row pad counts over WRITTEN tokens only (the unfilled zero tail
+    # must not count as EOS); position 0 is written at entry
+    pads = [int(seq[i, 0] == 0) for i in range(B)]
+    for pos in range(length - 1):
+        logits = forward_step(model, seq[:, pos].to(dev), cache)  # (B, V)
+        rows = logits.float().cpu()
+        if greedy:
+            nxt = rows.argmax(dim=-1)
+        else:
+            noise = R.gumbel_noise(rows.shape, generator=generator)
+            if top_k is not None:
+                mask, rows = R.select_top_k(rows, top_k)
+                noise = noise * mask
+            nxt = (rows + noise).argmax(dim=-1)
+        w = pos + 1
+        # keep prime tokens where the prime extends past this position
+        for i in range(B):
+            if w >= lens[i]:
+                seq[i, w] = nxt[i]
+            pads[i] += int(seq[i, w] == 0)
+        if all(p >= 2 for p in pads):
+            break  # every row has emitted its EOS
+
+    remove_after_eos = (seq == 0).long().cumsum(dim=-1) > 1
+    return seq * (~remove_after_eos).long()

@@ -89,3 +89,24 @@ def model_fn(model):
         with torch.no_grad():
             return model(seq.unsqueeze(0))[0]
     return fn
+
+
+def test_sample_cached_batch_greedy_parity():
+    from progen_amd.decode import sample_cached_batch, DecodeCache, forward_step
+    model = _tiny()
+    primes = [torch.randint(1, 256, (4,)), torch.randint(1, 256, (7,))]
+    out = sample_cached_batch(model, primes, 24)
+
+    # each row must equal its own single-row greedy decode
+    for i, p in enumerate(primes):
+        cache = DecodeCache(model, batch=1)
+        seq = torch.zeros(24, dtype=torch.long)
+        seq[:p.shape[0]] = p
+        for pos in range(23):
+            logits = forward_step(model, seq[pos:pos + 1], cache)
+            nxt = logits[0].float().argmax()
+            if pos + 1 >= p.shape[0]:
+                seq[pos + 1] = nxt
+        z = (seq == 0).long().cumsum(-1) > 1
+        seq = seq * (~z).long()
+        assert torch.equal(out[i], seq)
