@@ -67,11 +67,15 @@ def main():
     module.rotary_sin = module.rotary_sin.float()
     module.rotary_cos = module.rotary_cos.float()
 
-    optim = ProGenAdamW(module, lr=2e-4, weight_decay=1e-3, max_grad_norm=0.5)
+    if world > 1 and os.environ.get("PROGEN_ZERO1") == "1":
+        from progen_amd.parallel.zero1 import Zero1AdamW
+        optim = Zero1AdamW(module, lr=2e-4, weight_decay=1e-3, max_grad_norm=0.5)
+    else:
+        optim = ProGenAdamW(module, lr=2e-4, weight_decay=1e-3, max_grad_norm=0.5)
     ddp = DistributedTrainer(optim.space)
     if world > 1:  # belt-and-braces: bitwise-identical replicas
         torch.distributed.broadcast(optim.space.flat, src=0)
-        optim.master.copy_(optim.space.flat.float())
+        optim.resync_master()
     torch.manual_seed(1234 + rank)  # rank-local data stream
 
     B, N = args.batch, cfg.seq_len

@@ -143,6 +143,11 @@ class ProGenAdamW:
     def zero_grad(self) -> None:
         self.space.zero_grad()
 
+    def resync_master(self) -> None:
+        """Re-derive the fp32 master from the (possibly just broadcast)
+        flat param buffer."""
+        self.master.copy_(self.space.flat.float())
+
     def grad_norm(self) -> torch.Tensor:
         return torch.linalg.vector_norm(self.space.flat_grad.float())
 

@@ -1,0 +1,86 @@
+"""ZeRO-1 optimizer-state sharding over the flat parameter space.
+
+The reference has no optimizer sharding (its optax state is fully
+replicated under pmap, reference: train.py:117-121). On MI355X the fp32
+master + two Adam moments for ProGen-1.2B are ~14 GB per GPU; sharding
+them across the DP group divides that by N at the cost of one
+all-gather of the updated bf16 params per step.
+
+Design (v1, opt-in via PROGEN_ZERO1=1 in train.py/bench.py):
+  - gradients are still bucket-all-reduced by DistributedTrainer (every
+    rank holds the full mean gradient — so the global-norm clip needs no
+    extra communication);
+  - each rank runs AdamW on its contiguous 1/N slice of the flat buffer
+    (fp32 master/exp_avg/exp_avg_sq exist only for that slice);
+  - the updated param slices are all-gathered back into every rank's
+    flat buffer (bf16: 2 bytes/param on the wire).
+
+A reduce-scatter of the gradients (instead of all-reduce) would also
+halve the gradient communication — that needs bucket/shard alignment in
+the DDP overlap path and RCCL reduce_scatter (gloo, used by the CPU
+tests, lacks it), so it stays on the round-2 list (TODO.md).
+"""
+
+from __future__ import annotations
+
+from typing import Optional
+
+import torch
+import torch.distributed as dist
+
+from ..optim import ProGenAdamW
+
+
+class Zero1AdamW(ProGenAdamW):
+    """ProGenAdamW with optimizer state sharded across the DP group.
+
+    Must be constructed AFTER init_distributed; gradients must be
+    all-reduced (mean) before ``step`` — exactly what DistributedTrainer
+    already does."""
+
+    def __init__(self, module: torch.nn.Module, **kwargs):
+        super().__init__(module, **kwargs)
+        assert dist.is_initialized() and dist.get_world_size() > 1, \
+            "Zero1AdamW requires an initialized process group (world > 1)"
+        self.world = dist.get_world_size()
+        self.rank = dist.get_rank()
+        n = self.space.numel
+        self.shard = -(-n // self.world)          # padded shard size
+        self.lo = min(self.rank * self.shard, n)
+        self.hi = min(self.lo + self.shard, n)
+
+        # replace the full-size fp32 state with the local slice
+        sl = slice(self.lo, self.hi)
+        self.master = self.master[sl].clone()
+        self.exp_avg = torch.zeros_like(self.master)
+        self.exp_avg_sq = torch.zeros_like(self.master)
+        self._decay_mask = self._decay_mask[sl].clone()
+
+    def _step_eager(self, grad_scale: float) -> None:
+        g_full = self.space.flat_grad.float() * grad_scale
+        coef = self._clip_coef(g_full, grad_scale)  # full-grad norm: no comm
+        g32 = g_full[self.lo:self.hi] * coef
+        self.master.sub_(self._adamw_update(g32))
+        # scatter the updated slice into this rank's padded gather buffer
+        flat = self.space.flat
+        buf = torch.zeros(self.shard, dtype=flat.dtype, device=flat.device)
+        buf[: self.hi - self.lo] = self.master.to(flat.dtype)
+        out = torch.empty(self.shard * self.world, dtype=flat.dtype,
+                          device=flat.device)
+        dist.all_gather_into_tensor(out, buf)
+        flat.copy_(out[: flat.numel()])
+
+    def _step_hip(self, grad_scale: float) -> None:
+        # v1 runs the sharded update with torch ops (the fused adamw
+        # kernel addresses full-space chunk tables; a shard-local chunk
+        # table is round-2 work — TODO.md). Correctness is identical.
+        self._step_eager(grad_scale)
+
+    def resync_master(self) -> None:
+        self.master.copy_(self.space.flat[self.lo:self.hi].float())
+
+    def state_dict(self):
+        d = super().state_dict()
+        d["zero1"] = {"world": self.world, "rank": self.rank,
+                      "lo": self.lo, "hi": self.hi}
+        return d

@@ -126,15 +126,20 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
     module.rotary_sin = module.rotary_sin.float()
     module.rotary_cos = module.rotary_cos.float()
 
-    optim = ProGenAdamW(module, lr=learning_rate, weight_decay=weight_decay,
-                        max_grad_norm=max_grad_norm,
-                        accum_mode=grad_accum_mode,
-                        grad_accum_every=grad_accum_every)
+    opt_kwargs = dict(lr=learning_rate, weight_decay=weight_decay,
+                      max_grad_norm=max_grad_norm,
+                      accum_mode=grad_accum_mode,
+                      grad_accum_every=grad_accum_every)
+    if world > 1 and os.environ.get('PROGEN_ZERO1') == '1':
+        from progen_amd.parallel.zero1 import Zero1AdamW
+        optim = Zero1AdamW(module, **opt_kwargs)
+    else:
+        optim = ProGenAdamW(module, **opt_kwargs)
     ddp = DistributedTrainer(optim.space)
     if world > 1:  # DP replicas must start bitwise-identical
         import torch.distributed as dist
         dist.broadcast(optim.space.flat, src=0)
-        optim.master.copy_(optim.space.flat.float())
+        optim.resync_master()
 
     start_seq_index = 0
     if exists(last_checkpoint):
@@ -142,7 +147,7 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
             k: torch.as_tensor(v).to(device=device)
             for k, v in numpy_to_tensors(last_checkpoint['params']).items()
         }, strict=False)  # copies in place -> flat param buffer updated
-        optim.master.copy_(optim.space.flat.float())
+        optim.resync_master()
         if exists(last_checkpoint.get('optim_state')):
             optim.load_state_dict(numpy_to_tensors(last_checkpoint['optim_state']))
         start_seq_index = last_checkpoint['next_seq_index']
