@@ -89,7 +89,11 @@ def test_generate_train_resume_sample(workdir):
              if ln.startswith("starting from sequence")][0]
     assert int(start.split()[-1]) > 0
 
-    # sample from the checkpoint
+    # sample from the checkpoint — all three decoder paths
     r = CliRunner().invoke(sample_cli.main, ["--prime", "# M"])
     assert r.exit_code == 0, r.output
     assert "params:" in r.output
+    r = CliRunner().invoke(sample_cli.main, ["--prime", "# M", "--fast"])
+    assert r.exit_code == 0, r.output
+    r = CliRunner().invoke(sample_cli.main, ["--prime", "# M", "--cached"])
+    assert r.exit_code == 0, r.output
