@@ -27,7 +27,6 @@ Sharding rules for the ProGen block (docs/tp_design.md):
 
 from __future__ import annotations
 
-import os
 from typing import Optional
 
 import torch

@@ -23,8 +23,6 @@ tests, lacks it), so it stays on the round-2 list (TODO.md).
 
 from __future__ import annotations
 
-from typing import Optional
-
 import torch
 import torch.distributed as dist
 

@@ -4,8 +4,6 @@ surface the code touches — the real client needs network + credentials,
 but the save/get-last/prune protocol is backend logic and is covered
 here (reference parity: checkpoint.py:41-81)."""
 
-import shutil
-
 import torch
 
 from progen_amd import checkpoint as cp

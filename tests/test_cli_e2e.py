@@ -1,10 +1,8 @@
 """End-to-end slice: synthetic fasta -> generate_data -> train (CPU,
 tiny) -> checkpoint -> resume -> sample  (SURVEY.md §7.3)."""
 
-import os
 import random
 
-import numpy as np
 import pytest
 from click.testing import CliRunner
 

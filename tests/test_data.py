@@ -1,7 +1,6 @@
 """TFRecord wire format, collate, tokenizer, resume-skip tests."""
 
 import gzip
-import struct
 
 import numpy as np
 import pytest

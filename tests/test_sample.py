@@ -1,6 +1,5 @@
 """Sampler semantics tests (reference: utils.py:97-135)."""
 
-import numpy as np
 import torch
 
 from progen_amd import ProGen, ProGenBase, ProGenConfig
