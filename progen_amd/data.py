@@ -224,6 +224,36 @@ def collate_fn(batch: List[bytes], pad_length: int, offset: int = 0) -> np.ndarr
     return np.stack(padded)
 
 
+def prefetch_iter(it: Iterator, depth: int = 2) -> Iterator:
+    """Background-thread prefetch (the reference pipeline's
+    tf.data .prefetch(AUTOTUNE), data.py:62): decompression/collate of
+    the next batches overlaps the training step. Exceptions in the
+    producer re-raise in the consumer."""
+    import queue
+    import threading
+
+    q: "queue.Queue" = queue.Queue(maxsize=max(1, depth))
+    sentinel = object()
+
+    def worker():
+        try:
+            for item in it:
+                q.put(item)
+        except BaseException as e:  # noqa: BLE001 — forwarded to consumer
+            q.put(e)
+            return
+        q.put(sentinel)
+
+    threading.Thread(target=worker, daemon=True).start()
+    while True:
+        item = q.get()
+        if item is sentinel:
+            return
+        if isinstance(item, BaseException):
+            raise item
+        yield item
+
+
 def iterator_from_tfrecords_folder(folder: str, data_type: str = "train"):
     """Returns (num_seqs, iter_fn) (reference: data.py:37-72).
 
@@ -239,7 +269,8 @@ def iterator_from_tfrecords_folder(folder: str, data_type: str = "train"):
     filenames = sorted(str(p) for p in folder_p.glob(f"**/*.{data_type}.tfrecord.gz"))
     num_seqs = sum(int(f.split(".")[-4]) for f in filenames)
 
-    def iter_fn(seq_len: int, batch_size: int, skip: int = 0, loop: bool = False):
+    def iter_fn(seq_len: int, batch_size: int, skip: int = 0,
+                loop: bool = False, prefetch: int = 2):
         def gen_sequences():
             while True:
                 for fname in filenames:
@@ -254,14 +285,20 @@ def iterator_from_tfrecords_folder(folder: str, data_type: str = "train"):
             except StopIteration:
                 return
 
-        batch: List[bytes] = []
-        for seq in gen:
-            batch.append(seq)
-            if len(batch) == batch_size:
+        def gen_batches():
+            batch: List[bytes] = []
+            for seq in gen:
+                batch.append(seq)
+                if len(batch) == batch_size:
+                    yield _finalize_batch(batch, seq_len)
+                    batch = []
+            if batch:
                 yield _finalize_batch(batch, seq_len)
-                batch = []
-        if batch:
-            yield _finalize_batch(batch, seq_len)
+
+        if prefetch > 0:
+            yield from prefetch_iter(gen_batches(), depth=prefetch)
+        else:
+            yield from gen_batches()
 
     return num_seqs, iter_fn
 

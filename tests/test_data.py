@@ -95,3 +95,19 @@ def test_tokenizer_roundtrip():
     assert D.decode_tokens(np.array(toks)) == s
     # token 0 decodes to '' (PAD/BOS/EOS; reference data.py:79-81)
     assert D.decode_tokens(np.array([0])) == ""
+
+
+def test_prefetch_iter_order_and_errors():
+    from progen_amd.data import prefetch_iter
+    assert list(prefetch_iter(iter(range(100)), depth=3)) == list(range(100))
+    assert list(prefetch_iter(iter([]), depth=2)) == []
+
+    def boom():
+        yield 1
+        raise ValueError("producer failed")
+
+    it = prefetch_iter(boom(), depth=2)
+    assert next(it) == 1
+    import pytest
+    with pytest.raises(ValueError, match="producer failed"):
+        next(it)
