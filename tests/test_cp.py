@@ -15,7 +15,7 @@ def _free_port() -> int:
         return s.getsockname()[1]
 
 
-def _worker(rank, world, port, q):
+def _worker(rank, world, port, q):  # noqa: C901
     import torch.distributed as dist
     os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -72,8 +72,10 @@ def _worker(rank, world, port, q):
 
 
 @pytest.mark.timeout(180)
-def test_cp_parity_world2():
-    world = 2
+@pytest.mark.parametrize("world", [2, 4])
+def test_cp_parity(world):
+    # world=4 exercises MIDDLE ranks (simultaneous halo send+recv),
+    # which world=2 cannot
     port = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.Queue()

@@ -109,3 +109,66 @@ def test_zero1_matches_single_process():
             # equally — ProGen's CE is per-sequence mean then batch mean,
             # so mean-of-half-means == full mean for equal halves
             assert abs(a - b) < 1e-9, (rank, got, want)
+
+
+def _odd_worker(rank, world, port, q):
+    import torch.distributed as dist
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from progen_amd.optim import ProGenAdamW
+    from progen_amd.parallel.zero1 import Zero1AdamW
+    try:
+        torch.manual_seed(17)
+
+        class M(torch.nn.Module):
+            def __init__(self):
+                super().__init__()
+                self.lin = torch.nn.Linear(3, 4)      # 16 params
+                self.scale = torch.nn.Parameter(torch.ones(1))  # -> 17 (odd)
+
+            def forward(self, x):
+                return self.lin(x) * self.scale
+
+        def run(cls, **kw):
+            torch.manual_seed(17)
+            m = M().double()
+            opt = cls(m, lr=1e-2, **kw)
+            torch.manual_seed(5)
+            losses = []
+            for _ in range(3):
+                x = torch.randn(8, 3, dtype=torch.float64)
+                loss = (m(x) ** 2).mean()
+                loss.backward()
+                opt.step()
+                opt.zero_grad()
+                losses.append(loss.item())
+            return losses, opt.space.flat.detach().clone()
+
+        want, want_flat = run(ProGenAdamW)
+        got, got_flat = run(Zero1AdamW)
+        assert got_flat.numel() == 17  # odd: last shard is smaller
+        for a, b in zip(got, want):
+            assert abs(a - b) < 1e-12, (got, want)
+        torch.testing.assert_close(got_flat, want_flat)
+        q.put((rank, "ok"))
+    except Exception as e:
+        import traceback
+        q.put((rank, f"fail: {e}\n{traceback.format_exc()[-1000:]}"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_zero1_odd_numel_padding():
+    world = 2
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_odd_worker, args=(r, world, port, q))
+          for r in range(world)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=90) for _ in range(world)]
+    for p in ps:
+        p.join(timeout=60)
+    assert all(msg == "ok" for _, msg in results), results
