@@ -22,9 +22,19 @@
 // With S=1 the tile is written directly (no atomics).
 //
 // Build/run (GPU box):
-//   hipcc --offload-arch=gfx950 -O3 -std=c++17 tools/wgrad_gemm.hip \
-//       -o /tmp/wg && /tmp/wg 4608 1536 65536 8 50
+//   hipcc --offload-arch=gfx950 -O3 -std=c++17 [-DVARIANT=1] \
+//       tools/wgrad_gemm.hip -o /tmp/wg && /tmp/wg 4608 1536 65536 8 50
 //   (args: M N K splits iters; prints TF/s + spot-check vs CPU dots)
+//
+// VARIANT 0: synchronous staging (stage -> barrier -> mfma -> barrier)
+// VARIANT 1: double-buffered staging with early loads (the validated
+//            attention_fwd T14 split: global loads for chunk t+1 issue
+//            before chunk t's MFMAs, LDS writes go to the alternate
+//            buffer after them — one barrier per chunk)
+
+#ifndef VARIANT
+#define VARIANT 0
+#endif
 
 #include <hip/hip_runtime.h>
 #include <cstdio>
@@ -43,27 +53,47 @@ __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return (byte_in_row ^ ((row & 7) << 4));
 }
 
-// stage a (BK x BM) k-major global slab into an [m][k] bf16 LDS image
-// (transpose-in-staging: contiguous global reads along m, scattered
-// column writes — the attention_bwd kt_lds pattern)
-__device__ __forceinline__ void stage_T(const short* __restrict__ src,
-                                        long long ld,  // row stride (elems)
-                                        long long k0, int m0, int mspan,
-                                        char* dst) {
-  // mspan * BK elements, bf16x8 per thread-pass along m
-  const int passes = (mspan * BK) / (BLOCK * 8);
+// staging of a (BK x BM) k-major global slab into an [m][k] bf16 LDS
+// image (transpose-in-staging: contiguous global reads along m,
+// scattered column writes — the attention_bwd kt_lds pattern), split
+// into load-to-registers / write-to-LDS halves so VARIANT 1 can issue
+// the loads a chunk early (T14)
+#define STAGE_PASSES ((BM * BK) / (BLOCK * 8))
+
+__device__ __forceinline__ void stage_load(const short* __restrict__ src,
+                                           long long ld, long long k0,
+                                           int m0, int mspan,
+                                           bf16x8* regs) {
 #pragma unroll
-  for (int p = 0; p < passes; ++p) {
+  for (int p = 0; p < STAGE_PASSES; ++p) {
     const int idx = (p * BLOCK + (int)threadIdx.x) * 8;
-    const int k = idx / mspan;          // 0..BK
-    const int m = idx % mspan;          // multiple of 8
-    bf16x8 v = *(const bf16x8*)(src + (k0 + k) * ld + m0 + m);
+    const int k = idx / mspan;
+    const int m = idx % mspan;
+    regs[p] = *(const bf16x8*)(src + (k0 + k) * ld + m0 + m);
+  }
+}
+
+__device__ __forceinline__ void stage_write(const bf16x8* regs, int mspan,
+                                            char* dst) {
+#pragma unroll
+  for (int p = 0; p < STAGE_PASSES; ++p) {
+    const int idx = (p * BLOCK + (int)threadIdx.x) * 8;
+    const int k = idx / mspan;
+    const int m = idx % mspan;
 #pragma unroll
     for (int j = 0; j < 8; ++j) {
       const int mm = m + j;
-      *(short*)(dst + mm * (BK * 2) + swz(mm, k * 2)) = ((short*)&v)[j];
+      *(short*)(dst + mm * (BK * 2) + swz(mm, k * 2)) = ((short*)&regs[p])[j];
     }
   }
+}
+
+__device__ __forceinline__ void stage_T(const short* __restrict__ src,
+                                        long long ld, long long k0, int m0,
+                                        int mspan, char* dst) {
+  bf16x8 regs[STAGE_PASSES];
+  stage_load(src, ld, k0, m0, mspan, regs);
+  stage_write(regs, mspan, dst);
 }
 
 __global__ __launch_bounds__(BLOCK) void wgrad_kernel(
@@ -92,8 +122,10 @@ __global__ __launch_bounds__(BLOCK) void wgrad_kernel(
   const int wn = (wid % 2) * 64;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* a_lds = smem;                       // [BM][BK] bf16, 16 KiB
-  char* b_lds = smem + BM * BK * 2;         // [BN][BK] bf16, 16 KiB
+#if VARIANT == 1
+  // double-buffered: two [BM][BK]+[BN][BK] sets, 64 KiB total
+  const long long nsteps = (k_hi - k_lo) / BK;
+#endif
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -101,10 +133,35 @@ __global__ __launch_bounds__(BLOCK) void wgrad_kernel(
 #pragma unroll
     for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
+#if VARIANT == 1
+  bf16x8 aregs[STAGE_PASSES], bregs[STAGE_PASSES];
+  {
+    // prologue: stage chunk 0 into buffer 0
+    char* a0 = smem;
+    char* b0 = smem + BM * BK * 2;
+    stage_load(dy, M, k_lo, tm * BM, BM, aregs);
+    stage_write(aregs, BM, a0);
+    stage_load(x, N, k_lo, tn * BN, BN, bregs);
+    stage_write(bregs, BN, b0);
+  }
+  __syncthreads();
+  for (long long t = 0; t < nsteps; ++t) {
+    const long long k0 = k_lo + t * BK;
+    char* base = smem + (t & 1) * (BM + BN) * BK * 2;
+    char* a_lds = base;
+    char* b_lds = base + BM * BK * 2;
+    if (t + 1 < nsteps) {  // T14: issue next chunk's loads before MFMAs
+      stage_load(dy, M, k0 + BK, tm * BM, BM, aregs);
+      stage_load(x, N, k0 + BK, tn * BN, BN, bregs);
+    }
+#else
+  char* a_lds = smem;
+  char* b_lds = smem + BM * BK * 2;
   for (long long k0 = k_lo; k0 < k_hi; k0 += BK) {
     stage_T(dy, M, k0, tm * BM, BM, a_lds);
     stage_T(x, N, k0, tn * BN, BN, b_lds);
     __syncthreads();
+#endif
 
     __builtin_amdgcn_s_setprio(1);
 #pragma unroll
@@ -131,8 +188,18 @@ __global__ __launch_bounds__(BLOCK) void wgrad_kernel(
               af[i], bf[j], acc[i][j], 0, 0, 0);
     }
     __builtin_amdgcn_s_setprio(0);
+#if VARIANT == 1
+    if (t + 1 < nsteps) {  // write next chunk into the alternate buffer
+      char* nbase = smem + ((t + 1) & 1) * (BM + BN) * BK * 2;
+      stage_write(aregs, BM, nbase);
+      stage_write(bregs, BN, nbase + BM * BK * 2);
+    }
     __syncthreads();
   }
+#else
+    __syncthreads();
+  }
+#endif
 
   // epilogue: C layout — col = lane&15, row = (lane>>4)*4 + reg
 #pragma unroll
@@ -187,7 +254,11 @@ int main(int argc, char** argv) {
   hipMemcpy(db, hb.data(), (size_t)K * N * 2, hipMemcpyHostToDevice);
 
   dim3 grid((M / BM) * (N / BN), 1, S), block(BLOCK);
+#if VARIANT == 1
+  size_t lds = 2 * (size_t)(BM + BN) * BK * 2;
+#else
   size_t lds = (size_t)(BM + BN) * BK * 2;
+#endif
 
   auto run = [&]() {
     if (S > 1) hipMemsetAsync(dacc, 0, (size_t)M * N * 4);
@@ -211,7 +282,8 @@ int main(int argc, char** argv) {
   hipEventElapsedTime(&ms, e0, e1);
   double us = ms * 1000.0 / iters;
   double tf = 2.0 * M * N * (double)K / (us * 1e-6) / 1e12;
-  printf("wgrad %dx%dx%lld S=%d: %.1f us  %.1f TF/s\n", M, N, K, S, us, tf);
+  printf("VARIANT %d wgrad %dx%dx%lld S=%d: %.1f us  %.1f TF/s\n",
+         VARIANT, M, N, K, S, us, tf);
 
   // spot-check ~64 random outputs against fp32 CPU dots over K
   std::vector<float> got((size_t)M * N);
