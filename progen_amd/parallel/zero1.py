@@ -82,3 +82,31 @@ class Zero1AdamW(ProGenAdamW):
         d["zero1"] = {"world": self.world, "rank": self.rank,
                       "lo": self.lo, "hi": self.hi}
         return d
+
+    def load_state_dict(self, sd) -> None:
+        z = sd.get("zero1")
+        if z is None:
+            raise ValueError(
+                "resume with PROGEN_ZERO1=1 needs a ZeRO-1 checkpoint "
+                "(optimizer state sharded per rank); this checkpoint "
+                "holds replicated optimizer state")
+        if (z["world"], z["rank"]) != (self.world, self.rank):
+            raise ValueError(
+                f"ZeRO-1 checkpoint is sharded for world={z['world']} "
+                f"rank={z['rank']}; this run is world={self.world} "
+                f"rank={self.rank}")
+        self.step_count = int(sd["step_count"])
+        if self.step_dev is not None:
+            self.step_dev.fill_(self.step_count)
+        self._micro = int(sd.get("micro", 0))
+        with torch.no_grad():
+            dev = self.master.device
+            self.master.copy_(torch.as_tensor(sd["master"]).to(dev))
+            self.exp_avg.copy_(torch.as_tensor(sd["exp_avg"]).to(dev))
+            self.exp_avg_sq.copy_(torch.as_tensor(sd["exp_avg_sq"]).to(dev))
+            if self.is_low_precision:
+                # the fp32 master is authoritative for THIS rank's slice;
+                # other slices were already loaded from the checkpoint's
+                # params (identical values after the bf16 rounding)
+                self.space.flat[self.lo:self.hi].copy_(
+                    self.master.to(self.space.flat.dtype))

@@ -172,3 +172,60 @@ def test_zero1_odd_numel_padding():
     for p in ps:
         p.join(timeout=60)
     assert all(msg == "ok" for _, msg in results), results
+
+
+def _resume_worker(rank, world, port, q):
+    import torch.distributed as dist
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from progen_amd.models.progen import ProGenBase
+    from progen_amd.parallel.zero1 import Zero1AdamW
+    from progen_amd.utils import compute_loss
+    try:
+        torch.manual_seed(41)
+        model = ProGenBase(_cfg()).double()
+        opt = Zero1AdamW(model, lr=1e-3)
+        data = _batches(1)[0]
+        loss = compute_loss(model, data[rank * 2:(rank + 1) * 2])
+        loss.backward()
+        opt.step()
+        opt.zero_grad()
+
+        sd = {k: (v.clone() if torch.is_tensor(v) else v)
+              for k, v in opt.state_dict().items()}
+        opt2 = Zero1AdamW(model, lr=1e-3)
+        opt2.load_state_dict(sd)
+        assert opt2.step_count == 1
+        torch.testing.assert_close(opt2.master, opt.master)
+
+        # replicated (non-ZeRO) state must be rejected with a clear error
+        bad = dict(sd)
+        bad.pop("zero1")
+        try:
+            opt2.load_state_dict(bad)
+            q.put((rank, "fail: no error raised"))
+            return
+        except ValueError as e:
+            assert "ZeRO-1" in str(e)
+        q.put((rank, "ok"))
+    except Exception as e:
+        import traceback
+        q.put((rank, f"fail: {e}\n{traceback.format_exc()[-1000:]}"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_zero1_state_roundtrip_and_guard():
+    world = 2
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_resume_worker, args=(r, world, port, q))
+          for r in range(world)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=90) for _ in range(world)]
+    for p in ps:
+        p.join(timeout=60)
+    assert all(msg == "ok" for _, msg in results), results
