@@ -31,6 +31,22 @@
 //            attention_fwd T14 split: global loads for chunk t+1 issue
 //            before chunk t's MFMAs, LDS writes go to the alternate
 //            buffer after them — one barrier per chunk)
+//
+// Round-2 upgrade path (derivation done, needs on-GPU iteration):
+//   glds (`global_load_lds` 16B) CANNOT build the transposed [m][k]
+//   image — its LDS destination is wave-uniform base + lane*16 and the
+//   transposed slot order would need an 8-element k-strided global
+//   GATHER per lane (one address per lane loads 16 CONTIGUOUS bytes).
+//   The guide's recipe for k-major operands is therefore: stage the
+//   slab LINEARLY with glds as a blocked [k/8][m/16][8][16] image
+//   (each 8x16 sub-tile contiguous, 256 B) and read fragments with
+//   `ds_read_b64_tr_b16` (lane l, elem j reads element
+//   (l&15) + j*16 + (l>>4)*64 past the sub-tile base: per 16-lane
+//   group a transposed 4x16 strip; two tr reads build one 8-deep k
+//   fragment). That replaces this file's scatter-write staging (the
+//   ds_write pass and its VGPRs) with async copies — the guide measured
+//   +67% from 16B glds alone on the 128^2 GEMM ladder. Verify with the
+//   spot-check harness below before trusting any tr mapping.
 
 #ifndef VARIANT
 #define VARIANT 0
