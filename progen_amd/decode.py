@@ -62,8 +62,14 @@ class DecodeCache:
         N, H, DH = cfg.seq_len, cfg.heads, cfg.dim_head
         self.k = []    # per attn layer: (B, H, N, DH) rotated keys
         self.v = []
-        self.attn_prev = [None] * cfg.depth   # prev LN row (attn branch)
-        self.ff_prev = [None] * cfg.depth     # prev LN row (ff branch)
+        # previous position's LN row per branch (token-shift halo);
+        # preallocated + written in place so a captured static step
+        # updates fixed buffers (zeros == the position-0 "no previous
+        # row" semantics)
+        self.attn_prev = [torch.zeros(batch, cfg.dim, device=device,
+                                      dtype=dtype) for _ in range(cfg.depth)]
+        self.ff_prev = [torch.zeros(batch, cfg.dim, device=device,
+                                    dtype=dtype) for _ in range(cfg.depth)]
         self.gate_hist = []  # per SGU layer: (B, N, d2) LN'd gate rows
         for _attn, ff in model.layers:
             self.k.append(torch.zeros(batch, H, N, DH, device=device, dtype=dtype))
@@ -82,7 +88,7 @@ def _attn_step(attn, x, sin_p, cos_p, cache: DecodeCache, li: int) -> torch.Tens
     H, DH, wsz = attn.heads, cache.cfg.dim_head, attn.window_size
     y = _ln_row(x, attn.norm_weight)
     y_in = _shift_row(y, cache.attn_prev[li]) if attn.shift_tokens else y
-    cache.attn_prev[li] = y
+    cache.attn_prev[li].copy_(y)
     qkv = F.linear(y_in, attn.to_qkv.weight)            # (B, 3*H*DH)
     B = qkv.shape[0]
     q, k, v = qkv.view(B, 3, H, DH).unbind(1)
@@ -113,7 +119,7 @@ def _ff_step(ff, x, cache: DecodeCache, li: int) -> torch.Tensor:
     """One FeedForward row at position cache.pos. x: (B, dim)."""
     y = _ln_row(x, ff.norm_weight)
     y_in = _shift_row(y, cache.ff_prev[li]) if ff.shift_tokens else y
-    cache.ff_prev[li] = y
+    cache.ff_prev[li].copy_(y)
     h = F.linear(y_in, ff.proj_in.weight, ff.proj_in.bias)
     if ff.glu:
         a, g = h.chunk(2, dim=-1)
@@ -155,6 +161,104 @@ def forward_step(model: ProGenBase, token: torch.Tensor,
     logits = F.linear(h, model.to_logits.weight, model.to_logits.bias)
     cache.pos = p + 1
     return logits
+
+
+@torch.no_grad()
+def forward_step_static(model: ProGenBase, token: torch.Tensor,
+                        cache: DecodeCache,
+                        pos_dev: torch.Tensor) -> torch.Tensor:
+    """Position-STATIC decode step: numerically identical to
+    ``forward_step`` but every operation has position-independent shapes
+    and reads the position from a device tensor — the form a hipGraph
+    can capture once and replay per token (TODO.md "Serving"; the
+    pure-replay rule of profiles/r01_graph_interleave_bug.md means the
+    position increment must itself be in-graph, which this function's
+    tensor-only indexing permits).
+
+    Mechanics per layer:
+      - attention gathers a FIXED 2*window_size key slice at indices
+        win_start - wsz + [0, 2wsz); indices < 0 fetch zeroed k/v and
+        stay UNMASKED (exactly the reference's window-0 zero-pad quirk,
+        progen.py:90-96); indices > pos are masked -inf;
+      - the SGU row uses the full spatial_weights row masked by
+        arange(N) <= pos (static O(N*d2) dot);
+      - rotary rows, shift halos and cache writes are index_put/gather
+        with tensor indices.
+
+    ``pos_dev``: 0-dim int64 tensor holding the current position; the
+    caller increments it AFTER the step (in-graph when captured).
+    Updates the same ``DecodeCache`` as ``forward_step``."""
+    cfg = model.cfg
+    wsz = cfg.window_size
+    dev = token.device
+    p = pos_dev  # 0-dim int64 on device
+    sin_p = model.rotary_sin.index_select(0, p.reshape(1))[0]
+    cos_p = model.rotary_cos.index_select(0, p.reshape(1))[0]
+    h = model.embed(token.long().reshape(-1))
+
+    win_start = (p // wsz) * wsz
+    gather_idx = win_start - wsz + torch.arange(2 * wsz, device=dev)
+    valid = (gather_idx >= 0)                       # zero-halo quirk keys
+    causal = (gather_idx <= p)                      # future keys masked
+    safe_idx = gather_idx.clamp(min=0)
+
+    for li, (attn, ff) in enumerate(model.layers):
+        # ---- attention branch ----
+        x = h
+        y = _ln_row(x, attn.norm_weight)
+        y_in = _shift_row(y, cache.attn_prev[li]) if attn.shift_tokens else y
+        cache.attn_prev[li].copy_(y)
+        qkv = F.linear(y_in, attn.to_qkv.weight)
+        B = qkv.shape[0]
+        H, DH = attn.heads, cfg.dim_head
+        q, k, v = qkv.view(B, 3, H, DH).unbind(1)
+        sc, ss = cos_p.to(q.dtype), sin_p.to(q.dtype)
+        q, k, v = (t * sc + R.rotate_every_two(t) * ss for t in (q, k, v))
+        cache.k[li].index_copy_(2, p.reshape(1), k.unsqueeze(2))
+        cache.v[li].index_copy_(2, p.reshape(1), v.unsqueeze(2))
+        keys = cache.k[li].index_select(2, safe_idx)    # (B, H, 2wsz, DH)
+        vals = cache.v[li].index_select(2, safe_idx)
+        kmask = valid.view(1, 1, -1, 1).to(keys.dtype)
+        keys = keys * kmask                             # zero halo keys
+        vals = vals * kmask
+        s_row = torch.einsum("bhd,bhnd->bhn", q, keys) * (DH ** -0.5)
+        s_row = torch.where(causal.view(1, 1, -1), s_row,
+                            torch.tensor(-1e30, dtype=s_row.dtype, device=dev))
+        s_row = s_row - s_row.amax(dim=-1, keepdim=True)
+        a = s_row.softmax(dim=-1)
+        out = torch.einsum("bhn,bhnd->bhd", a, vals).reshape(B, H * DH)
+        h = h + F.linear(out, attn.to_out.weight, attn.to_out.bias)
+
+        # ---- ff branch ----
+        y = _ln_row(h, ff.norm_weight)
+        y_in = _shift_row(y, cache.ff_prev[li]) if ff.shift_tokens else y
+        cache.ff_prev[li].copy_(y)
+        t = F.linear(y_in, ff.proj_in.weight, ff.proj_in.bias)
+        if ff.glu:
+            a2, g2 = t.chunk(2, dim=-1)
+            t = a2 * F.gelu(g2, approximate="tanh")
+        else:
+            t = F.gelu(t, approximate="tanh")
+        if ff.sgu is not None:
+            sgu = ff.sgu
+            xa, gate = t.chunk(2, dim=-1)
+            gate_ln = _ln_row(gate, sgu.norm_weight)
+            hist = cache.gate_hist[li]
+            hist.index_copy_(1, p.reshape(1), gate_ln.unsqueeze(1))
+            n = hist.shape[1]
+            w_row = sgu.spatial_weights.index_select(0, p.reshape(1))[0, :n]
+            past = (torch.arange(n, device=dev) <= p).to(hist.dtype)
+            gate_out = torch.einsum("n,bnd->bd", w_row.to(hist.dtype) * past,
+                                    hist) + \
+                sgu.spatial_biases.index_select(0, p.reshape(1))[0]
+            t = xa * gate_out
+            t = F.linear(t, sgu.proj_out.weight, sgu.proj_out.bias)
+        h = h + F.linear(t, ff.proj_out.weight, ff.proj_out.bias)
+
+    y = _ln_row(h, model.final_norm_weight)
+    return F.linear(y, model.to_logits.weight, model.to_logits.bias)
+    # NOTE: cache.pos is NOT advanced — the caller owns pos_dev and
+    # increments it (in-graph when captured)
 
 
 @torch.no_grad()

@@ -110,3 +110,23 @@ def test_sample_cached_batch_greedy_parity():
         z = (seq == 0).long().cumsum(-1) > 1
         seq = seq * (~z).long()
         assert torch.equal(out[i], seq)
+
+
+def test_forward_step_static_matches_dynamic():
+    """The position-static step (fixed shapes, tensor indexing — the
+    hipGraph-capturable form) must be numerically identical to
+    forward_step, including the window-0 zero-halo quirk and window
+    boundaries."""
+    from progen_amd.decode import forward_step_static
+    model = _tiny(global_mlp_depth=2)
+    seq = torch.randint(1, 256, (48,))
+
+    cache_d = decode.DecodeCache(model, batch=1)
+    cache_s = decode.DecodeCache(model, batch=1)
+    for p in range(48):
+        ref = decode.forward_step(model, seq[p:p + 1], cache_d)
+        pos = torch.tensor(p, dtype=torch.long)
+        got = forward_step_static(model, seq[p:p + 1], cache_s, pos)
+        # fixed-band vs variable-band einsum: same math, different fp32
+        # reduction order -> ulp-level noise only
+        torch.testing.assert_close(got, ref, rtol=1e-5, atol=1e-5)
