@@ -66,9 +66,16 @@ class DistributedTrainer:
     divides by world size.
     """
 
-    def __init__(self, space: FlatParamSpace, bucket_mb: Optional[float] = None):
+    def __init__(self, space: FlatParamSpace, bucket_mb: Optional[float] = None,
+                 group: Optional[dist.ProcessGroup] = None):
+        """``group``: the data-parallel process group to reduce over
+        (default: WORLD). Under a TP x DP mesh pass the DP group
+        (parallel/tp.py::dp_group) so gradients average over replicas
+        only — TP-sharded slices are rank-local and must not mix across
+        the TP group."""
         self.space = space
-        self.world = dist.get_world_size() if is_distributed() else 1
+        self.group = group
+        self.world = dist.get_world_size(group) if is_distributed() else 1
         self._sync = True
         self._works: List[dist.Work] = []
         if bucket_mb is None:
@@ -114,7 +121,8 @@ class DistributedTrainer:
             if self._pending[b] == 0:
                 start, end, _ = self.buckets[b]
                 work = dist.all_reduce(self.space.flat_grad[start:end],
-                                       op=dist.ReduceOp.SUM, async_op=True)
+                                       op=dist.ReduceOp.SUM, async_op=True,
+                                       group=self.group)
                 self._works.append(work)
         return hook
 
@@ -143,7 +151,8 @@ class DistributedTrainer:
                     start, end, _ = self.buckets[b]
                     self._works.append(
                         dist.all_reduce(self.space.flat_grad[start:end],
-                                        op=dist.ReduceOp.SUM, async_op=True))
+                                        op=dist.ReduceOp.SUM, async_op=True,
+                                        group=self.group))
         for w in self._works:
             w.wait()
         self._works.clear()
@@ -156,6 +165,6 @@ class DistributedTrainer:
         host-side masked mean over devices, utils.py:90-91)."""
         if self.world > 1:
             t = t.detach().clone()
-            dist.all_reduce(t, op=dist.ReduceOp.SUM)
+            dist.all_reduce(t, op=dist.ReduceOp.SUM, group=self.group)
             t /= self.world
         return t

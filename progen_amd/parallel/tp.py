@@ -36,13 +36,14 @@ import torch.nn.functional as F
 from ..models.progen import _haiku_linear_init_
 
 _TP_GROUP: Optional[dist.ProcessGroup] = None
+_DP_GROUP: Optional[dist.ProcessGroup] = None
 
 
 def init_tensor_parallel(tp_size: int) -> None:
     """Split WORLD into contiguous TP groups of ``tp_size`` ranks (ranks
     [0..tp-1], [tp..2tp-1], ... — contiguous ranks share a node's xGMI
     mesh under torchrun's rank assignment)."""
-    global _TP_GROUP
+    global _TP_GROUP, _DP_GROUP
     world = dist.get_world_size()
     assert world % tp_size == 0, (world, tp_size)
     rank = dist.get_rank()
@@ -51,10 +52,24 @@ def init_tensor_parallel(tp_size: int) -> None:
         group = dist.new_group(ranks)
         if rank in ranks:
             _TP_GROUP = group
+    # the orthogonal DP axis: ranks sharing a TP position across
+    # replicas ({r, r+tp, r+2tp, ...}) — every rank must create every
+    # group (new_group is collective)
+    for pos in range(tp_size):
+        ranks = list(range(pos, world, tp_size))
+        group = dist.new_group(ranks)
+        if rank in ranks:
+            _DP_GROUP = group
 
 
 def tp_group() -> Optional[dist.ProcessGroup]:
     return _TP_GROUP
+
+
+def dp_group() -> Optional[dist.ProcessGroup]:
+    """The orthogonal data-parallel group of a TP x DP mesh (ranks with
+    the same TP position across replicas)."""
+    return _DP_GROUP
 
 
 def tp_size() -> int:
