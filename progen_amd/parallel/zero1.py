@@ -26,6 +26,8 @@ from __future__ import annotations
 import torch
 import torch.distributed as dist
 
+from typing import Optional
+
 from ..optim import ProGenAdamW
 
 
@@ -36,12 +38,18 @@ class Zero1AdamW(ProGenAdamW):
     all-reduced (mean) before ``step`` — exactly what DistributedTrainer
     already does."""
 
-    def __init__(self, module: torch.nn.Module, **kwargs):
+    def __init__(self, module: torch.nn.Module,
+                 group: Optional["dist.ProcessGroup"] = None, **kwargs):
+        """``group``: the data-parallel group to shard over (default:
+        WORLD). Under a TP x DP mesh pass parallel/tp.py::dp_group() —
+        the optimizer state shards across REPLICAS; TP shards are
+        already rank-local."""
         super().__init__(module, **kwargs)
-        assert dist.is_initialized() and dist.get_world_size() > 1, \
+        assert dist.is_initialized() and dist.get_world_size(group) > 1, \
             "Zero1AdamW requires an initialized process group (world > 1)"
-        self.world = dist.get_world_size()
-        self.rank = dist.get_rank()
+        self.group = group
+        self.world = dist.get_world_size(group)
+        self.rank = dist.get_rank(group)
         n = self.space.numel
         self.shard = -(-n // self.world)          # padded shard size
         self.lo = min(self.rank * self.shard, n)
@@ -65,7 +73,7 @@ class Zero1AdamW(ProGenAdamW):
         buf[: self.hi - self.lo] = self.master.to(flat.dtype)
         out = torch.empty(self.shard * self.world, dtype=flat.dtype,
                           device=flat.device)
-        dist.all_gather_into_tensor(out, buf)
+        dist.all_gather_into_tensor(out, buf, group=self.group)
         flat.copy_(out[: flat.numel()])
 
     def _step_hip(self, grad_scale: float) -> None:

@@ -50,7 +50,7 @@ def _losses_single(steps=3):
     return losses
 
 
-def _worker(rank, world, port, q, steps=3):
+def _worker(rank, world, port, q, steps=3, zero1=False):
     import torch.distributed as dist
     os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
     dist.init_process_group("gloo", rank=rank, world_size=world)
@@ -65,7 +65,12 @@ def _worker(rank, world, port, q, steps=3):
         dp_rank = rank // TP  # contiguous TP groups -> replica index
         torch.manual_seed(51)  # identical full init everywhere
         model = tp_model.tp_shard_(ProGenBase(_cfg()).double())
-        optim = ProGenAdamW(model, lr=1e-3)
+        if zero1:
+            from progen_amd.parallel.zero1 import Zero1AdamW
+            optim = Zero1AdamW(model, lr=1e-3, group=tp.dp_group())
+            assert optim.world == 2  # shards over replicas only
+        else:
+            optim = ProGenAdamW(model, lr=1e-3)
         optim.norm_sumsq_fn = tp_model.tp_grad_sumsq_fn(model)
         ddp = DistributedTrainer(optim.space, group=tp.dp_group())
         assert ddp.world == 2  # reduces over replicas, not the world
@@ -88,7 +93,10 @@ def _worker(rank, world, port, q, steps=3):
 
 
 @pytest.mark.timeout(240)
-def test_mesh_tp2_dp2_matches_single():
+@pytest.mark.parametrize("zero1", [False, True])
+def test_mesh_tp2_dp2_matches_single(zero1):
+    # zero1=True additionally shards the optimizer state over the DP
+    # axis of the mesh (TP=2 x DP=2 x ZeRO-1)
     want = _losses_single()
     assert want[0] != want[-1]
 
@@ -96,7 +104,7 @@ def test_mesh_tp2_dp2_matches_single():
     port = _free_port()
     ctx = mp.get_context("spawn")
     q = ctx.Queue()
-    ps = [ctx.Process(target=_worker, args=(r, world, port, q))
+    ps = [ctx.Process(target=_worker, args=(r, world, port, q, 3, zero1))
           for r in range(world)]
     for p in ps:
         p.start()
