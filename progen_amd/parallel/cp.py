@@ -37,10 +37,11 @@ from ..models.progen import ProGenBase
 from ..ops import reference as R
 
 _CP_GROUP: Optional[dist.ProcessGroup] = None
+_DP_GROUP: Optional[dist.ProcessGroup] = None
 
 
 def init_context_parallel(cp_size: int) -> None:
-    global _CP_GROUP
+    global _CP_GROUP, _DP_GROUP
     world = dist.get_world_size()
     assert world % cp_size == 0
     rank = dist.get_rank()
@@ -49,6 +50,18 @@ def init_context_parallel(cp_size: int) -> None:
         group = dist.new_group(ranks)
         if rank in ranks:
             _CP_GROUP = group
+    # orthogonal DP axis (ranks at the same sequence position across
+    # replicas); every rank must create every group (collective)
+    for pos in range(cp_size):
+        ranks = list(range(pos, world, cp_size))
+        group = dist.new_group(ranks)
+        if rank in ranks:
+            _DP_GROUP = group
+
+
+def dp_group() -> Optional[dist.ProcessGroup]:
+    """The orthogonal data-parallel group of a CP x DP mesh."""
+    return _DP_GROUP
 
 
 def cp_size() -> int:
