@@ -19,13 +19,18 @@
 // 64-row q-chunk. Per 64-key tile the per-row phases (S, P, dS, dQ) are
 // wave-local; the per-KEY phases (dV, dK) are computed as per-wave
 // 16-key OUTPUT SLICES whose MFMA K-dim spans ALL chunks' P/dS/dO^T/Q^T
-// LDS regions — so every dV/dK element is produced by exactly ONE wave
-// and hits global memory with ONE atomicAdd per overlapping window
-// (adjacent windows share keys through the lookback, progen.py:90-91;
-// the naive per-wave accumulation was 4x more atomic traffic and was the
-// top kernel in the step profile). dQ rows are exclusively owned ->
-// plain fp32 stores. Window 0's lookback keys are the zero pad; their
-// gradients are discarded.
+// LDS regions — every dV/dK element is produced by exactly ONE wave, so
+// the stores are PLAIN (no atomics): each window's own-band gradients go
+// to dacc, its lookback-band gradients to a separate dlook buffer, and
+// attn_bwd_finalize_kernel sums the two (adjacent windows share keys
+// through the lookback, progen.py:90-91 — writing both bands to one
+// buffer would need atomics across blocks; the earlier atomic versions
+// were 4x/1x the traffic and the plain-store split measured fastest;
+// cross-round accumulation within a block is a read-modify-write, which
+// tools/ablate_attn_bwd.hip showed is latency-hidden at the production
+// grid depth). dQ rows are exclusively owned -> plain fp32 stores.
+// Window 0's lookback keys are the zero pad; their gradients are
+// discarded.
 //
 // MFMA operand LDS images (XOR-swizzled, byte ^= (row&7)<<4):
 //   k_lds  [key][dh]   k'           (S B-fragments)
