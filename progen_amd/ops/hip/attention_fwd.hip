@@ -15,8 +15,9 @@
 //   - fp32 online softmax (max-subtract parity with progen.py:98-99);
 //   - P·V accumulation and the '(w n) (h d)' output merge (progen.py:102).
 //
-// Geometry (CDNA4): block = 4 waves = one window; each wave owns a
-// 64-row Q chunk (round-robins chunks when wsz > 256). Per 64-key tile:
+// Geometry (CDNA4): block = 4 waves; each wave owns a 32-row Q chunk
+// (MF=2 m-fragments; 193 VGPRs -> 2 waves/SIMD), sub_per_win blocks
+// cover a window and round-robin chunks when wsz > 128. Per 64-key tile:
 //   - QK^T is computed SWAPPED — S^T = mfma(K, Q) — so the MFMA C-layout
 //     holds 4 consecutive KEYS of one q-row per register quad: the P
 //     tile is written to LDS [row][key] with ds_write_b64 (the
@@ -24,9 +25,10 @@
 //     softmax reduce is 16 in-lane values + a 2-step cross-lane shuffle;
 //   - K staged in LDS [key][dh] and V transposed [dh][key], both
 //     XOR-swizzled (byte ^= (row&7)<<4) -> <=2-way bank conflicts on the
-//     ds_read_b128 fragments; staging uses the async-STAGE split (T14):
-//     tile t+1's global loads issue before tile t's compute, the LDS
-//     writes land after the barrier — hiding HBM latency at 1 wave/SIMD;
+//     ds_read_b128 fragments; staging is DOUBLE-BUFFERED with the
+//     async-stage split (T14): tile t+1's global loads issue before
+//     tile t's MFMAs and the LDS writes land in the alternate buffer
+//     after them — ONE barrier per tile;
 //   - per-row softmax state (m, l) lives in the 4 lanes of the row's
 //     shuffle group; the O rescale factor crosses to the PV C-layout
 //     rows through a tiny per-wave LDS broadcast array.
