@@ -47,6 +47,10 @@ class Zero1AdamW(ProGenAdamW):
         super().__init__(module, **kwargs)
         assert dist.is_initialized() and dist.get_world_size(group) > 1, \
             "Zero1AdamW requires an initialized process group (world > 1)"
+        assert self.accum_mode == "sum", \
+            "Zero1AdamW supports accum_mode='sum' only (apply_every " \
+            "advances moments from rank-local grads; with sharded state " \
+            "the shapes don't even line up — use the replicated optimizer)"
         self.group = group
         self.world = dist.get_world_size(group)
         self.rank = dist.get_rank(group)
@@ -85,36 +89,54 @@ class Zero1AdamW(ProGenAdamW):
     def resync_master(self) -> None:
         self.master.copy_(self.space.flat[self.lo:self.hi].float())
 
+    # Checkpoints use the REPLICATED format: state_dict() all-gathers
+    # the shards into full master/exp_avg/exp_avg_sq tensors, so a
+    # ZeRO-1 checkpoint resumes at any world size, with or without
+    # ZeRO-1 (the earlier per-rank format only saved rank
+    # 0's shard and could not resume at all). The gather is a
+    # COLLECTIVE: train.py must call state_dict() on every rank (see
+    # state_dict_is_collective).
+    state_dict_is_collective = True
+
+    def _gather_full(self, t: torch.Tensor) -> torch.Tensor:
+        buf = torch.zeros(self.shard, dtype=t.dtype, device=t.device)
+        buf[: self.hi - self.lo] = t
+        out = torch.empty(self.shard * self.world, dtype=t.dtype,
+                          device=t.device)
+        dist.all_gather_into_tensor(out, buf, group=self.group)
+        return out[: self.space.numel].clone()
+
     def state_dict(self):
-        d = super().state_dict()
-        d["zero1"] = {"world": self.world, "rank": self.rank,
-                      "lo": self.lo, "hi": self.hi}
-        return d
+        if self.step_dev is not None:
+            self.step_count = max(self.step_count, int(self.step_dev.item()))
+        return {
+            "step_count": self.step_count,
+            "micro": self._micro,
+            "master": self._gather_full(self.master),
+            "exp_avg": self._gather_full(self.exp_avg),
+            "exp_avg_sq": self._gather_full(self.exp_avg_sq),
+        }
 
     def load_state_dict(self, sd) -> None:
-        z = sd.get("zero1")
-        if z is None:
+        if "zero1" in sd:
             raise ValueError(
-                "resume with PROGEN_ZERO1=1 needs a ZeRO-1 checkpoint "
-                "(optimizer state sharded per rank); this checkpoint "
-                "holds replicated optimizer state")
-        if (z["world"], z["rank"]) != (self.world, self.rank):
-            raise ValueError(
-                f"ZeRO-1 checkpoint is sharded for world={z['world']} "
-                f"rank={z['rank']}; this run is world={self.world} "
-                f"rank={self.rank}")
+                "this checkpoint uses the removed per-rank ZeRO-1 shard "
+                "format (round 1); it only ever held rank 0's shard and "
+                "cannot be resumed — restart from a params-only load")
         self.step_count = int(sd["step_count"])
         if self.step_dev is not None:
             self.step_dev.fill_(self.step_count)
         self._micro = int(sd.get("micro", 0))
         with torch.no_grad():
             dev = self.master.device
-            self.master.copy_(torch.as_tensor(sd["master"]).to(dev))
-            self.exp_avg.copy_(torch.as_tensor(sd["exp_avg"]).to(dev))
-            self.exp_avg_sq.copy_(torch.as_tensor(sd["exp_avg_sq"]).to(dev))
+            sl = slice(self.lo, self.hi)
+            self.master.copy_(torch.as_tensor(sd["master"])[sl].to(dev))
+            self.exp_avg.copy_(torch.as_tensor(sd["exp_avg"])[sl].to(dev))
+            self.exp_avg_sq.copy_(
+                torch.as_tensor(sd["exp_avg_sq"])[sl].to(dev))
             if self.is_low_precision:
                 # the fp32 master is authoritative for THIS rank's slice;
                 # other slices were already loaded from the checkpoint's
                 # params (identical values after the bf16 rounding)
-                self.space.flat[self.lo:self.hi].copy_(
+                self.space.flat[sl].copy_(
                     self.master.to(self.space.flat.dtype))

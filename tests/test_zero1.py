@@ -191,16 +191,30 @@ def _resume_worker(rank, world, port, q):
         opt.step()
         opt.zero_grad()
 
+        # state_dict() is a COLLECTIVE all-gather into the replicated
+        # format (the old per-rank format only saved rank 0's
+        # shard and could never resume)
+        assert opt.state_dict_is_collective
         sd = {k: (v.clone() if torch.is_tensor(v) else v)
               for k, v in opt.state_dict().items()}
+        assert sd["master"].numel() == opt.space.numel  # full, not shard
         opt2 = Zero1AdamW(model, lr=1e-3)
         opt2.load_state_dict(sd)
         assert opt2.step_count == 1
         torch.testing.assert_close(opt2.master, opt.master)
+        torch.testing.assert_close(opt2.exp_avg, opt.exp_avg)
 
-        # replicated (non-ZeRO) state must be rejected with a clear error
+        # the replicated format also resumes WITHOUT ZeRO-1 (world-size
+        # independent): full master must match the gathered shards
+        from progen_amd.optim import ProGenAdamW
+        opt3 = ProGenAdamW(model, lr=1e-3)
+        opt3.load_state_dict(sd)
+        torch.testing.assert_close(
+            opt3.master[opt.lo:opt.hi], opt.master)
+
+        # the removed round-1 per-rank format is rejected with a clear error
         bad = dict(sd)
-        bad.pop("zero1")
+        bad["zero1"] = {"world": world, "rank": rank}
         try:
             opt2.load_state_dict(bad)
             q.put((rank, "fail: no error raised"))

@@ -57,6 +57,7 @@
 #include <cstdlib>
 #include <cmath>
 #include <vector>
+#include <cstring>
 #include "../progen_amd/ops/hip/common.h"
 
 #define BM 128
@@ -243,22 +244,26 @@ int main(int argc, char** argv) {
   int M = argc > 1 ? atoi(argv[1]) : 4608;
   int N = argc > 2 ? atoi(argv[2]) : 1536;
   long long K = argc > 3 ? atoll(argv[3]) : 65536;
-  int S = argc > 4 ? atoi(argv[4]) : 8;
+  const char* slist = argc > 4 ? argv[4] : "8";  // comma-separated splits
   int iters = argc > 5 ? atoi(argv[5]) : 50;
-  if (M % BM || N % BN || K % (long long)(BK * S)) {
-    printf("shape must divide: M%%%d N%%%d K%%(%d*S)\n", BM, BN, BK);
+  if (M % BM || N % BN || K % BK) {
+    printf("shape must divide: M%%%d N%%%d K%%%d\n", BM, BN, BK);
     return 1;
   }
 
   std::vector<short> ha((size_t)K * M), hb((size_t)K * N);
-  srand(3);
-  auto rb = []() {
-    float f = ((float)rand() / (float)RAND_MAX - 0.5f) * 2.0f;  // [-1,1)
-    union { float f; unsigned u; } c; c.f = f;
-    return (short)(c.u >> 16);
+  // fast xorshift fill (host rand() is too slow for GB-scale buffers)
+  auto fill = [](std::vector<short>& v, unsigned seed) {
+    unsigned s = seed;
+    for (auto& e : v) {
+      s ^= s << 13; s ^= s >> 17; s ^= s << 5;
+      float f = ((float)(s & 0xffffff) / 8388608.0f - 1.0f);  // [-1,1)
+      union { float f; unsigned u; } c; c.f = f;
+      e = (short)(c.u >> 16);
+    }
   };
-  for (auto& v : ha) v = rb();
-  for (auto& v : hb) v = rb();
+  fill(ha, 3u);
+  fill(hb, 77u);
 
   short *da, *db, *dout;
   float* dacc;
@@ -269,56 +274,66 @@ int main(int argc, char** argv) {
   hipMemcpy(da, ha.data(), (size_t)K * M * 2, hipMemcpyHostToDevice);
   hipMemcpy(db, hb.data(), (size_t)K * N * 2, hipMemcpyHostToDevice);
 
-  dim3 grid((M / BM) * (N / BN), 1, S), block(BLOCK);
 #if VARIANT == 1
   size_t lds = 2 * (size_t)(BM + BN) * BK * 2;
 #else
   size_t lds = (size_t)(BM + BN) * BK * 2;
 #endif
 
-  auto run = [&]() {
-    if (S > 1) hipMemsetAsync(dacc, 0, (size_t)M * N * 4);
-    wgrad_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
-    long long n = (long long)M * N;
-    cast_bf16_kernel<<<(int)((n + 255) / 256), 256>>>(dacc, dout, n);
-  };
+  int rc = 0;
+  char sbuf[256];
+  snprintf(sbuf, sizeof sbuf, "%s", slist);
+  for (char* tok = strtok(sbuf, ","); tok; tok = strtok(nullptr, ",")) {
+    int S = atoi(tok);
+    if (K % (long long)(BK * S)) { printf("skip S=%d (K)\n", S); continue; }
+    dim3 grid((M / BM) * (N / BN), 1, S), block(BLOCK);
 
-  for (int i = 0; i < 5; ++i) run();
-  hipDeviceSynchronize();
-  hipError_t err = hipGetLastError();
-  if (err != hipSuccess) { printf("HIP ERR %s\n", hipGetErrorString(err)); return 1; }
+    auto run = [&]() {
+      if (S > 1) hipMemsetAsync(dacc, 0, (size_t)M * N * 4);
+      wgrad_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
+      long long n = (long long)M * N;
+      cast_bf16_kernel<<<(int)((n + 255) / 256), 256>>>(dacc, dout, n);
+    };
 
-  hipEvent_t e0, e1;
-  hipEventCreate(&e0); hipEventCreate(&e1);
-  hipEventRecord(e0);
-  for (int i = 0; i < iters; ++i) run();
-  hipEventRecord(e1);
-  hipEventSynchronize(e1);
-  float ms;
-  hipEventElapsedTime(&ms, e0, e1);
-  double us = ms * 1000.0 / iters;
-  double tf = 2.0 * M * N * (double)K / (us * 1e-6) / 1e12;
-  printf("VARIANT %d wgrad %dx%dx%lld S=%d: %.1f us  %.1f TF/s\n",
-         VARIANT, M, N, K, S, us, tf);
+    for (int i = 0; i < 5; ++i) run();
+    hipDeviceSynchronize();
+    hipError_t err = hipGetLastError();
+    if (err != hipSuccess) { printf("HIP ERR %s\n", hipGetErrorString(err)); return 1; }
 
-  // spot-check ~64 random outputs against fp32 CPU dots over K
-  std::vector<float> got((size_t)M * N);
-  hipMemcpy(got.data(), dacc, (size_t)M * N * 4, hipMemcpyDeviceToHost);
-  auto b2f = [](short s) {
-    union { unsigned u; float f; } c; c.u = ((unsigned)(unsigned short)s) << 16;
-    return c.f;
-  };
-  double maxrel = 0;
-  for (int t = 0; t < 64; ++t) {
-    int m = rand() % M, n = rand() % N;
-    double ref = 0;
-    for (long long k = 0; k < K; ++k)
-      ref += (double)b2f(ha[(size_t)k * M + m]) * (double)b2f(hb[(size_t)k * N + n]);
-    double g = got[(size_t)m * N + n];
-    double rel = fabs(g - ref) / (fabs(ref) + 1e-3);
-    if (rel > maxrel) maxrel = rel;
+    hipEvent_t e0, e1;
+    hipEventCreate(&e0); hipEventCreate(&e1);
+    hipEventRecord(e0);
+    for (int i = 0; i < iters; ++i) run();
+    hipEventRecord(e1);
+    hipEventSynchronize(e1);
+    float ms;
+    hipEventElapsedTime(&ms, e0, e1);
+    double us = ms * 1000.0 / iters;
+    double tf = 2.0 * M * N * (double)K / (us * 1e-6) / 1e12;
+    printf("VARIANT %d wgrad %dx%dx%lld S=%d: %.1f us  %.1f TF/s\n",
+           VARIANT, M, N, K, S, us, tf);
+
+    // spot-check ~64 random outputs against fp32 CPU dots over K
+    std::vector<float> got((size_t)M * N);
+    hipMemcpy(got.data(), dacc, (size_t)M * N * 4, hipMemcpyDeviceToHost);
+    auto b2f = [](short s) {
+      union { unsigned u; float f; } c; c.u = ((unsigned)(unsigned short)s) << 16;
+      return c.f;
+    };
+    double maxrel = 0;
+    for (int t = 0; t < 64; ++t) {
+      int m = rand() % M, n = rand() % N;
+      double ref = 0;
+      for (long long k = 0; k < K; ++k)
+        ref += (double)b2f(ha[(size_t)k * M + m]) * (double)b2f(hb[(size_t)k * N + n]);
+      double g = got[(size_t)m * N + n];
+      double rel = fabs(g - ref) / (fabs(ref) + 1e-3);
+      if (rel > maxrel) maxrel = rel;
+    }
+    printf("spot-check max rel err (64 samples): %.3e %s\n", maxrel,
+           maxrel < 2e-2 ? "OK" : "FAIL");
+    if (maxrel >= 2e-2) rc = 1;
   }
-  printf("spot-check max rel err (64 samples): %.3e %s\n", maxrel,
-         maxrel < 2e-2 ? "OK" : "FAIL");
-  return maxrel < 2e-2 ? 0 : 1;
+  hipFree(da); hipFree(db); hipFree(dacc); hipFree(dout);
+  return rc;
 }

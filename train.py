@@ -77,7 +77,9 @@ def _wandb(wandb_off):
 @click.option('--graph/--no-graph', default=False,
               help='hipGraph-capture the training step (single-GPU, pure replay '
                    'only: disables in-loop validation/sampling — eager kernels '
-                   'between replays corrupt replay state on this ROCm stack)')
+                   'between replays corrupt replay state on this ROCm stack; '
+                   'checkpoints still happen: the graph is dropped and '
+                   're-captured around each save)')
 @click.option('--yes', default=False, is_flag=True, help='skip the --new confirmation prompt')
 def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
          data_parallel, max_grad_norm, validate_every, sample_every,
@@ -179,18 +181,23 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
 
     # hipGraph-captured step (progen_amd/runtime.py): replays the whole
     # fwd+bwd(+optimizer) sequence as one graph when shapes are static
-    graphed = None
-    if graph and device.type == 'cuda' and grad_accum_every == 1 \
-            and grad_accum_mode == 'sum' and world == 1:
+    graph_ok = graph and device.type == 'cuda' and grad_accum_every == 1 \
+        and grad_accum_mode == 'sum' and world == 1
+
+    def make_graphed():
         from progen_amd.runtime import GraphedTrainStep
         try:
-            graphed = GraphedTrainStep(module, optim, ddp, batch_size,
-                                       seq_len, device)
+            g = GraphedTrainStep(module, optim, ddp, batch_size,
+                                 seq_len, device)
             if is_main:
                 print('hipGraph training step captured')
+            return g
         except Exception as e:  # noqa: BLE001
             if is_main:
                 print(f'hipGraph capture failed ({e}); running eager')
+            return None
+
+    graphed = make_graphed() if graph_ok else None
 
     if is_main:
         print(f'params: {num_params}')
@@ -198,11 +205,16 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
         print(f'num sequences: {total_train_seqs}')
         print(f'starting from sequence {start_seq_index}')
 
-    def my_shard(batch_np):
+    def my_shard(batch_np, drop_ragged=True):
         t = torch.from_numpy(batch_np.astype(np.int64))
+        if drop_ragged and world > 1 and t.shape[0] < global_batch:
+            # tf.data drop_remainder parity: duplicating rows to fill the
+            # ragged tail would double-count them in the averaged
+            # gradient — end the epoch instead
+            raise StopIteration
         lo = rank * batch_size
         shard = t[lo:lo + batch_size]
-        if shard.shape[0] == 0:  # tail batch smaller than world*bs
+        if shard.shape[0] == 0:  # ragged validation batch: reuse row 0
             shard = t[:1]
         return shard.to(device)
 
@@ -224,8 +236,12 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
             if graphed is not None:
                 loss = graphed.run(data)
                 continue
+            # apply_every advances Adam moments EVERY micro-batch, so the
+            # grads it consumes must already be all-reduced each micro —
+            # the reference pmap reduces per micro-batch too (ADVICE r1:
+            # no_sync here would diverge the DP replicas)
             last_micro = micro == grad_accum_every - 1
-            if last_micro:
+            if last_micro or grad_accum_mode == 'apply_every':
                 loss = compute_loss(module, data)
                 loss.backward()
                 ddp.finish_backward()
@@ -249,18 +265,33 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
             if wandb is not None:
                 wandb.log({'loss': loss_val, 'tokens_per_sec': toks_per_sec})
 
-        if graphed is None and i % checkpoint_every == 0 and is_main:
-            package = {
-                'next_seq_index': seq_index + effective_batch_size,
-                'params': tensors_to_numpy(
-                    {k: v for k, v in module.state_dict().items()}),
-                'optim_state': tensors_to_numpy(optim.state_dict()),
-                'model_config': model_kwargs,
-                'run_id': run_id,
-            }
-            save_checkpoint(package, checkpoint_keep_n)
-            print(f"checkpoint to start at sequence index of "
-                  f"{package['next_seq_index']}")
+        # checkpointing: a live hipGraph forbids eager work between
+        # replays (pure-replay rule), so the graph is dropped first and
+        # re-captured after the save (--graph used to skip
+        # checkpoints entirely, losing all progress on interruption)
+        do_ckpt = i % checkpoint_every == 0 and (graphed is None or i > 0)
+        if do_ckpt:
+            if graphed is not None:
+                torch.cuda.synchronize()
+                graphed = None  # free replay state before eager/D2H work
+            # ZeRO-1's state_dict all-gathers shards: collective call
+            optim_sd = optim.state_dict() \
+                if (is_main or getattr(optim, 'state_dict_is_collective',
+                                       False)) else None
+            if is_main:
+                package = {
+                    'next_seq_index': seq_index + effective_batch_size,
+                    'params': tensors_to_numpy(
+                        {k: v for k, v in module.state_dict().items()}),
+                    'optim_state': tensors_to_numpy(optim_sd),
+                    'model_config': model_kwargs,
+                    'run_id': run_id,
+                }
+                save_checkpoint(package, checkpoint_keep_n)
+                print(f"checkpoint to start at sequence index of "
+                      f"{package['next_seq_index']}")
+            if graph_ok:
+                graphed = make_graphed()
 
         if graphed is None and i % validate_every == 0:
             valid_data = my_shard(next(valid_dataset))
