@@ -49,13 +49,35 @@ def _crc32c_table() -> np.ndarray:
     return _CRC_TABLE
 
 
+_NATIVE_CRC = None
+_NATIVE_CRC_TRIED = False
+
+
+def _native_crc():
+    # the _C extension's slicing-by-8 CRC is ~1000x the Python loop;
+    # soft import: data prep must still work before the extension is
+    # built (the GPU-op dispatch in ops/dispatch.py stays strict)
+    global _NATIVE_CRC, _NATIVE_CRC_TRIED
+    if not _NATIVE_CRC_TRIED:
+        _NATIVE_CRC_TRIED = True
+        try:
+            from progen_amd import _C  # noqa: WPS433
+            _NATIVE_CRC = _C.crc32c
+        except Exception:  # noqa: BLE001
+            _NATIVE_CRC = None
+    return _NATIVE_CRC
+
+
 def crc32c(data: bytes) -> int:
+    native = _native_crc()
+    if native is not None:
+        return int(native(data))
     table = _crc32c_table()
     crc = np.uint32(0xFFFFFFFF)
     buf = np.frombuffer(data, dtype=np.uint8)
     tbl = table
     c = int(crc)
-    for b in buf.tobytes():  # byte loop; fine for data-prep volumes
+    for b in buf.tobytes():  # byte loop; fallback when _C is not built
         c = tbl[(c ^ b) & 0xFF] ^ (c >> 8)
         c = int(c)
     return c ^ 0xFFFFFFFF

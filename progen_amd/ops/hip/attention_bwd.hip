@@ -16,32 +16,42 @@
 // (ops/hip/rope_qkv.hip).
 //
 // Geometry: block = one (batch, head, window), 4 waves, each wave owns a
-// 64-row q-chunk. Per 64-key tile the per-row phases (S, P, dS, dQ) are
-// wave-local; the per-KEY phases (dV, dK) are computed as per-wave
-// 16-key OUTPUT SLICES whose MFMA K-dim spans ALL chunks' P/dS/dO^T/Q^T
-// LDS regions — every dV/dK element is produced by exactly ONE wave, so
-// the stores are PLAIN (no atomics): each window's own-band gradients go
-// to dacc, its lookback-band gradients to a separate dlook buffer, and
+// 64-row q-chunk. Round-2 restructure (harness VARIANT 4,
+// tools/ablate_attn_bwd.hip — 1965 -> 1500 us/call at the production
+// grid B=64 H=24 wsz=256, bitwise-identical output; PMC showed the old
+// 5-barriers-per-tile schedule parked waves on barriers 50% of cycles):
+// per 64-key tile there are TWO barriers —
+//   phase 1 (wave-local, no barrier): S, P (P^T b64-written), dP,
+//     dS = P o (dP - D) (dS^T b64-written), and dQ += dS k_s where the
+//     dS A-fragments come from ds_read_b64_tr_b16 transposed reads of
+//     the wave's OWN just-written dS^T region (same-wave DS ordering is
+//     program order). The old separate row-major dS image (dsrl, 16
+//     scattered b16 writes per m,n) is gone.
+//   barrier A; then tile t+1's k/v/kt staging (phase 2 reads none of
+//     them) overlaps with phase 2: the dV and dK 16-key OUTPUT SLICES,
+//     merged into one chunk-loop, whose MFMA K-dim spans ALL chunks'
+//     P^T/dS^T/dO^T/Q^T regions; barrier B.
+// Every dV/dK element is produced by exactly ONE wave, so the stores
+// are PLAIN (no atomics): each window's own-band gradients go to dacc,
+// its lookback-band gradients to a separate dlook buffer, and
 // attn_bwd_finalize_kernel sums the two (adjacent windows share keys
-// through the lookback, progen.py:90-91 — writing both bands to one
-// buffer would need atomics across blocks; the earlier atomic versions
-// were 4x/1x the traffic and the plain-store split measured fastest;
-// cross-round accumulation within a block is a read-modify-write, which
-// tools/ablate_attn_bwd.hip showed is latency-hidden at the production
-// grid depth). dQ rows are exclusively owned -> plain fp32 stores.
-// Window 0's lookback keys are the zero pad; their gradients are
-// discarded.
+// through the lookback, progen.py:90-91). dQ rows are exclusively
+// owned -> plain fp32 stores. Window 0's lookback keys are the zero
+// pad; their gradients are discarded.
 //
-// MFMA operand LDS images (XOR-swizzled, byte ^= (row&7)<<4):
+// MFMA operand LDS images (XOR-swizzled, byte ^= (row&7)<<4 except ds2):
 //   k_lds  [key][dh]   k'           (S B-fragments)
 //   kt_lds [dh][key]   scaled k'    (dQ B-fragments)
 //   v_lds  [key][dh]   v'           (dP B-fragments)
 //   qt_lds [dh][row]   scaled q'    (dK B-fragments, per chunk)
 //   dot_lds[dh][row]   dO           (dV B-fragments, per chunk)
-//   pds_lds [key][row] P then dS^T  (b64-written from the MFMA C-layout:
-//                                    4 consecutive rows at a fixed key =
-//                                    one 8-B write; per chunk)
-//   dsrl_lds [row][key] dS          (scattered b16 writes; per chunk)
+//   pds_lds [key][row] P^T          (b64-written from the MFMA C-layout;
+//                                    per chunk)
+//   ds2_lds [key][row] dS^T         (b64-written, col ^= v(key)*32 with
+//                                    v(k)=((k&2)>>1)|((k&8)>>2) so the
+//                                    tr reads land conflict-free; dK
+//                                    A-frags read it b128, dQ A-frags
+//                                    via tr; per chunk)
 
 #include "common.h"
 
@@ -64,13 +74,21 @@ __device__ __forceinline__ void load_rope(const float* rsin,
   *(f32x4*)(cv + 4) = *(const f32x4*)(rcos + pos * DH + d0 + 4);
 }
 
+// per-key 32-B XOR window for the dS^T image (128-B rows): the half-
+// wave's 8 key-rows {kb..kb+3, kb+8..kb+11} split 4/4 by parity (row
+// base 32*(key&1) dwords) and within a parity v(key) is a bijection
+// onto 0..3, so the 8 rows cover all 64 banks -> zero-conflict tr
+// reads. v*32 <= 96 B stays inside the 128-B row (uk-style *32 from
+// the wgrad image would escape it).
+__device__ __forceinline__ int uk4(int k) { return ((k & 2) >> 1) | ((k & 8) >> 2); }
+
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4t;
+#define AS3 __attribute__((address_space(3)))
+
 __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
-    const short* __restrict__ dout,  // (B, N, H*DH) bf16
-    const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16, PRE-ROTATED
-    const short* __restrict__ out,   // (B, N, H*DH) bf16 (fwd output)
-    const float* __restrict__ lse,   // (B, H, N)
-    float* __restrict__ dacc,        // (B, N, 3*H*DH) fp32 (own + dQ)
-    float* __restrict__ dlook,       // (B, N, 2*H*DH) fp32 (lookback k/v)
+    const short* __restrict__ dout, const short* __restrict__ qkv,
+    const short* __restrict__ out, const float* __restrict__ lse,
+    float* __restrict__ dacc, float* __restrict__ dlook,
     int B, int N, int H, int wsz) {
   const int window = blockIdx.x;
   const int head = blockIdx.y;
@@ -89,44 +107,43 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
   const int k_off = H * DH + head * DH;
   const int v_off = 2 * H * DH + head * DH;
   const long long look_bn = (long long)batch * N * (2LL * H * DH);
-  const int lk_off = head * DH;            // k slot in dlook
-  const int lv_off = H * DH + head * DH;   // v slot in dlook
+  const int lk_off = head * DH;
+  const int lv_off = H * DH + head * DH;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* k_lds = smem;                                   // 8 KiB
-  char* kt_lds = smem + 8192;                           // 8 KiB
-  char* v_lds = smem + 16384;                           // 8 KiB
-  char* qt_base = smem + 24576;                         // 32 KiB (4 chunks)
-  char* dot_base = smem + 24576 + 32768;                // 32 KiB
-  char* pds_base = smem + 24576 + 65536;                // 32 KiB
-  char* dsrl_base = smem + 24576 + 98304;               // 32 KiB
-  float* d_lds = (float*)(smem + 24576 + 131072 + wid * 256);      // 1 KiB
-  float* lse_lds = (float*)(smem + 24576 + 131072 + 1024 + wid * 256);
+  char* k_lds = smem;
+  char* kt_lds = smem + 8192;
+  char* v_lds = smem + 16384;
+  char* qt_base = smem + 24576;                  // 32 KiB
+  char* dot_base = qt_base + 4 * 8192;           // 32 KiB
+  char* pds_base = dot_base + 4 * 8192;          // 32 KiB (P^T)
+  char* ds2_base = pds_base + 4 * 8192;          // 32 KiB (dS^T, XOR-u)
+  float* d_lds = (float*)(ds2_base + 4 * 8192 + wid * 256);
+  float* lse_lds = (float*)(ds2_base + 4 * 8192 + 1024 + wid * 256);
 
   char* qt_lds = qt_base + wid * 8192;
   char* dot_lds = dot_base + wid * 8192;
   char* pds_lds = pds_base + wid * 8192;
-  char* dsrl_lds = dsrl_base + wid * 8192;
+  char* ds2_lds = ds2_base + wid * 8192;
 
   const float scale = rsqrtf((float)DH);
   const int tiles = 2 * wsz / KT;
   const int chunks = wsz / 64;
-  const int rounds = (chunks + ATTN_WAVES - 1) / ATTN_WAVES;
+  const int rounds = (chunks + 3) / 4;
 
-  // T14 staging registers (pure copies of pre-rotated k/v)
   const int su_key[2] = {(int)threadIdx.x >> 3,
                          (int)(threadIdx.x + ATTN_BLOCK) >> 3};
   const int su_d0[2] = {((int)threadIdx.x & 7) * 8,
                         (((int)threadIdx.x + ATTN_BLOCK) & 7) * 8};
 
   for (int round = 0; round < rounds; ++round) {
-    const int chunk = round * ATTN_WAVES + wid;
+    const int chunk = round * 4 + wid;
     const bool active = chunk < chunks;
-    const int nactive = min(ATTN_WAVES, chunks - round * ATTN_WAVES);
+    const int nactive = min(4, chunks - round * 4);
     const int chunk_off = chunk * 64;
     const int q0 = window * wsz + chunk_off;
 
-    bf16x8 qfrag[4][2];  // scaled pre-rotated q fragments
+    bf16x8 qfrag[4][2];
     f32x4 dqacc[4][4];
 #pragma unroll
     for (int m = 0; m < 4; ++m)
@@ -149,8 +166,6 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           qfrag[m][ks] = o;
         }
       }
-
-      // per-round chunk staging: Q^T (scaled), dO^T, D, lse; one lane/row
       {
         const int row = lane;
         const long long gq = qkv_bn + (long long)(q0 + row) * HD3 + q_off;
@@ -175,11 +190,10 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
         lse_lds[row] = lse[((long long)batch * H + head) * N + q0 + row];
       }
     }
-    __syncthreads();  // qt/dot/pds regions ready & previous round done
+    __syncthreads();
 
     const int max_tile = active ? ((chunk_off + 63 + wsz) / KT) : -1;
 
-    // ---- staging prologue (T14): tile 0 ----
     bf16x8 kreg[2], vreg[2];
     auto issue_loads = [&](int t) {
 #pragma unroll
@@ -221,17 +235,17 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
     for (int t = 0; t < tiles; ++t) {
       if (t + 1 < tiles) issue_loads(t + 1);
       const int kb = t * KT;
-      // chunks whose causal range covers this tile: chunk >= c_min
-      const int c_min = max(0, (t * KT - wsz - 63 + 63) / 64 - round * ATTN_WAVES);
+      const int c_min = max(0, (t * KT - wsz) / 64 - round * 4);
       const bool i_compute = active && t <= max_tile;
 
-      f32x4 s[4][4];  // S -> P for this wave's rows
+      // ---- phase 1 (wave-local): S, P, P^T, dP, dS, dS^T, dQ ----
       if (i_compute) {
+        f32x4 s[4][4];
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n) s[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
-        __builtin_amdgcn_s_setprio(1);  // T5: favor MFMA clusters
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
@@ -246,7 +260,6 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           }
         __builtin_amdgcn_s_setprio(0);
 
-        // P = exp(S - lse) masked; b64-write P^T into own pds region
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -273,57 +286,8 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             *(unsigned long long*)(pds_lds + key * 128 + swz(key, row0 * 2)) =
                 *(unsigned long long*)pk;
           }
-      }
-      __syncthreads();  // all P regions ready
 
-      // ---- dV slice: this wave owns keys [wid*16, wid*16+16) of the
-      // tile; K-dim spans contributing chunks' rows ----
-      {
-        f32x4 dv[4];
-#pragma unroll
-        for (int n = 0; n < 4; ++n) dv[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
-        __builtin_amdgcn_s_setprio(1);
-        for (int c = c_min; c < nactive; ++c) {
-          char* pds_c = pds_base + c * 8192;
-          char* dot_c = dot_base + c * 8192;
-#pragma unroll
-          for (int ks = 0; ks < 2; ++ks) {
-            const int key = wid * 16 + l15;
-            const int r0 = ks * 32 + 8 * l4;
-            bf16x8 pf = *(const bf16x8*)(pds_c + key * 128 + swz(key, r0 * 2));
-#pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              const int d = n * 16 + l15;
-              bf16x8 dof = *(const bf16x8*)(dot_c + d * 128 + swz(d, r0 * 2));
-              dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
-                                                              0, 0, 0);
-            }
-          }
-        }
-        __builtin_amdgcn_s_setprio(0);
-        const bool lookback = kb < wsz;  // tile-uniform half of the band
-#pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
-          if (kpos >= 0) {
-            float* dst = lookback
-                ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lv_off
-                : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
-#pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              float v = ((float*)&dv[n])[r];
-              if (round > 0) v += dst[n * 16 + l15];  // later chunk rounds
-              dst[n * 16 + l15] = v;
-            }
-          }
-        }
-      }
-
-      __syncthreads();  // dV reads of every pds region complete before
-                        // any wave overwrites its own with dS
-
-      // ---- dP = dO V'^T ; dS = P o (dP - D); write dS^T + dS ----
-      if (i_compute) {
+        // dP = dO V'^T (wave-local: v_lds staged, dout from global)
         f32x4 dp[4][4];
 #pragma unroll
         for (int m = 0; m < 4; ++m)
@@ -359,6 +323,7 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
               ((float*)&dp[m][n])[r] = p * (d - dval);  // now dS
             }
           }
+        // write dS^T once: b64 into [key][row] image, col ^= u(key)*32
 #pragma unroll
         for (int m = 0; m < 4; ++m)
 #pragma unroll
@@ -368,34 +333,77 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             short dk4[4];
 #pragma unroll
             for (int r = 0; r < 4; ++r) dk4[r] = f2bf(((float*)&dp[m][n])[r]);
-            *(unsigned long long*)(pds_lds + key * 128 + swz(key, row0 * 2)) =
+            *(unsigned long long*)(ds2_lds + key * 128 +
+                                   ((row0 * 2) ^ (uk4(key) * 32))) =
                 *(unsigned long long*)dk4;
-#pragma unroll
-            for (int r = 0; r < 4; ++r)
-              *(short*)(dsrl_lds + (row0 + r) * 128 + swz(row0 + r, key * 2)) =
-                  dk4[r];
           }
-      }
-      __syncthreads();  // all dS regions ready
 
-      // ---- dK slice (keys [wid*16, wid*16+16)): K spans chunks ----
-      {
-        f32x4 dk[4];
+        // dQ += dS k_s: A-frags by tr reads of the wave's OWN dS^T
+        // (same-wave DS ordering; no barrier needed)
+        __builtin_amdgcn_s_setprio(1);
 #pragma unroll
-        for (int n = 0; n < 4; ++n) dk[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int m = 0; m < 4; ++m) {
+            const int krow1 = ks * 32 + l4 * 8 + (l15 >> 2);
+            const int krow2 = krow1 + 4;
+            const int colb = (m * 16 + (l15 & 3) * 4) * 2;
+            auto p1 = (AS3 bf16x4t*)(ds2_lds + krow1 * 128 +
+                                     (colb ^ (uk4(krow1) * 32)));
+            auto p2 = (AS3 bf16x4t*)(ds2_lds + krow2 * 128 +
+                                     (colb ^ (uk4(krow2) * 32)));
+            bf16x4t f1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+            bf16x4t f2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
+            bf16x8 dsf;
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              ((__bf16*)&dsf)[j] = f1[j];
+              ((__bf16*)&dsf)[j + 4] = f2[j];
+            }
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int d = n * 16 + l15;
+              bf16x8 kf = *(const bf16x8*)(kt_lds + d * 128 +
+                                           swz(d, (ks * 32 + 8 * l4) * 2));
+              dqacc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  dsf, kf, dqacc[m][n], 0, 0, 0);
+            }
+          }
+        __builtin_amdgcn_s_setprio(0);
+      }
+      __syncthreads();  // barrier A: P^T/dS^T ready; k/v/kt reads done
+
+      // stage t+1 into k/v/kt while phase 2 runs (phase 2 reads none)
+      if (t + 1 < tiles) write_lds();
+
+      // ---- phase 2: merged dV + dK key-slices ----
+      {
+        f32x4 dv[4], dk[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          dv[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+          dk[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        }
         __builtin_amdgcn_s_setprio(1);
         for (int c = c_min; c < nactive; ++c) {
           char* pds_c = pds_base + c * 8192;
+          char* ds2_c = ds2_base + c * 8192;
+          char* dot_c = dot_base + c * 8192;
           char* qt_c = qt_base + c * 8192;
 #pragma unroll
           for (int ks = 0; ks < 2; ++ks) {
             const int key = wid * 16 + l15;
             const int r0 = ks * 32 + 8 * l4;
-            bf16x8 dsf = *(const bf16x8*)(pds_c + key * 128 + swz(key, r0 * 2));
+            bf16x8 pf = *(const bf16x8*)(pds_c + key * 128 + swz(key, r0 * 2));
+            bf16x8 dsf = *(const bf16x8*)(ds2_c + key * 128 +
+                                          ((r0 * 2) ^ (uk4(key) * 32)));
 #pragma unroll
             for (int n = 0; n < 4; ++n) {
               const int d = n * 16 + l15;
+              bf16x8 dof = *(const bf16x8*)(dot_c + d * 128 + swz(d, r0 * 2));
               bf16x8 qf = *(const bf16x8*)(qt_c + d * 128 + swz(d, r0 * 2));
+              dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
+                                                              0, 0, 0);
               dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, dk[n],
                                                               0, 0, 0);
             }
@@ -407,47 +415,29 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
         for (int r = 0; r < 4; ++r) {
           const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
           if (kpos >= 0) {
-            float* dst = lookback
+            float* dstv = lookback
+                ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lv_off
+                : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
+            float* dstk = lookback
                 ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lk_off
                 : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
 #pragma unroll
             for (int n = 0; n < 4; ++n) {
-              float v = ((float*)&dk[n])[r];
-              if (round > 0) v += dst[n * 16 + l15];
-              dst[n * 16 + l15] = v;
+              float vv = ((float*)&dv[n])[r];
+              float vk = ((float*)&dk[n])[r];
+              if (round > 0) {
+                vv += dstv[n * 16 + l15];
+                vk += dstk[n * 16 + l15];
+              }
+              dstv[n * 16 + l15] = vv;
+              dstk[n * 16 + l15] = vk;
             }
           }
         }
       }
-
-      // ---- dQ += dS k_s (own rows; accumulates across tiles) ----
-      if (i_compute) {
-#pragma unroll
-        for (int ks = 0; ks < 2; ++ks)
-#pragma unroll
-          for (int m = 0; m < 4; ++m) {
-            const int row = m * 16 + l15;
-            bf16x8 dsf = *(const bf16x8*)(dsrl_lds + row * 128 +
-                                          swz(row, (ks * 32 + 8 * l4) * 2));
-#pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              const int d = n * 16 + l15;
-              bf16x8 kf = *(const bf16x8*)(kt_lds + d * 128 +
-                                           swz(d, (ks * 32 + 8 * l4) * 2));
-              dqacc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  dsf, kf, dqacc[m][n], 0, 0, 0);
-            }
-          }
-      }
-
-      __syncthreads();  // done reading k/v/kt LDS for tile t
-      if (t + 1 < tiles) {
-        write_lds();
-        __syncthreads();
-      }
+      __syncthreads();  // barrier B: slices done; t+1 staged
     }
 
-    // ---- store dQ (rows exclusively owned -> plain fp32 stores) ----
     if (active) {
 #pragma unroll
       for (int m = 0; m < 4; ++m)

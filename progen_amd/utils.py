@@ -113,6 +113,7 @@ def sample(
     add_bos: bool = False,
     generator: Optional[torch.Generator] = None,
     device=None,
+    reference_add_bos_quirk: bool = False,
 ) -> torch.Tensor:
     """Gumbel-max top-k autoregressive decoding, reference semantics
     (reference: utils.py:106-135):
@@ -123,6 +124,15 @@ def sample(
         logits to 0, not -inf (utils.py:97-100);
       - everything after the second pad/EOS token is zeroed (utils.py:132-133).
 
+    ``reference_add_bos_quirk``: reproduce the reference's add_bos
+    off-by-one bit-for-bit (utils.py:110-116): the BOS pad shifts the
+    prime right but start_pos is NOT advanced, so the loop starts at
+    the position now holding the LAST PRIME TOKEN, and the one-hot
+    `seq += one_hot * sampled` ADDS the first sample onto it. Off by
+    default (the fixed semantics keep the prime intact); on request the
+    reference behavior is reproduced exactly (each quirk
+    deviation needs an explicit test + flag — tests/test_sample.py).
+
     fn: (n,) int64 tensor -> (n, V) logits (e.g. a closure over
     TransformedProGen.apply or the bare module).
     """
@@ -130,7 +140,7 @@ def sample(
     start_pos = prime.shape[-1]
     pad = (0, length - start_pos) if not add_bos else (1, length - start_pos - 1)
     seq = torch.nn.functional.pad(prime, pad)
-    if add_bos:
+    if add_bos and not reference_add_bos_quirk:
         start_pos += 1
 
     for curr_pos in range(start_pos, length):
@@ -145,7 +155,10 @@ def sample(
             noise = noise * mask
 
         sampled = (logits + noise).argmax(dim=-1)
-        seq[curr_pos] = sampled
+        # reference uses `seq += one_hot * sampled` (utils.py:128-129):
+        # an ADD, which only differs from assignment at the quirk's
+        # first position (everywhere else the slot is 0)
+        seq[curr_pos] = seq[curr_pos] + sampled
 
     # zero after 2nd pad token (the 1st learned pad acts as EOS)
     remove_after_eos = (seq == 0).long().cumsum(dim=-1) > 1

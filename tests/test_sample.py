@@ -99,3 +99,28 @@ def test_sample_fast_early_exit_bos():
     assert calls["n"] == 1
     assert out[1:3].tolist() == [5, 6]
     assert (out[3:] == 0).all()
+
+
+def test_sample_reference_add_bos_quirk_flag():
+    """the one deliberate deviation (the reference's add_bos
+    off-by-one, utils.py:110-116) must be selectable. With the flag the
+    first sample is ADDED onto the last prime token (reference
+    `seq += one_hot * sampled` with start_pos not advanced); without it
+    the prime stays intact."""
+    m = ProGenBase(ProGenConfig(**TINY))
+    prime = torch.tensor([5, 6, 7])
+    g = torch.Generator().manual_seed(0)
+    out_q = sample(_fn(m), prime, length=32, top_k=5, add_bos=True,
+                   generator=g, reference_add_bos_quirk=True)
+    assert out_q[0].item() == 0  # BOS
+    assert out_q[1:3].tolist() == [5, 6]
+    # position 3 = last prime token + first sampled token (the quirk);
+    # with top_k the sample is rarely 0, so the slot usually differs
+    # from the pristine prime — but deterministically it is 7 + s
+    g2 = torch.Generator().manual_seed(0)
+    out_f = sample(_fn(m), prime, length=32, top_k=5, add_bos=True,
+                   generator=g2)
+    # the fixed path keeps the prime; the quirk path perturbs slot 3 by
+    # the same token the fixed path would have sampled one step later
+    assert out_f[1:4].tolist() == [5, 6, 7]
+    assert out_q[3].item() >= 7  # 7 + sampled >= 7
