@@ -53,6 +53,11 @@ class Linear(nn.Linear):
         _haiku_linear_init_(self.weight, self.bias)
 
     def forward(self, x: torch.Tensor) -> torch.Tensor:
+        from ..ops import fp8
+        if fp8.fp8_eligible(x, self.weight):
+            # PROGEN_FP8=1: big projections run e4m3 hipBLASLt GEMMs
+            # (fwd + dgrad at the 2x fp8 MFMA rate, wgrad bf16)
+            return fp8.fp8_linear(x, self.weight, self.bias)
         from ..ops.overlap import overlap_linear
         return overlap_linear(x, self.weight, self.bias)
 

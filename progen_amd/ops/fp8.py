@@ -55,3 +55,84 @@ def matmul_sim_fp8(a: torch.Tensor, b: torch.Tensor,
     qb, sb = quantize_e4m3(b)
     acc = qa.float() @ qb.float()
     return (acc * (sa * sb)).to(out_dtype)
+
+
+# ---------------------------------------------------------------------------
+# real fp8 GEMM path (round 2): hipBLASLt e4m3 via torch._scaled_mm
+# ---------------------------------------------------------------------------
+#
+# MI355X fp8 MFMA is 2x the bf16 rate (~5 PF/s dense). The projections
+# route through torch._scaled_mm, which on ROCm lowers to hipBLASLt
+# fp8 GEMMs with fp32 accumulation:
+#   forward: y = x8 @ W8^T        (x row-major, W (N,K) row-major, so
+#                                  W.t() is naturally column-major as
+#                                  _scaled_mm requires)
+#   dgrad:   dx = dy8 @ W8        (needs a column-major W copy)
+#   wgrad:   bf16 (dW feeds the fp32 master update; keeping it bf16
+#            avoids the e5m2-vs-e4m3 gradient-range question — measured
+#            fwd+dgrad are where the 2x rate pays; TODO: fp8 wgrad)
+# Scaling is per-tensor dynamic amax (device-side, graph-capturable).
+# Enable with PROGEN_FP8=1 (train.py/bench.py); only GEMMs with every
+# dim >= FP8_MIN_DIM route (the V=256 head and tiny test models stay
+# bf16).
+
+import os
+
+FP8_MIN_DIM = 1024
+ENABLED = os.environ.get("PROGEN_FP8", "0") == "1"
+
+
+def scaled_mm(a: torch.Tensor, b_colmajor: torch.Tensor,
+              out_dtype: torch.dtype = torch.bfloat16,
+              bias: torch.Tensor = None) -> torch.Tensor:
+    """(M,K) row-major @ (K,N) column-major in e4m3; fp32 accumulate."""
+    qa, sa = quantize_e4m3(a)
+    qb, sb = quantize_e4m3(b_colmajor)
+    return torch._scaled_mm(qa, qb, scale_a=sa, scale_b=sb, bias=bias,
+                            out_dtype=out_dtype)
+
+
+class _Fp8LinearFn(torch.autograd.Function):
+    @staticmethod
+    def forward(ctx, x, weight, bias):
+        x2 = x.reshape(-1, x.shape[-1])
+        ctx.save_for_backward(x2, weight)
+        ctx.has_bias = bias is not None
+        ctx.x_shape = x.shape
+        # W (N,K) row-major -> W.t() is (K,N) column-major
+        qx, sx = quantize_e4m3(x2)
+        qw, sw = quantize_e4m3(weight)
+        y = torch._scaled_mm(qx, qw.t(), scale_a=sx, scale_b=sw,
+                             bias=bias.to(torch.bfloat16) if bias is not None else None,
+                             out_dtype=x.dtype)
+        return y.reshape(*x.shape[:-1], weight.shape[0])
+
+    @staticmethod
+    def backward(ctx, dy):
+        x2, weight = ctx.saved_tensors
+        dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
+        # dgrad in fp8: dy (M,N) @ W (N,K); _scaled_mm wants the second
+        # operand column-major, so quantize W^T and transpose the fp8
+        qdy, sdy = quantize_e4m3(dy2)
+        qwt, swt = quantize_e4m3(weight.t().contiguous())
+        dx = torch._scaled_mm(qdy, qwt.t(), scale_a=sdy, scale_b=swt,
+                              out_dtype=dy.dtype)
+        # wgrad in bf16 (library GEMM)
+        dw = dy2.transpose(0, 1).matmul(x2)
+        db = dy2.sum(dim=0) if ctx.has_bias else None
+        return dx.reshape(ctx.x_shape), dw, db
+
+
+def fp8_linear(x: torch.Tensor, weight: torch.Tensor,
+               bias: torch.Tensor = None) -> torch.Tensor:
+    return _Fp8LinearFn.apply(x, weight, bias)
+
+
+def fp8_eligible(x: torch.Tensor, weight: torch.Tensor) -> bool:
+    if not (ENABLED and x.is_cuda and x.dtype == torch.bfloat16):
+        return False
+    m = x.numel() // x.shape[-1]
+    n, k = weight.shape
+    # _scaled_mm needs 16-divisible dims; route only big projections
+    return (min(m, n, k) >= FP8_MIN_DIM and m % 16 == 0 and n % 16 == 0
+            and k % 16 == 0)
