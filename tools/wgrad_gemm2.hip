@@ -30,8 +30,14 @@
 // VARIANT 0: 2 LDS buffers, glds for chunk t+1 issued before compute
 //            of chunk t, vmcnt(0) + plain __syncthreads() per K-step
 //            (the guide's "glds, 2 buffers, BK=64" configuration).
-// VARIANT 1: 3 buffers, counted vmcnt leaves one tile in flight across
-//            each raw s_barrier (the 8-phase template's sync scheme).
+// VARIANT 1: 2 buffers, counted vmcnt across raw s_barriers keeps one
+//            tile's glds in flight (the 8-phase template's sync
+//            scheme, 2-deep).
+// VARIANT 2: ABLATION — stage tile 0 once, loop MFMAs+barriers over it
+//            (wrong results; isolates the MFMA+barrier skeleton time).
+// VARIANT 3: ABLATION — staging+sync only, MFMAs skipped with the
+//            fragments kept alive via asm (wrong results; isolates the
+//            staging/HBM side).
 //
 // Build/run (GPU box):
 //   hipcc --offload-arch=gfx950 -O3 -std=c++17 [-DVARIANT=1] \
