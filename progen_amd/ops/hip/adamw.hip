@@ -27,7 +27,10 @@ __global__ __launch_bounds__(ADAMW_BLOCK) void fused_adamw_kernel(
     const long long* __restrict__ ends, const int* __restrict__ decay_flags,
     int nchunks, float lr, float b1, float b2, float eps, float wd,
     const int* __restrict__ step_ptr, float grad_scale,
-    const float* __restrict__ clip_coef) {
+    const float* __restrict__ clip_coef, long long shard_off) {
+  // shard_off: ZeRO-1 — master/exp_avg/exp_avg_sq hold only the rank's
+  // [lo, hi) slice of the flat space; chunk tables are pre-clipped to
+  // that range and the fp32 state is indexed at (i - shard_off).
   const float clip = *clip_coef;
   const int step = *step_ptr;
   const float bc1 = 1.f - powf(b1, (float)step);
@@ -39,13 +42,14 @@ __global__ __launch_bounds__(ADAMW_BLOCK) void fused_adamw_kernel(
     for (long long i = s + threadIdx.x; i < e; i += ADAMW_BLOCK) {
       float g = IS_BF16 ? bf2f(((const short*)grads)[i]) : ((const float*)grads)[i];
       g *= grad_scale * clip;
-      float m = exp_avg[i] = b1 * exp_avg[i] + (1.f - b1) * g;
-      float v = exp_avg_sq[i] = b2 * exp_avg_sq[i] + (1.f - b2) * g * g;
+      const long long si = i - shard_off;
+      float m = exp_avg[si] = b1 * exp_avg[si] + (1.f - b1) * g;
+      float v = exp_avg_sq[si] = b2 * exp_avg_sq[si] + (1.f - b2) * g * g;
       float mhat = m / bc1;
       float vhat = v / bc2;
-      float p = master[i];
+      float p = master[si];
       p -= lr * (mhat / (sqrtf(vhat) + eps) + wdc * p);
-      master[i] = p;
+      master[si] = p;
       if (IS_BF16) ((short*)params)[i] = f2bf(p);
       else ((float*)params)[i] = p;
     }
@@ -109,17 +113,19 @@ void fused_adamw_launch(float* master, void* params, const void* grads,
                         float b1, float b2, float eps, float wd,
                         int* step_dev, float grad_scale,
                         const float* clip_coef, bool is_bf16,
-                        hipStream_t stream) {
+                        long long shard_off, hipStream_t stream) {
   step_inc_kernel<<<1, 1, 0, stream>>>(step_dev);
   int grid = nchunks < 2048 ? nchunks : 2048;
   if (is_bf16)
     fused_adamw_kernel<true><<<grid, ADAMW_BLOCK, 0, stream>>>(
         master, params, grads, exp_avg, exp_avg_sq, starts, ends, decay_flags,
-        nchunks, lr, b1, b2, eps, wd, step_dev, grad_scale, clip_coef);
+        nchunks, lr, b1, b2, eps, wd, step_dev, grad_scale, clip_coef,
+        shard_off);
   else
     fused_adamw_kernel<false><<<grid, ADAMW_BLOCK, 0, stream>>>(
         master, params, grads, exp_avg, exp_avg_sq, starts, ends, decay_flags,
-        nchunks, lr, b1, b2, eps, wd, step_dev, grad_scale, clip_coef);
+        nchunks, lr, b1, b2, eps, wd, step_dev, grad_scale, clip_coef,
+        shard_off);
 }
 
 }  // extern "C"

@@ -214,7 +214,7 @@ void fused_adamw(at::Tensor& master, at::Tensor& params,
                  const at::Tensor& chunk_ends, const at::Tensor& chunk_decay,
                  double lr, double b1, double b2, double eps, double wd,
                  at::Tensor& step_dev, double grad_scale,
-                 const at::Tensor& clip_coef) {
+                 const at::Tensor& clip_coef, int64_t shard_off) {
   TORCH_CHECK(step_dev.scalar_type() == at::kInt && step_dev.is_cuda());
   TORCH_CHECK(master.is_cuda() && master.scalar_type() == at::kFloat);
   bool bf = params.scalar_type() == at::kBFloat16;
@@ -225,7 +225,8 @@ void fused_adamw(at::Tensor& master, at::Tensor& params,
                      chunk_decay.data_ptr<int>(), chunk_starts.size(0),
                      (float)lr, (float)b1, (float)b2, (float)eps, (float)wd,
                      step_dev.data_ptr<int>(), (float)grad_scale,
-                     clip_coef.data_ptr<float>(), bf, cur_stream());
+                     clip_coef.data_ptr<float>(), bf, (long long)shard_off,
+                     cur_stream());
 }
 
 // ---------------------------------------------------------------------------
@@ -391,7 +392,13 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("ce_fwd", &ce_fwd, "fused CE forward (nll, lse)");
   m.def("ce_bwd", &ce_bwd, "fused CE backward");
   m.def("grad_sumsq", &grad_sumsq, "sum of squares of flat grads");
-  m.def("fused_adamw", &fused_adamw, "fused clip+AdamW over flat space");
+  m.def("fused_adamw", &fused_adamw, "fused clip+AdamW over flat space",
+        py::arg("master"), py::arg("params"), py::arg("grads"),
+        py::arg("exp_avg"), py::arg("exp_avg_sq"), py::arg("chunk_starts"),
+        py::arg("chunk_ends"), py::arg("chunk_decay"), py::arg("lr"),
+        py::arg("b1"), py::arg("b2"), py::arg("eps"), py::arg("wd"),
+        py::arg("step_dev"), py::arg("grad_scale"), py::arg("clip_coef"),
+        py::arg("shard_off") = 0);
   m.def("rope_qkv", &rope_qkv, "pre-rotation of qkv (rotary on q,k,v)");
   m.def("attn_fwd", &attn_fwd, "fused local attention forward");
   m.def("attn_bwd", &attn_bwd, "fused local attention backward");

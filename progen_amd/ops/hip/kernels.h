@@ -38,7 +38,7 @@ void fused_adamw_launch(float* master, void* params, const void* grads,
                         float b1, float b2, float eps, float wd,
                         int* step_dev, float grad_scale,
                         const float* clip_coef, bool is_bf16,
-                        hipStream_t stream);
+                        long long shard_off, hipStream_t stream);
 
 void rope_qkv_launch(const void* qkv, const float* rsin, const float* rcos,
                      void* qkv_rot, int B, int N, int H, hipStream_t stream);

@@ -66,25 +66,63 @@ class Zero1AdamW(ProGenAdamW):
         self.exp_avg_sq = torch.zeros_like(self.master)
         self._decay_mask = self._decay_mask[sl].clone()
 
-    def _step_eager(self, grad_scale: float) -> None:
-        g_full = self.space.flat_grad.float() * grad_scale
-        coef = self._clip_coef(g_full, grad_scale)  # full-grad norm: no comm
-        g32 = g_full[self.lo:self.hi] * coef
-        self.master.sub_(self._adamw_update(g32))
-        # scatter the updated slice into this rank's padded gather buffer
+        # shard-clipped chunk tables for the fused kernel: each full-
+        # space chunk intersected with [lo, hi); the kernel indexes the
+        # fp32 state at (i - shard_off) (ops/hip/adamw.hip)
+        starts = self.chunk_starts.clamp(min=self.lo, max=self.hi)
+        ends = self.chunk_ends.clamp(min=self.lo, max=self.hi)
+        keep = ends > starts
+        self.shard_chunk_starts = starts[keep].clone()
+        self.shard_chunk_ends = ends[keep].clone()
+        self.shard_chunk_decay = self.chunk_decay[keep].clone()
+
+    def _gather_params(self) -> None:
+        """All-gather the updated bf16 param slice into every rank's
+        flat buffer (the one collective of the ZeRO-1 step)."""
         flat = self.space.flat
         buf = torch.zeros(self.shard, dtype=flat.dtype, device=flat.device)
-        buf[: self.hi - self.lo] = self.master.to(flat.dtype)
+        buf[: self.hi - self.lo] = flat[self.lo:self.hi] \
+            if self.is_low_precision else self.master.to(flat.dtype)
         out = torch.empty(self.shard * self.world, dtype=flat.dtype,
                           device=flat.device)
         dist.all_gather_into_tensor(out, buf, group=self.group)
         flat.copy_(out[: flat.numel()])
 
+    def _step_eager(self, grad_scale: float) -> None:
+        g_full = self.space.flat_grad.float() * grad_scale
+        coef = self._clip_coef(g_full, grad_scale)  # full-grad norm: no comm
+        g32 = g_full[self.lo:self.hi] * coef
+        self.master.sub_(self._adamw_update(g32))
+        if self.is_low_precision:
+            self.space.flat[self.lo:self.hi].copy_(
+                self.master.to(self.space.flat.dtype))
+        else:
+            self.space.flat[self.lo:self.hi].copy_(self.master)
+        self._gather_params()
+
     def _step_hip(self, grad_scale: float) -> None:
-        # v1 runs the sharded update with torch ops (the fused adamw
-        # kernel addresses full-space chunk tables; a shard-local chunk
-        # table is round-2 work — TODO.md). Correctness is identical.
-        self._step_eager(grad_scale)
+        """Sharded fused update: the kernel touches only this rank's
+        chunk slices (master/moments indexed at i - lo), writes the
+        updated bf16 params in place at [lo, hi), then the slice is
+        all-gathered (no more eager fallback)."""
+        from ..ops import dispatch
+        C = dispatch.ext()
+        if self.max_grad_norm is None:
+            clip_coef = torch.ones(1, device=self.master.device)
+        else:
+            norm = C.grad_sumsq(self.space.flat_grad).sqrt_() * grad_scale
+            clip_coef = self.max_grad_norm / torch.clamp_min(
+                norm, self.max_grad_norm)
+        C.fused_adamw(
+            self.master, self.space.flat, self.space.flat_grad,
+            self.exp_avg, self.exp_avg_sq,
+            self.shard_chunk_starts, self.shard_chunk_ends,
+            self.shard_chunk_decay,
+            float(self.lr), float(self.betas[0]), float(self.betas[1]),
+            float(self.eps), float(self.weight_decay), self.step_dev,
+            float(grad_scale), clip_coef, shard_off=self.lo,
+        )
+        self._gather_params()
 
     def resync_master(self) -> None:
         self.master.copy_(self.space.flat[self.lo:self.hi].float())

@@ -173,6 +173,9 @@ __global__ __launch_bounds__(BLOCK) void wgrad2_kernel(
   asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
   __syncthreads();
   for (long long t = 0; t < nsteps; ++t) {
+#if VARIANT == 2
+    char* cur = smem;  // ablation: same staged tile every step
+#else
     char* cur = smem + (t & 1) * 65536;
     if (t + 1 < nsteps) {
       const long long koff = (t + 1) * BK;
@@ -180,6 +183,7 @@ __global__ __launch_bounds__(BLOCK) void wgrad2_kernel(
       stage_glds(asrc + koff * lda, lda, nxt);
       stage_glds(bsrc + koff * ldb2, ldb2, nxt + 32768);
     }
+#endif
 #endif
 
     __builtin_amdgcn_s_setprio(1);
@@ -192,12 +196,21 @@ __global__ __launch_bounds__(BLOCK) void wgrad2_kernel(
 #pragma unroll
       for (int mb = 0; mb < 8; ++mb)
         afr[mb] = frag_tr(cur, ks * 32, wr * 128 + mb * 16, l15, l4);
+#if VARIANT == 3
+      // ablation: keep fragments alive without MFMAs (guide rule 17:
+      // plain #if-out would DCE the tr reads and the staging)
+#pragma unroll
+      for (int mb = 0; mb < 8; ++mb) asm volatile("" :: "v"(afr[mb]));
+#pragma unroll
+      for (int nb = 0; nb < 4; ++nb) asm volatile("" :: "v"(bfr[nb]));
+#else
 #pragma unroll
       for (int mb = 0; mb < 8; ++mb)
 #pragma unroll
         for (int nb = 0; nb < 4; ++nb)
           acc[mb][nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
               afr[mb], bfr[nb], acc[mb][nb], 0, 0, 0);
+#endif
     }
     __builtin_amdgcn_s_setprio(0);
 
@@ -364,6 +377,10 @@ int main(int argc, char** argv) {
     printf("G2 VARIANT %d wgrad %dx%dx%lld S=%d: %.1f us  %.1f TF/s\n",
            VARIANT, M, N, K, S, us, tf);
 
+#if VARIANT >= 2
+    printf("(ablation variant: results not checked)\n");
+    continue;
+#endif
     std::vector<float> got((size_t)M * N);
     hipMemcpy(got.data(), dacc, (size_t)M * N * 4, hipMemcpyDeviceToHost);
     auto b2f = [](short s) {
