@@ -261,6 +261,84 @@ def forward_step_static(model: ProGenBase, token: torch.Tensor,
     # increments it (in-graph when captured)
 
 
+class GraphedDecodeStep:
+    """hipGraph-captured decode step (TODO.md "Serving"; VERDICT r1 item
+    6): one replay = forward_step_static + in-graph pos_dev increment.
+    The eager per-token step is launch-bound (~500 kernel launches); the
+    graph replays them as one unit.
+
+    Pure-replay discipline (profiles/r01_graph_interleave_bug.md): no
+    eager KERNEL may run between replays. The per-token loop only does
+    host<->device MEMCPYs (token in, logits out) and CPU-side sampling,
+    which keeps replay state intact — validated by
+    tests/test_gpu_decode_graph.py against the eager cached decoder.
+
+    Usage:
+        cache = DecodeCache(model, batch=B)
+        ... eager prefill via forward_step ...
+        g = GraphedDecodeStep(model, cache, start_pos=cache.pos)
+        for _ in range(n):
+            logits = g.step(tokens)        # (B, V) device tensor
+            tokens = <sample on host>      # no device kernels!
+    """
+
+    def __init__(self, model: ProGenBase, cache: DecodeCache,
+                 start_pos: int, warmup: int = 2):
+        dev = next(model.parameters()).device
+        batch = cache.attn_prev[0].shape[0]
+        self.cache = cache
+        self.token = torch.zeros(batch, dtype=torch.long, device=dev)
+        self.pos_dev = torch.tensor(start_pos, dtype=torch.long, device=dev)
+
+        # warmup + capture perturb the cache and pos; snapshot & restore
+        snap = {
+            "k": [t.clone() for t in cache.k],
+            "v": [t.clone() for t in cache.v],
+            "ap": [t.clone() for t in cache.attn_prev],
+            "fp": [t.clone() for t in cache.ff_prev],
+            "gh": [t.clone() for t in cache.gate_hist if t is not None],
+            "pos": self.pos_dev.clone(),
+        }
+
+        side = torch.cuda.Stream()
+        side.wait_stream(torch.cuda.current_stream())
+        with torch.cuda.stream(side):
+            for _ in range(warmup):
+                forward_step_static(model, self.token, cache, self.pos_dev)
+                self.pos_dev += 1
+        torch.cuda.current_stream().wait_stream(side)
+        torch.cuda.synchronize()
+
+        self.graph = torch.cuda.CUDAGraph()
+        with torch.cuda.graph(self.graph):
+            self.static_logits = forward_step_static(
+                model, self.token, cache, self.pos_dev)
+            self.pos_dev += 1  # in-graph position advance
+
+        with torch.no_grad():
+            for dst, src in zip(cache.k, snap["k"]):
+                dst.copy_(src)
+            for dst, src in zip(cache.v, snap["v"]):
+                dst.copy_(src)
+            for dst, src in zip(cache.attn_prev, snap["ap"]):
+                dst.copy_(src)
+            for dst, src in zip(cache.ff_prev, snap["fp"]):
+                dst.copy_(src)
+            gh = [t for t in cache.gate_hist if t is not None]
+            for dst, src in zip(gh, snap["gh"]):
+                dst.copy_(src)
+            self.pos_dev.copy_(snap["pos"])
+        torch.cuda.synchronize()
+
+    def step(self, tokens: torch.Tensor) -> torch.Tensor:
+        """Replay one decode step. ``tokens``: (B,) int64 (host or
+        device); returns the (B, V) static logits tensor — copy it out
+        before the next replay."""
+        self.token.copy_(tokens.reshape(-1), non_blocking=False)
+        self.graph.replay()
+        return self.static_logits
+
+
 @torch.no_grad()
 def sample_cached(
     model: ProGenBase,
@@ -270,6 +348,7 @@ def sample_cached(
     add_bos: bool = False,
     generator: Optional[torch.Generator] = None,
     device=None,
+    graph: bool = False,
 ) -> torch.Tensor:
     """Drop-in ``utils.sample`` with O(window + n_sgu) per-token cost.
 
@@ -293,6 +372,11 @@ def sample_cached(
     for p in range(start_pos):                           # prefill
         logits = forward_step(model, seq[p:p + 1].to(dev), cache)
 
+    graphed = None
+    if graph and dev.type == "cuda":
+        logits = logits.clone()  # the static buffer will be reused
+        graphed = GraphedDecodeStep(model, cache, start_pos=start_pos)
+
     pads_seen = int((seq[:start_pos] == 0).sum())
     for curr_pos in range(start_pos, length):
         logits_row = logits[0].float().cpu()
@@ -308,8 +392,12 @@ def sample_cached(
             if pads_seen >= 2:
                 break
         if curr_pos + 1 < length:
-            logits = forward_step(model, seq[curr_pos:curr_pos + 1].to(dev),
-                                  cache)
+            if graphed is not None:
+                logits = graphed.step(seq[curr_pos:curr_pos + 1])
+            else:
+                logits = forward_step(model,
+                                      seq[curr_pos:curr_pos + 1].to(dev),
+                                      cache)
 
     remove_after_eos = (seq == 0).long().cumsum(dim=-1) > 1
     return seq * (~remove_after_eos).long()
@@ -323,6 +411,7 @@ def sample_cached_batch(
     top_k: Optional[int] = None,
     generator: Optional[torch.Generator] = None,
     device=None,
+    graph: bool = False,
 ) -> torch.Tensor:
     """Batched incremental decode: one forward_step advances ALL rows
     (the per-layer caches are batch-first), so serving throughput scales
@@ -347,12 +436,17 @@ def sample_cached_batch(
         seq[i, :lens[i]] = p
 
     cache = DecodeCache(model, batch=B, device=dev)
+    graphed = GraphedDecodeStep(model, cache, start_pos=0) \
+        if (graph and dev.type == "cuda") else None
     greedy = top_k is None and generator is None
     # per-row pad counts over WRITTEN tokens only (the unfilled zero tail
     # must not count as EOS); position 0 is written at entry
     pads = [int(seq[i, 0] == 0) for i in range(B)]
     for pos in range(length - 1):
-        logits = forward_step(model, seq[:, pos].to(dev), cache)  # (B, V)
+        if graphed is not None:
+            logits = graphed.step(seq[:, pos])
+        else:
+            logits = forward_step(model, seq[:, pos].to(dev), cache)  # (B, V)
         rows = logits.float().cpu()
         if greedy:
             nxt = rows.argmax(dim=-1)

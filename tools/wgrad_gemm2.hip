@@ -38,6 +38,18 @@
 // VARIANT 3: ABLATION — staging+sync only, MFMAs skipped with the
 //            fragments kept alive via asm (wrong results; isolates the
 //            staging/HBM side).
+// VARIANT 5: hand-scheduled pipeline. The round-2 PMC + .s inspection
+//            found hipcc emits `s_waitcnt vmcnt(0)` before the FIRST
+//            ds_read of every k-step whenever a glds is in flight (the
+//            guide's §6 trap), draining the prefetched tile and making
+//            VARIANT 1's counted-vmcnt scheme identical to VARIANT 0
+//            (both measured 860 TF/s vs the 1463 TF/s compute-only
+//            skeleton). Here the tr reads are INLINE ASM (invisible to
+//            hipcc's alias analysis, so it stops injecting the drain),
+//            with manual `s_waitcnt lgkmcnt(0)` + sched_barrier before
+//            each MFMA cluster, raw s_barriers, and a counted
+//            vmcnt(8) that keeps one tile's glds in flight across the
+//            barrier pair.
 //
 // Build/run (GPU box):
 //   hipcc --offload-arch=gfx950 -O3 -std=c++17 [-DVARIANT=1] \
@@ -107,6 +119,140 @@ __device__ __forceinline__ bf16x8 frag_tr(const char* img, int kb, int m0,
   }
   return o;
 }
+
+#if VARIANT == 5
+// one 8-bf16 MFMA A/B fragment from two inline-asm tr reads at integer
+// LDS byte offsets (compiler cannot see these as LDS reads)
+union FragU {
+  bf16x8 v;
+  struct { unsigned long long lo, hi; } u;
+};
+
+__device__ __forceinline__ bf16x8 frag_tr_asm(unsigned off1, unsigned off2) {
+  FragU f;
+  // early-clobber (=&v) is required: instruction 1's destination must
+  // not be allocated over instruction 2's address operand (%3) — plain
+  // "=v" produced exactly that aliasing and garbage fragments
+  asm volatile("ds_read_b64_tr_b16 %0, %2\n\tds_read_b64_tr_b16 %1, %3"
+               : "=&v"(f.u.lo), "=&v"(f.u.hi)
+               : "v"(off1), "v"(off2));
+  return f.v;
+}
+
+__global__ __launch_bounds__(BLOCK) void wgrad2_v5_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    float* __restrict__ dw_acc, int M, int N, long long K, int splits) {
+  const int ntiles_n = N / BN;
+  const int nwg = (M / BM) * ntiles_n;
+  const int orig = blockIdx.x;
+  const int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+               + orig / 8;
+  const int tm = wg / ntiles_n, tn = wg % ntiles_n;
+
+  const long long kchunk = K / splits;
+  const long long k_lo = blockIdx.z * kchunk;
+  const long long nsteps = kchunk / BK;
+
+  const int lane = (int)threadIdx.x & 63;
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+  const int wid = (int)threadIdx.x >> 6;
+  const int wr = wid >> 2;
+  const int wc = wid & 3;
+
+  const long long lda = 2LL * M;
+  const long long ldb2 = 2LL * N;
+  const char* asrc = (const char*)dy + k_lo * lda + (long long)tm * 512;
+  const char* bsrc = (const char*)x + k_lo * ldb2 + (long long)tn * 512;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+
+  // per-lane row constants: rows r(ks,j) = ks*32 + l4*8 + (l15>>2) + 4j
+  // (j=0,1); swizzled col = base + 32*(blk ^ u(row)) per §file header
+  unsigned arow[2][2], brow[2][2], au[2][2];
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int rr = ks * 32 + l4 * 8 + (l15 >> 2) + 4 * j;
+      au[ks][j] = (unsigned)uk(rr);
+      arow[ks][j] = (unsigned)(rr * 512 + wr * 256 + (l15 & 3) * 8);
+      brow[ks][j] = (unsigned)(32768 + rr * 512 + wc * 128 + (l15 & 3) * 8);
+    }
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  {
+    stage_glds(asrc, lda, smem);
+    stage_glds(bsrc, ldb2, smem + 32768);
+    if (nsteps > 1) {
+      stage_glds(asrc + BK * lda, lda, smem + 65536);
+      stage_glds(bsrc + BK * ldb2, ldb2, smem + 65536 + 32768);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(8)" ::: "memory");  // tile 0 landed (mine)
+  __builtin_amdgcn_s_barrier();                     // ...and everyone's
+
+  for (long long t = 0; t < nsteps; ++t) {
+    const unsigned tb = (unsigned)((t & 1) * 65536);
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 bfr[4], afr[8];
+#pragma unroll
+      for (int nb = 0; nb < 4; ++nb)
+        bfr[nb] = frag_tr_asm(tb + brow[ks][0] + 32 * (nb ^ au[ks][0]),
+                              tb + brow[ks][1] + 32 * (nb ^ au[ks][1]));
+#pragma unroll
+      for (int mb = 0; mb < 8; ++mb)
+        afr[mb] = frag_tr_asm(tb + arow[ks][0] + 32 * (mb ^ au[ks][0]),
+                              tb + arow[ks][1] + 32 * (mb ^ au[ks][1]));
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);  // §5.4 rule 18: pin MFMAs after
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mb = 0; mb < 8; ++mb)
+#pragma unroll
+        for (int nb = 0; nb < 4; ++nb)
+          acc[mb][nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[mb], bfr[nb], acc[mb][nb], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    __builtin_amdgcn_s_barrier();  // raw: buffer t&1 reads done everywhere
+    if (t + 2 < nsteps) {
+      const long long koff = (t + 2) * BK;
+      char* nxt = smem + (t & 1) * 65536;
+      stage_glds(asrc + koff * lda, lda, nxt);
+      stage_glds(bsrc + koff * ldb2, ldb2, nxt + 32768);
+      asm volatile("s_waitcnt vmcnt(8)" ::: "memory");  // t+1 landed (mine)
+    } else if (t + 1 < nsteps) {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    if (t + 1 < nsteps)
+      __builtin_amdgcn_s_barrier();  // t+1 landed in every wave's rows
+  }
+
+  const long long mbase = (long long)tm * BM + wr * 128;
+  const long long nbase = (long long)tn * BN + wc * 64;
+#pragma unroll
+  for (int mb = 0; mb < 8; ++mb)
+#pragma unroll
+    for (int r4 = 0; r4 < 4; ++r4) {
+      const long long m = mbase + mb * 16 + l4 * 4 + r4;
+#pragma unroll
+      for (int nb = 0; nb < 4; ++nb) {
+        const long long n = nbase + nb * 16 + l15;
+        const float v = ((float*)&acc[mb][nb])[r4];
+        if (splits > 1) atomicAdd(dw_acc + m * N + n, v);
+        else dw_acc[m * N + n] = v;
+      }
+    }
+}
+#endif  // VARIANT == 5
 
 __global__ __launch_bounds__(BLOCK) void wgrad2_kernel(
     const short* __restrict__ dy,  // (K, M) bf16
@@ -354,7 +500,11 @@ int main(int argc, char** argv) {
 
     auto run = [&]() {
       if (S > 1) hipMemsetAsync(dacc, 0, (size_t)M * N * 4);
+#if VARIANT == 5
+      wgrad2_v5_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
+#else
       wgrad2_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
+#endif
       long long n = (long long)M * N;
       cast_bf16_kernel<<<(int)((n + 255) / 256), 256>>>(dacc, dout, n);
     };
@@ -377,7 +527,7 @@ int main(int argc, char** argv) {
     printf("G2 VARIANT %d wgrad %dx%dx%lld S=%d: %.1f us  %.1f TF/s\n",
            VARIANT, M, N, K, S, us, tf);
 
-#if VARIANT >= 2
+#if VARIANT == 2 || VARIANT == 3
     printf("(ablation variant: results not checked)\n");
     continue;
 #endif
