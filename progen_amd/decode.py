@@ -379,7 +379,7 @@ def sample_cached(
 
     pads_seen = int((seq[:start_pos] == 0).sum())
     for curr_pos in range(start_pos, length):
-        logits_row = logits[0].float().cpu()
+        logits_row = logits[0].cpu().float()  # D2H then host cast: no device kernel between replays
         noise = R.gumbel_noise(logits_row.shape, generator=generator,
                                device=logits_row.device)
         if top_k is not None:
@@ -447,7 +447,7 @@ def sample_cached_batch(
             logits = graphed.step(seq[:, pos])
         else:
             logits = forward_step(model, seq[:, pos].to(dev), cache)  # (B, V)
-        rows = logits.float().cpu()
+        rows = logits.cpu().float()  # no device kernel between replays
         if greedy:
             nxt = rows.argmax(dim=-1)
         else:

@@ -57,12 +57,12 @@ def main():
         # between replays — a device argmax kernel would poison replay)
         for _ in range(args.warmup):
             logits = g.step(tok)
-            tok = logits.float().cpu().argmax(dim=-1)
+            tok = logits.cpu().float().argmax(dim=-1)
         torch.cuda.synchronize()
         t0 = time.perf_counter()
         for _ in range(args.tokens):
             logits = g.step(tok)
-            tok = logits.float().cpu().argmax(dim=-1)
+            tok = logits.cpu().float().argmax(dim=-1)
         torch.cuda.synchronize()
         mode = "graphed"
     else:
