@@ -103,6 +103,22 @@ def main():
             traceback.print_exc()
             print("[bench] hipGraph capture failed; eager fallback",
                   file=sys.stderr)
+        if world > 1:
+            # all ranks must agree on the mode: the graphed step's
+            # collective sequence (one whole-buffer all-reduce) differs
+            # from eager's bucketed one, so a mixed fleet would deadlock
+            # at the first step. MIN-reduce a success flag; any failure
+            # degrades every rank to eager (the
+            # graphed-DP opt-in must work or degrade CLEANLY).
+            flag = torch.tensor([1 if graphed is not None else 0],
+                                device=device)
+            torch.distributed.all_reduce(
+                flag, op=torch.distributed.ReduceOp.MIN)
+            if int(flag.item()) == 0 and graphed is not None:
+                graphed = None
+                import sys
+                print("[bench] a peer rank failed capture; all ranks "
+                      "running eager", file=sys.stderr)
 
     def step():
         if graphed is not None:
@@ -163,10 +179,19 @@ def main():
         }))
 
     # synchronized teardown (see train.py): a fast rank exiting early can
-    # SIGABRT a slower rank still inside process-group destruction
+    # SIGABRT a slower rank still inside process-group destruction; the
+    # result JSON is already printed, so teardown failures must not turn
+    # a measured run into a nonzero exit (world=8 first-shot hardening)
     if world > 1:
-        torch.distributed.barrier()
-        torch.distributed.destroy_process_group()
+        try:
+            torch.distributed.barrier()
+            torch.distributed.destroy_process_group()
+        except Exception:  # noqa: BLE001
+            import sys
+            import traceback
+            traceback.print_exc()
+            print("[bench] teardown error ignored (result already "
+                  "reported)", file=sys.stderr)
 
 
 if __name__ == "__main__":
