@@ -169,17 +169,24 @@ __global__ __launch_bounds__(BLOCK) void wgrad2_v5_kernel(
   extern __shared__ __attribute__((aligned(16))) char smem[];
 
   // per-lane row constants: rows r(ks,j) = ks*32 + l4*8 + (l15>>2) + 4j
-  // (j=0,1); swizzled col = base + 32*(blk ^ u(row)) per §file header
-  unsigned arow[2][2], brow[2][2], au[2][2];
+  // (j=0,1). The column offset must be XORed as a WHOLE with u(row)*32
+  // (bits 5..7): wc*128 carries bit 7, so it CANNOT be hoisted out of
+  // the XOR additively (the first V5 did, and B-frags with wc in {1,3}
+  // and u >= 4 read garbage — spot-check FAIL). arow/brow hold only the
+  // XOR-neutral parts (row base, bits 0..4 and >= 8-safe A term).
+  unsigned arowb[2][2], browb[2][2], au32[2][2];
+  const unsigned lp = (unsigned)((l15 & 3) * 8);
 #pragma unroll
   for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
     for (int j = 0; j < 2; ++j) {
       const int rr = ks * 32 + l4 * 8 + (l15 >> 2) + 4 * j;
-      au[ks][j] = (unsigned)uk(rr);
-      arow[ks][j] = (unsigned)(rr * 512 + wr * 256 + (l15 & 3) * 8);
-      brow[ks][j] = (unsigned)(32768 + rr * 512 + wc * 128 + (l15 & 3) * 8);
+      au32[ks][j] = (unsigned)(uk(rr) * 32);
+      arowb[ks][j] = (unsigned)(rr * 512);
+      browb[ks][j] = (unsigned)(32768 + rr * 512);
     }
+  const unsigned acol = (unsigned)(wr * 256) + lp;  // + mb*32, then ^u32
+  const unsigned bcol = (unsigned)(wc * 128) + lp;  // + nb*32, then ^u32
 
   f32x4 acc[8][4];
 #pragma unroll
@@ -205,12 +212,14 @@ __global__ __launch_bounds__(BLOCK) void wgrad2_v5_kernel(
       bf16x8 bfr[4], afr[8];
 #pragma unroll
       for (int nb = 0; nb < 4; ++nb)
-        bfr[nb] = frag_tr_asm(tb + brow[ks][0] + 32 * (nb ^ au[ks][0]),
-                              tb + brow[ks][1] + 32 * (nb ^ au[ks][1]));
+        bfr[nb] = frag_tr_asm(
+            tb + browb[ks][0] + ((bcol + 32 * nb) ^ au32[ks][0]),
+            tb + browb[ks][1] + ((bcol + 32 * nb) ^ au32[ks][1]));
 #pragma unroll
       for (int mb = 0; mb < 8; ++mb)
-        afr[mb] = frag_tr_asm(tb + arow[ks][0] + 32 * (mb ^ au[ks][0]),
-                              tb + arow[ks][1] + 32 * (mb ^ au[ks][1]));
+        afr[mb] = frag_tr_asm(
+            tb + arowb[ks][0] + ((acol + 32 * mb) ^ au32[ks][0]),
+            tb + arowb[ks][1] + ((acol + 32 * mb) ^ au32[ks][1]));
       asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
       __builtin_amdgcn_sched_barrier(0);  // §5.4 rule 18: pin MFMAs after
       __builtin_amdgcn_s_setprio(1);
