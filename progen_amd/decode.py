@@ -222,8 +222,9 @@ def forward_step_static(model: ProGenBase, token: torch.Tensor,
         keys = keys * kmask                             # zero halo keys
         vals = vals * kmask
         s_row = torch.einsum("bhd,bhnd->bhn", q, keys) * (DH ** -0.5)
-        s_row = torch.where(causal.view(1, 1, -1), s_row,
-                            torch.tensor(-1e30, dtype=s_row.dtype, device=dev))
+        # masked_fill with a python scalar: no host->device transfer, so
+        # the op is hipGraph-capturable (torch.tensor(scalar) is not)
+        s_row = s_row.masked_fill(~causal.view(1, 1, -1), -1e30)
         s_row = s_row - s_row.amax(dim=-1, keepdim=True)
         a = s_row.softmax(dim=-1)
         out = torch.einsum("bhn,bhnd->bhd", a, vals).reshape(B, H * DH)

@@ -87,8 +87,13 @@ typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4t;
 
 __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
     const short* __restrict__ dout, const short* __restrict__ qkv,
+    const short* __restrict__ halo,   // CP lookback for window 0 (see
+                                      // attention_fwd.hip) or nullptr
     const short* __restrict__ out, const float* __restrict__ lse,
     float* __restrict__ dacc, float* __restrict__ dlook,
+    float* __restrict__ dhalo,        // (B, wsz, 2*H*DH) fp32 window-0
+                                      // lookback grads (CP) or nullptr
+                                      // (quirk path: grads discarded)
     int B, int N, int H, int wsz) {
   const int window = blockIdx.x;
   const int head = blockIdx.y;
@@ -203,6 +208,12 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           const long long base = qkv_bn + (long long)kpos * HD3;
           kreg[u] = *(const bf16x8*)(qkv + base + k_off + su_d0[u]);
           vreg[u] = *(const bf16x8*)(qkv + base + v_off + su_d0[u]);
+        } else if (halo != nullptr) {
+          const long long hb =
+              ((long long)batch * wsz + (kpos + wsz)) * (2LL * H * DH);
+          kreg[u] = *(const bf16x8*)(halo + hb + head * DH + su_d0[u]);
+          vreg[u] = *(const bf16x8*)(halo + hb + (long long)H * DH +
+                                     head * DH + su_d0[u]);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
@@ -414,24 +425,33 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
+          float* dstv;
+          float* dstk;
           if (kpos >= 0) {
-            float* dstv = lookback
+            dstv = lookback
                 ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lv_off
                 : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
-            float* dstk = lookback
+            dstk = lookback
                 ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lk_off
                 : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
+          } else if (dhalo != nullptr) {
+            const long long hb =
+                ((long long)batch * wsz + (kpos + wsz)) * (2LL * H * DH);
+            dstv = dhalo + hb + lv_off;
+            dstk = dhalo + hb + lk_off;
+          } else {
+            continue;  // window-0 zero-pad quirk: grads discarded
+          }
 #pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              float vv = ((float*)&dv[n])[r];
-              float vk = ((float*)&dk[n])[r];
-              if (round > 0) {
-                vv += dstv[n * 16 + l15];
-                vk += dstk[n * 16 + l15];
-              }
-              dstv[n * 16 + l15] = vv;
-              dstk[n * 16 + l15] = vk;
+          for (int n = 0; n < 4; ++n) {
+            float vv = ((float*)&dv[n])[r];
+            float vk = ((float*)&dk[n])[r];
+            if (round > 0) {
+              vv += dstv[n * 16 + l15];
+              vk += dstk[n * 16 + l15];
             }
+            dstv[n * 16 + l15] = vv;
+            dstk[n * 16 + l15] = vk;
           }
         }
       }
@@ -504,15 +524,16 @@ __global__ __launch_bounds__(256) void attn_bwd_finalize_kernel(
 
 extern "C" {
 
-void attn_bwd_launch(const void* dout, const void* qkv, const float* rsin,
-                     const float* rcos, const void* out, const float* lse,
-                     float* dacc, float* dlook, void* dqkv, int B, int N,
+void attn_bwd_launch(const void* dout, const void* qkv, const void* halo,
+                     const float* rsin, const float* rcos, const void* out,
+                     const float* lse, float* dacc, float* dlook,
+                     float* dhalo, void* dqkv, int B, int N,
                      int H, int wsz, hipStream_t stream) {
   dim3 grid(N / wsz, H, B), block(ATTN_BLOCK);
   size_t lds = 24576 + 131072 + 2048;  // 154 KiB
   attn_bwd_kernel<<<grid, block, lds, stream>>>(
-      (const short*)dout, (const short*)qkv, (const short*)out, lse, dacc,
-      dlook, B, N, H, wsz);
+      (const short*)dout, (const short*)qkv, (const short*)halo,
+      (const short*)out, lse, dacc, dlook, dhalo, B, N, H, wsz);
   long long total = (long long)B * N * 3 * H * (DH / 8);
   int fin_grid = (int)((total + 255) / 256);
   if (fin_grid > 2048) fin_grid = 2048;

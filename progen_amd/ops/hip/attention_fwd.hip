@@ -52,6 +52,12 @@ __device__ __forceinline__ int swz(int row, int byte_in_row) {
 
 __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
     const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16, PRE-ROTATED
+    const short* __restrict__ halo,  // (B, wsz, 2*H*DH) rotated [k|v]
+                                     // lookback for window 0 (context
+                                     // parallelism: the previous rank's
+                                     // last window) or nullptr (rank 0 /
+                                     // single rank: the reference's
+                                     // zero-pad quirk, progen.py:90-96)
     short* __restrict__ out,         // (B, N, H*DH) bf16
     float* __restrict__ lse_out,     // (B, H, N)
     int B, int N, int H, int wsz) {
@@ -145,6 +151,12 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
           const long long base = qkv_bn + (long long)kpos * HD3;
           kreg[u] = *(const bf16x8*)(qkv + base + k_off + su_d0[u]);
           vreg[u] = *(const bf16x8*)(qkv + base + v_off + su_d0[u]);
+        } else if (halo != nullptr) {
+          const long long hb =
+              ((long long)batch * wsz + (kpos + wsz)) * (2LL * H * DH);
+          kreg[u] = *(const bf16x8*)(halo + hb + head * DH + su_d0[u]);
+          vreg[u] = *(const bf16x8*)(halo + hb + (long long)H * DH +
+                                     head * DH + su_d0[u]);
         } else {
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
@@ -335,14 +347,16 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
 
 extern "C" {
 
-void attn_fwd_launch(const void* qkv_rot, void* out, float* lse, int B, int N,
-                     int H, int wsz, hipStream_t stream) {
+void attn_fwd_launch(const void* qkv_rot, const void* halo, void* out,
+                     float* lse, int B, int N, int H, int wsz,
+                     hipStream_t stream) {
   const int sub_per_win = (wsz + ATTN_WAVES * QB - 1) / (ATTN_WAVES * QB);
   dim3 grid((N / wsz) * sub_per_win, H, B), block(ATTN_BLOCK);
   size_t lds = (size_t)(4 * KT * DH * 2) + (size_t)ATTN_WAVES * QB * KT * 2 +
                ATTN_WAVES * 2 * QB * 4;
   attn_fwd_kernel<<<grid, block, lds, stream>>>(
-      (const short*)qkv_rot, (short*)out, lse, B, N, H, wsz);
+      (const short*)qkv_rot, (const short*)halo, (short*)out, lse, B, N, H,
+      wsz);
 }
 
 }  // extern "C"
