@@ -94,37 +94,52 @@ def ln_shift_res(x: torch.Tensor, res, weight: torch.Tensor,
 
 class _LocalAttnFn(torch.autograd.Function):
     @staticmethod
-    def forward(ctx, qkv, sin, cos, heads: int, window_size: int):
+    def forward(ctx, qkv, sin, cos, heads: int, window_size: int, halo):
         C = dispatch.ext()
         # pre-rotation pass (rotary on q, k AND v — progen.py:87); the
         # attention kernels then stage pure bf16 copies
         qkv_rot = C.rope_qkv(qkv.contiguous(), sin, cos)
-        out, lse = C.attn_fwd(qkv_rot, int(heads), int(window_size))
-        ctx.save_for_backward(qkv_rot, sin, cos, out, lse)
+        out, lse = C.attn_fwd(qkv_rot, int(heads), int(window_size),
+                              halo=halo)
+        ctx.save_for_backward(qkv_rot, sin, cos, out, lse,
+                              *([halo] if halo is not None else []))
         ctx.heads = int(heads)
         ctx.window_size = int(window_size)
+        ctx.has_halo = halo is not None
         return out
 
     @staticmethod
     def backward(ctx, dout):
-        qkv_rot, sin, cos, out, lse = ctx.saved_tensors
+        qkv_rot, sin, cos, out, lse = ctx.saved_tensors[:5]
+        halo = ctx.saved_tensors[5] if ctx.has_halo else None
         C = dispatch.ext()
         # bwd finalize applies the inverse rotation (rotary is linear)
-        dqkv = C.attn_bwd(dout.contiguous(), qkv_rot, sin, cos, out, lse,
-                          ctx.heads, ctx.window_size)
-        return dqkv, None, None, None, None
+        res = C.attn_bwd(dout.contiguous(), qkv_rot, sin, cos, out, lse,
+                         ctx.heads, ctx.window_size, halo=halo)
+        dqkv = res[0]
+        # the halo holds PRE-ROTATED [k|v]: its grad stays in rotated
+        # space (the peer applies the inverse rotation after receiving
+        # it, with its own absolute positions — parallel/cp.py)
+        dhalo = res[1].to(dout.dtype) if ctx.has_halo else None
+        return dqkv, None, None, None, None, dhalo
 
 
 def local_attention(qkv: torch.Tensor, sin: torch.Tensor, cos: torch.Tensor,
-                    heads: int, window_size: int) -> torch.Tensor:
+                    heads: int, window_size: int,
+                    halo: torch.Tensor = None) -> torch.Tensor:
     """Fused windowed-causal attention core (reference: progen.py:83-103).
 
     Applies interleaved rotary to q, k AND v (quirk, progen.py:87), windows
     the sequence with one-window lookback (window 0's zero lookback keys
     UNMASKED, progen.py:90-96), runs online-softmax attention on MFMA and
-    returns the merged (B, N, h*dh) context."""
+    returns the merged (B, N, h*dh) context.
+
+    ``halo``: optional (B, wsz, 2*H*dh) ROTATED [k|v] band that replaces
+    window 0's zero lookback (context parallelism, parallel/cp.py); its
+    gradient (same shape/space) is returned to the autograd graph."""
     if dispatch.use_hip(qkv):
-        return _LocalAttnFn.apply(qkv, sin, cos, heads, window_size)
+        return _LocalAttnFn.apply(qkv, sin, cos, heads, window_size, halo)
+    assert halo is None, "halo is a kernel-path (GPU) feature"
     return reference.local_attention(qkv, sin, cos, heads, window_size)
 
 

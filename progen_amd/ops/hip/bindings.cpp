@@ -249,7 +249,8 @@ at::Tensor rope_qkv(const at::Tensor& qkv, const at::Tensor& rsin,
 }
 
 std::vector<at::Tensor> attn_fwd(const at::Tensor& qkv_rot, long heads,
-                                 long window) {
+                                 long window,
+                                 const c10::optional<at::Tensor>& halo) {
   TORCH_CHECK(qkv_rot.is_cuda() && qkv_rot.is_contiguous() && qkv_rot.dim() == 3);
   TORCH_CHECK(qkv_rot.scalar_type() == at::kBFloat16,
               "attn_fwd: bf16 only (MFMA path)");
@@ -258,17 +259,28 @@ std::vector<at::Tensor> attn_fwd(const at::Tensor& qkv_rot, long heads,
   TORCH_CHECK(qkv_rot.size(2) == 3LL * H * 64, "attn: dim_head must be 64");
   TORCH_CHECK(N % wsz == 0 && wsz % 64 == 0,
               "attn: seq divisible by window, window divisible by 64");
+  const void* halo_ptr = nullptr;
+  if (halo.has_value()) {
+    // context parallelism: the previous rank's last window of rotated
+    // [k|v] replaces window 0's zero lookback (parallel/cp.py)
+    TORCH_CHECK(halo->is_cuda() && halo->is_contiguous() &&
+                halo->scalar_type() == at::kBFloat16);
+    TORCH_CHECK(halo->dim() == 3 && halo->size(0) == B &&
+                halo->size(1) == wsz && halo->size(2) == 2LL * H * 64,
+                "halo must be (B, wsz, 2*H*64) rotated [k|v]");
+    halo_ptr = halo->data_ptr();
+  }
   auto out = at::empty({B, N, (long)H * 64}, qkv_rot.options());
   auto lse = at::empty({B, (long)H, N}, qkv_rot.options().dtype(at::kFloat));
-  attn_fwd_launch(qkv_rot.data_ptr(), out.data_ptr(), lse.data_ptr<float>(),
-                  B, N, H, wsz, cur_stream());
+  attn_fwd_launch(qkv_rot.data_ptr(), halo_ptr, out.data_ptr(),
+                  lse.data_ptr<float>(), B, N, H, wsz, cur_stream());
   return {out, lse};
 }
 
-at::Tensor attn_bwd(const at::Tensor& dout, const at::Tensor& qkv,
+std::vector<at::Tensor> attn_bwd(const at::Tensor& dout, const at::Tensor& qkv,
                     const at::Tensor& rsin, const at::Tensor& rcos,
                     const at::Tensor& out, const at::Tensor& lse, long heads,
-                    long window) {
+                    long window, const c10::optional<at::Tensor>& halo) {
   TORCH_CHECK(dout.is_cuda() && dout.is_contiguous());
   const int B = qkv.size(0), N = qkv.size(1);
   const int H = (int)heads, wsz = (int)window;
@@ -278,12 +290,26 @@ at::Tensor attn_bwd(const at::Tensor& dout, const at::Tensor& qkv,
   auto dlook = at::empty({(long)B, (long)N, 2L * H * 64},
                          qkv.options().dtype(at::kFloat));
   auto dqkv = at::empty_like(qkv);
-  attn_bwd_launch(dout.data_ptr(), qkv.data_ptr(), rsin.data_ptr<float>(),
-                  rcos.data_ptr<float>(), out.data_ptr(),
-                  lse.data_ptr<float>(), dacc.data_ptr<float>(),
-                  dlook.data_ptr<float>(), dqkv.data_ptr(), B, N, H, wsz,
-                  cur_stream());
-  return dqkv;
+  const void* halo_ptr = nullptr;
+  float* dhalo_ptr = nullptr;
+  at::Tensor dhalo;
+  if (halo.has_value()) {
+    TORCH_CHECK(halo->is_cuda() && halo->is_contiguous() &&
+                halo->scalar_type() == at::kBFloat16);
+    halo_ptr = halo->data_ptr();
+    // window-0 slices write every element of their bands (plain
+    // stores), so no zeroing needed here either
+    dhalo = at::empty({(long)B, (long)wsz, 2L * H * 64},
+                      qkv.options().dtype(at::kFloat));
+    dhalo_ptr = dhalo.data_ptr<float>();
+  }
+  attn_bwd_launch(dout.data_ptr(), qkv.data_ptr(), halo_ptr,
+                  rsin.data_ptr<float>(), rcos.data_ptr<float>(),
+                  out.data_ptr(), lse.data_ptr<float>(),
+                  dacc.data_ptr<float>(), dlook.data_ptr<float>(), dhalo_ptr,
+                  dqkv.data_ptr(), B, N, H, wsz, cur_stream());
+  if (halo.has_value()) return {dqkv, dhalo};
+  return {dqkv};
 }
 
 std::vector<at::Tensor> sgu_fwd(const at::Tensor& xa, const at::Tensor& g_ln,
@@ -400,7 +426,12 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("step_dev"), py::arg("grad_scale"), py::arg("clip_coef"),
         py::arg("shard_off") = 0);
   m.def("rope_qkv", &rope_qkv, "pre-rotation of qkv (rotary on q,k,v)");
-  m.def("attn_fwd", &attn_fwd, "fused local attention forward");
-  m.def("attn_bwd", &attn_bwd, "fused local attention backward");
+  m.def("attn_fwd", &attn_fwd, "fused local attention forward",
+        py::arg("qkv_rot"), py::arg("heads"), py::arg("window"),
+        py::arg("halo") = py::none());
+  m.def("attn_bwd", &attn_bwd, "fused local attention backward",
+        py::arg("dout"), py::arg("qkv"), py::arg("rsin"), py::arg("rcos"),
+        py::arg("out"), py::arg("lse"), py::arg("heads"), py::arg("window"),
+        py::arg("halo") = py::none());
   m.def("crc32c", &crc32c_host, "CRC-32C (slicing-by-8, host)");
 }

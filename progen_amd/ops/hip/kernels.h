@@ -42,12 +42,14 @@ void fused_adamw_launch(float* master, void* params, const void* grads,
 
 void rope_qkv_launch(const void* qkv, const float* rsin, const float* rcos,
                      void* qkv_rot, int B, int N, int H, hipStream_t stream);
-void attn_fwd_launch(const void* qkv_rot, void* out, float* lse, int B, int N,
-                     int H, int wsz, hipStream_t stream);
-void attn_bwd_launch(const void* dout, const void* qkv, const float* rsin,
-                     const float* rcos, const void* out, const float* lse,
-                     float* dacc, float* dlook, void* dqkv, int B, int N,
-                     int H, int wsz, hipStream_t stream);
+void attn_fwd_launch(const void* qkv_rot, const void* halo, void* out,
+                     float* lse, int B, int N, int H, int wsz,
+                     hipStream_t stream);
+void attn_bwd_launch(const void* dout, const void* qkv, const void* halo,
+                     const float* rsin, const float* rcos, const void* out,
+                     const float* lse, float* dacc, float* dlook,
+                     float* dhalo, void* dqkv, int B, int N, int H, int wsz,
+                     hipStream_t stream);
 
 void sgu_fwd_launch(const void* xa, const void* g_ln, const void* w,
                     const float* bias, void* out, void* gate_out, int B,

@@ -38,6 +38,13 @@
 // VARIANT 3: ABLATION — staging+sync only, MFMAs skipped with the
 //            fragments kept alive via asm (wrong results; isolates the
 //            staging/HBM side).
+// VARIANT 7: VARIANT 6 + ONE barrier per K-step (the counted
+//            vmcnt(4) moves BEFORE the barrier, so the same barrier
+//            orders both "reads of t done" and "tile t+1 landed in
+//            every wave"; B(t+2)'s glds issue after it) + split
+//            read/MFMA phases per ks: the A mb4-7 tr reads issue
+//            before the mb0-3 MFMA cluster and land under it
+//            (lgkmcnt(8) leaves exactly them outstanding).
 // VARIANT 6: VARIANT 5 + A-operand 3-slot LDS ring (96 KiB A + 64 KiB
 //            B = 160 KiB, the full CU LDS): slot (t+2)%3 is free
 //            DURING compute of step t (its readers finished at step
@@ -127,7 +134,7 @@ __device__ __forceinline__ bf16x8 frag_tr(const char* img, int kb, int m0,
   return o;
 }
 
-#if VARIANT == 5 || VARIANT == 6
+#if VARIANT >= 5
 // one 8-bf16 MFMA A/B fragment from two inline-asm tr reads at integer
 // LDS byte offsets (compiler cannot see these as LDS reads)
 union FragU {
@@ -270,6 +277,147 @@ __global__ __launch_bounds__(BLOCK) void wgrad2_v5_kernel(
 }
 #endif  // VARIANT == 5
 
+
+
+#if VARIANT == 7
+__global__ __launch_bounds__(BLOCK) void wgrad2_v7_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    float* __restrict__ dw_acc, int M, int N, long long K, int splits) {
+  const int ntiles_n = N / BN;
+  const int nwg = (M / BM) * ntiles_n;
+  const int orig = blockIdx.x;
+  const int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+               + orig / 8;
+  const int tm = wg / ntiles_n, tn = wg % ntiles_n;
+
+  const long long kchunk = K / splits;
+  const long long k_lo = blockIdx.z * kchunk;
+  const long long nsteps = kchunk / BK;
+
+  const int lane = (int)threadIdx.x & 63;
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+  const int wid = (int)threadIdx.x >> 6;
+  const int wr = wid >> 2;
+  const int wc = wid & 3;
+
+  const long long lda = 2LL * M;
+  const long long ldb2 = 2LL * N;
+  const char* asrc = (const char*)dy + k_lo * lda + (long long)tm * 512;
+  const char* bsrc = (const char*)x + k_lo * ldb2 + (long long)tn * 512;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  const unsigned ABASE[3] = {0u, 32768u, 65536u};
+  const unsigned BBASE[2] = {98304u, 131072u};
+
+  unsigned arowb[2][2], brow_r[2][2], au32[2][2];
+  const unsigned lp = (unsigned)((l15 & 3) * 8);
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int rr = ks * 32 + l4 * 8 + (l15 >> 2) + 4 * j;
+      au32[ks][j] = (unsigned)(uk(rr) * 32);
+      arowb[ks][j] = (unsigned)(rr * 512);
+      brow_r[ks][j] = (unsigned)(rr * 512);
+    }
+  const unsigned acol = (unsigned)(wr * 256) + lp;
+  const unsigned bcol = (unsigned)(wc * 128) + lp;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  {
+    stage_glds(asrc, lda, smem + ABASE[0]);
+    stage_glds(bsrc, ldb2, smem + BBASE[0]);
+    if (nsteps > 1) {
+      stage_glds(asrc + BK * lda, lda, smem + ABASE[1]);
+      stage_glds(bsrc + BK * ldb2, ldb2, smem + BBASE[1]);
+    }
+  }
+  asm volatile("s_waitcnt vmcnt(8)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (long long t = 0; t < nsteps; ++t) {
+    const unsigned ta = ABASE[t % 3];
+    const unsigned tbb = BBASE[t & 1];
+#pragma unroll
+    for (int ks = 0; ks < 2; ++ks) {
+      bf16x8 bfr[4], afr[8];
+#pragma unroll
+      for (int nb = 0; nb < 4; ++nb)
+        bfr[nb] = frag_tr_asm(
+            tbb + brow_r[ks][0] + ((bcol + 32 * nb) ^ au32[ks][0]),
+            tbb + brow_r[ks][1] + ((bcol + 32 * nb) ^ au32[ks][1]));
+#pragma unroll
+      for (int mb = 0; mb < 4; ++mb)
+        afr[mb] = frag_tr_asm(
+            ta + arowb[ks][0] + ((acol + 32 * mb) ^ au32[ks][0]),
+            ta + arowb[ks][1] + ((acol + 32 * mb) ^ au32[ks][1]));
+      // issue the second A half now; it lands under the first cluster
+#pragma unroll
+      for (int mb = 4; mb < 8; ++mb)
+        afr[mb] = frag_tr_asm(
+            ta + arowb[ks][0] + ((acol + 32 * mb) ^ au32[ks][0]),
+            ta + arowb[ks][1] + ((acol + 32 * mb) ^ au32[ks][1]));
+      asm volatile("s_waitcnt lgkmcnt(8)" ::: "memory");  // halves 0-3 in
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mb = 0; mb < 4; ++mb)
+#pragma unroll
+        for (int nb = 0; nb < 4; ++nb)
+          acc[mb][nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[mb], bfr[nb], acc[mb][nb], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      if (ks == 0 && t + 2 < nsteps)
+        stage_glds(asrc + (t + 2) * BK * lda, lda,
+                   smem + ABASE[(t + 2) % 3]);
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");  // halves 4-7 in
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mb = 4; mb < 8; ++mb)
+#pragma unroll
+        for (int nb = 0; nb < 4; ++nb)
+          acc[mb][nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[mb], bfr[nb], acc[mb][nb], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+    }
+    // single barrier: per-wave vmcnt(4) first (A(t+1)/B(t+1) landed;
+    // A(t+2)'s 4 glds stay in flight), then the barrier orders BOTH
+    // "everyone done reading t" and "t+1 landed everywhere"
+    if (t + 2 < nsteps) {
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");
+    } else {
+      asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+    }
+    __builtin_amdgcn_s_barrier();
+    if (t + 2 < nsteps)
+      stage_glds(bsrc + (t + 2) * BK * ldb2, ldb2, smem + BBASE[t & 1]);
+  }
+
+  const long long mbase = (long long)tm * BM + wr * 128;
+  const long long nbase = (long long)tn * BN + wc * 64;
+#pragma unroll
+  for (int mb = 0; mb < 8; ++mb)
+#pragma unroll
+    for (int r4 = 0; r4 < 4; ++r4) {
+      const long long m = mbase + mb * 16 + l4 * 4 + r4;
+#pragma unroll
+      for (int nb = 0; nb < 4; ++nb) {
+        const long long n = nbase + nb * 16 + l15;
+        const float v = ((float*)&acc[mb][nb])[r4];
+        if (splits > 1) atomicAdd(dw_acc + m * N + n, v);
+        else dw_acc[m * N + n] = v;
+      }
+    }
+}
+#endif  // VARIANT == 7
 
 #if VARIANT == 6
 __global__ __launch_bounds__(BLOCK) void wgrad2_v6_kernel(
@@ -628,7 +776,7 @@ int main(int argc, char** argv) {
   hipMemcpy(da, ha.data(), (size_t)K * M * 2, hipMemcpyHostToDevice);
   hipMemcpy(db, hb.data(), (size_t)K * N * 2, hipMemcpyHostToDevice);
 
-#if VARIANT == 6
+#if VARIANT == 6 || VARIANT == 7
   const size_t lds = 163840;  // 96 KiB A ring + 64 KiB B pair (full CU LDS)
 #else
   const size_t lds = 2 * 65536;  // 128 KiB
@@ -648,6 +796,8 @@ int main(int argc, char** argv) {
       wgrad2_v5_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
 #elif VARIANT == 6
       wgrad2_v6_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
+#elif VARIANT == 7
+      wgrad2_v7_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
 #else
       wgrad2_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
 #endif
