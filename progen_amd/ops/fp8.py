@@ -31,11 +31,26 @@ def amax_scale(t: torch.Tensor, margin: float = 1.0) -> torch.Tensor:
     return torch.where(amax > 0, s, torch.ones_like(s))
 
 
+def _ext_or_none():
+    try:
+        from progen_amd import _C  # noqa: WPS433
+        return _C
+    except Exception:  # noqa: BLE001
+        return None
+
+
 def quantize_e4m3(t: torch.Tensor,
                   scale: torch.Tensor = None) -> Tuple[torch.Tensor, torch.Tensor]:
-    """t -> (e4m3 tensor, fp32 scale) with t ≈ fp8 * scale."""
+    """t -> (e4m3 tensor, fp32 scale) with t ≈ fp8 * scale. On GPU the
+    scale+cast is ONE fused kernel pass (ops/hip/fp8_quant.hip —
+    v_cvt_pk_fp8_f32 saturates, so no separate clamp; the eager chain's
+    fp32 intermediate copy is gone)."""
     if scale is None:
         scale = amax_scale(t)
+    C = _ext_or_none() if (t.is_cuda and t.dtype == torch.bfloat16
+                           and t.numel() % 8 == 0) else None
+    if C is not None:
+        return C.fp8_quantize(t.contiguous(), scale.reshape(1)), scale
     q = (t.float() / scale).clamp(-E4M3_MAX, E4M3_MAX)
     return q.to(torch.float8_e4m3fn), scale
 
@@ -112,9 +127,16 @@ class _Fp8LinearFn(torch.autograd.Function):
         x2, weight = ctx.saved_tensors
         dy2 = dy.reshape(-1, dy.shape[-1]).contiguous()
         # dgrad in fp8: dy (M,N) @ W (N,K); _scaled_mm wants the second
-        # operand column-major, so quantize W^T and transpose the fp8
+        # operand column-major — the fused transpose-quantize kernel
+        # emits (K,N) e4m3 directly from the (N,K) bf16 weight (one
+        # LDS-tiled pass instead of a bf16 transpose copy + 4-pass cast)
         qdy, sdy = quantize_e4m3(dy2)
-        qwt, swt = quantize_e4m3(weight.t().contiguous())
+        C = _ext_or_none() if weight.is_cuda else None
+        if C is not None and weight.dtype == torch.bfloat16:
+            swt = amax_scale(weight)
+            qwt = C.fp8_quantize_t(weight.contiguous(), swt.reshape(1))
+        else:
+            qwt, swt = quantize_e4m3(weight.t().contiguous())
         dx = torch._scaled_mm(qdy, qwt.t(), scale_a=sdy, scale_b=swt,
                               out_dtype=dy.dtype)
         # wgrad in bf16 (library GEMM)

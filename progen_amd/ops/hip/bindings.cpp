@@ -401,6 +401,28 @@ uint32_t crc32c_host(py::bytes data) {
   return c ^ 0xFFFFFFFFu;
 }
 
+at::Tensor fp8_quantize(const at::Tensor& t, const at::Tensor& scale) {
+  TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
+              t.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(t.numel() % 8 == 0, "fp8_quantize: numel % 8");
+  TORCH_CHECK(scale.is_cuda() && scale.scalar_type() == at::kFloat);
+  auto out = at::empty_like(t, t.options().dtype(at::kFloat8_e4m3fn));
+  quant_e4m3_launch(t.data_ptr(), out.data_ptr(), scale.data_ptr<float>(),
+                    t.numel(), cur_stream());
+  return out;
+}
+
+at::Tensor fp8_quantize_t(const at::Tensor& w, const at::Tensor& scale) {
+  TORCH_CHECK(w.is_cuda() && w.is_contiguous() && w.dim() == 2 &&
+              w.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(w.size(0) % 2 == 0, "fp8_quantize_t: rows % 2");
+  auto out = at::empty({w.size(1), w.size(0)},
+                       w.options().dtype(at::kFloat8_e4m3fn));
+  quant_e4m3_t_launch(w.data_ptr(), out.data_ptr(), scale.data_ptr<float>(),
+                      (int)w.size(0), (int)w.size(1), cur_stream());
+  return out;
+}
+
 PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
   m.def("sgu_fwd", &sgu_fwd, "SGU causal spatial matmul forward");
   m.def("sgu_dgate", &sgu_dgate, "SGU backward: dgate");
@@ -434,4 +456,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out"), py::arg("lse"), py::arg("heads"), py::arg("window"),
         py::arg("halo") = py::none());
   m.def("crc32c", &crc32c_host, "CRC-32C (slicing-by-8, host)");
+  m.def("fp8_quantize", &fp8_quantize, "fused bf16 -> e4m3 at device scale");
+  m.def("fp8_quantize_t", &fp8_quantize_t,
+        "fused transpose-quantize (N,K) bf16 -> (K,N) e4m3");
 }

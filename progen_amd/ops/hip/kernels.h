@@ -40,6 +40,11 @@ void fused_adamw_launch(float* master, void* params, const void* grads,
                         const float* clip_coef, bool is_bf16,
                         long long shard_off, hipStream_t stream);
 
+void quant_e4m3_launch(const void* in, void* out, const float* scale,
+                       long long numel, hipStream_t stream);
+void quant_e4m3_t_launch(const void* in, void* out, const float* scale,
+                         int Nr, int Kc, hipStream_t stream);
+
 void rope_qkv_launch(const void* qkv, const float* rsin, const float* rcos,
                      void* qkv_rot, int B, int N, int H, hipStream_t stream);
 void attn_fwd_launch(const void* qkv_rot, const void* halo, void* out,

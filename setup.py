@@ -24,6 +24,7 @@ HIP_SOURCES = [
     "progen_amd/ops/hip/attention_fwd.hip",
     "progen_amd/ops/hip/attention_bwd.hip",
     "progen_amd/ops/hip/sgu.hip",
+    "progen_amd/ops/hip/fp8_quant.hip",
 ]
 
 setup(
