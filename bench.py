@@ -35,6 +35,11 @@ CONFIGS = {
                          global_mlp_depth=2),
     "tiny-cpu": dict(num_tokens=256, dim=128, depth=2, heads=2, dim_head=64,
                      window_size=64, seq_len=256, global_mlp_depth=1),
+    # BASELINE config #5 vehicle (TP=8 target; single-GPU bench runs it
+    # data-parallel-1 at a small batch, optionally with --fp8)
+    "progen-6b": dict(num_tokens=256, dim=4096, depth=24, heads=64,
+                      dim_head=64, window_size=512, seq_len=2048,
+                      ff_glu=True, global_mlp_depth=2),
 }
 
 
@@ -46,7 +51,15 @@ def main():
     p.add_argument("--model", default="progen-1.2b", choices=list(CONFIGS))
     p.add_argument("--batch", type=int, default=64, help="per-GPU batch size")
     p.add_argument("--dtype", default="bf16", choices=["bf16", "fp32"])
+    p.add_argument("--fp8", action="store_true",
+                   help="route big projections through e4m3 hipBLASLt "
+                        "GEMMs (fwd+dgrad; wgrad stays bf16) — the "
+                        "BASELINE #5 fp8 mode; reported dtype becomes "
+                        "bf16+fp8")
     args = p.parse_args()
+    if args.fp8:
+        from progen_amd.ops import fp8 as _fp8
+        _fp8.ENABLED = True
 
     from progen_amd.tuning import enable_tuned_gemms
     enable_tuned_gemms()
@@ -168,7 +181,8 @@ def main():
             "higher_is_better": True,
             "scaling": "weak",
             "vs_baseline": None,
-            "dtype": "bf16" if dtype == torch.bfloat16 else "fp32",
+            "dtype": ("bf16+fp8" if args.fp8 else
+                      "bf16" if dtype == torch.bfloat16 else "fp32"),
             "data": "synthetic",
             "config": {
                 "model": "ProGen-1.2B" if args.model == "progen-1.2b" else args.model,
