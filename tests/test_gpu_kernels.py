@@ -259,8 +259,11 @@ def test_fused_adamw_shard_off_matches_full():
     m_a.rotary_cos = m_a.rotary_cos.float()
     m_b = copy.deepcopy(m_a)
 
-    o_full = ProGenAdamW(m_a, lr=1e-3, max_grad_norm=0.5)
-    o_shard = ProGenAdamW(m_b, lr=1e-3, max_grad_norm=0.5)
+    # no grad clipping: the grad-norm reduce is atomicAdd-ordered and
+    # its last-ulp nondeterminism would break the bitwise comparison
+    # (clip correctness is covered by test_fused_adamw_matches_eager)
+    o_full = ProGenAdamW(m_a, lr=1e-3, max_grad_norm=None)
+    o_shard = ProGenAdamW(m_b, lr=1e-3, max_grad_norm=None)
     torch.manual_seed(18)
     fake = torch.randn(o_full.space.flat_grad.shape, device=dev())
     o_full.space.flat_grad.copy_(fake.to(o_full.space.flat_grad.dtype))
@@ -272,8 +275,7 @@ def test_fused_adamw_shard_off_matches_full():
     C = dispatch.ext()
     n = o_shard.space.numel
     lo_hi = [(0, n // 2), (n // 2, n)]
-    norm = C.grad_sumsq(o_shard.space.flat_grad).sqrt_()
-    clip = 0.5 / torch.clamp_min(norm, 0.5)
+    clip = torch.ones(1, device=dev())
     o_shard.step_count += 1
     for i, (lo, hi) in enumerate(lo_hi):
         starts = o_shard.chunk_starts.clamp(min=lo, max=hi)

@@ -38,6 +38,13 @@
 // VARIANT 3: ABLATION — staging+sync only, MFMAs skipped with the
 //            fragments kept alive via asm (wrong results; isolates the
 //            staging/HBM side).
+// VARIANT 8: the 8-phase template's fine interleave (guide: 'the
+//            per-phase interleave is the lever', -7..27% without it).
+//            Each K-step = 4 phases of [16 tr reads | 2 glds of one
+//            half of tile t+1 | raw barrier | lgkmcnt(0) | 16 MFMA];
+//            staging order A0,B0,A1,B1 matches the next step's read
+//            order so counted vmcnt(4) at phases 0 and 2 suffices.
+//            2 full buffers (ping-pong), stage depth 1 tile.
 // VARIANT 7: VARIANT 6 + ONE barrier per K-step (the counted
 //            vmcnt(4) moves BEFORE the barrier, so the same barrier
 //            orders both "reads of t done" and "tile t+1 landed in
@@ -278,6 +285,160 @@ __global__ __launch_bounds__(BLOCK) void wgrad2_v5_kernel(
 #endif  // VARIANT == 5
 
 
+
+
+#if VARIANT == 8
+// stage HALF a slab (32 k-rows): 2 wave-instructions per wave
+__device__ __forceinline__ void stage_glds_half(const char* __restrict__ src,
+                                                long long ldb, char* img,
+                                                int khalf) {
+  const int lane = (int)threadIdx.x & 63;
+  const int wid = (int)threadIdx.x >> 6;
+#pragma unroll
+  for (int p = 0; p < 2; ++p) {
+    const int inst = khalf * 16 + p * NWAVES + wid;
+    const int krow = inst * 2 + (lane >> 5);
+    const int colb = ((lane & 31) * 16) ^ (uk(krow) * 32);
+    __builtin_amdgcn_global_load_lds(
+        (const AS1 unsigned int*)(src + (long long)krow * ldb + colb),
+        (AS3 unsigned int*)(img + inst * 1024), 16, 0, 0);
+  }
+}
+
+__global__ __launch_bounds__(BLOCK) void wgrad2_v8_kernel(
+    const short* __restrict__ dy, const short* __restrict__ x,
+    float* __restrict__ dw_acc, int M, int N, long long K, int splits) {
+  const int ntiles_n = N / BN;
+  const int nwg = (M / BM) * ntiles_n;
+  const int orig = blockIdx.x;
+  const int xcd = orig % 8, q = nwg / 8, r = nwg % 8;
+  const int wg = (xcd < r ? xcd * (q + 1) : r * (q + 1) + (xcd - r) * q)
+               + orig / 8;
+  const int tm = wg / ntiles_n, tn = wg % ntiles_n;
+
+  const long long kchunk = K / splits;
+  const long long k_lo = blockIdx.z * kchunk;
+  const long long nsteps = kchunk / BK;
+
+  const int lane = (int)threadIdx.x & 63;
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+  const int wid = (int)threadIdx.x >> 6;
+  const int wr = wid >> 2;
+  const int wc = wid & 3;
+
+  const long long lda = 2LL * M;
+  const long long ldb2 = 2LL * N;
+  const char* asrc = (const char*)dy + k_lo * lda + (long long)tm * 512;
+  const char* bsrc = (const char*)x + k_lo * ldb2 + (long long)tn * 512;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  // [buf][A 32K | B 32K]
+
+  unsigned arowb[2][2], brow_r[2][2], au32[2][2];
+  const unsigned lp = (unsigned)((l15 & 3) * 8);
+#pragma unroll
+  for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+    for (int j = 0; j < 2; ++j) {
+      const int rr = ks * 32 + l4 * 8 + (l15 >> 2) + 4 * j;
+      au32[ks][j] = (unsigned)(uk(rr) * 32);
+      arowb[ks][j] = (unsigned)(rr * 512);
+      brow_r[ks][j] = (unsigned)(32768 + rr * 512);
+    }
+  const unsigned acol = (unsigned)(wr * 256) + lp;
+  const unsigned bcol = (unsigned)(wc * 128) + lp;
+
+  f32x4 acc[8][4];
+#pragma unroll
+  for (int i = 0; i < 8; ++i)
+#pragma unroll
+    for (int j = 0; j < 4; ++j) acc[i][j] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+  {
+    stage_glds(asrc, lda, smem);
+    stage_glds(bsrc, ldb2, smem + 32768);
+  }
+  asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+  __builtin_amdgcn_s_barrier();
+
+  for (long long t = 0; t < nsteps; ++t) {
+    const unsigned tb = (unsigned)((t & 1) * 65536);
+    char* nxt = smem + ((t + 1) & 1) * 65536;
+    const char* an = asrc + (t + 1) * BK * lda;
+    const char* bn = bsrc + (t + 1) * BK * ldb2;
+    bf16x8 bfr[4];
+#pragma unroll
+    for (int p = 0; p < 4; ++p) {
+      const int ks = p >> 1, mh = p & 1;
+      bf16x8 afr[4];
+      if (mh == 0) {
+#pragma unroll
+        for (int nb = 0; nb < 4; ++nb)
+          bfr[nb] = frag_tr_asm(
+              tb + brow_r[ks][0] + ((bcol + 32 * nb) ^ au32[ks][0]),
+              tb + brow_r[ks][1] + ((bcol + 32 * nb) ^ au32[ks][1]));
+      }
+#pragma unroll
+      for (int mb = 0; mb < 4; ++mb) {
+        const int mbb = mh * 4 + mb;
+        afr[mb] = frag_tr_asm(
+            tb + arowb[ks][0] + ((acol + 32 * mbb) ^ au32[ks][0]),
+            tb + arowb[ks][1] + ((acol + 32 * mbb) ^ au32[ks][1]));
+      }
+      // stage one half of tile t+1 (order A0,B0,A1,B1 = read order)
+      if (t + 1 < nsteps) {
+        if (p == 0) stage_glds_half(an, lda, nxt, 0);
+        else if (p == 1) stage_glds_half(bn, ldb2, nxt + 32768, 0);
+        else if (p == 2) stage_glds_half(an, lda, nxt, 1);
+        else stage_glds_half(bn, ldb2, nxt + 32768, 1);
+      }
+      if (p == 2) {
+        // phase 2 reads A1/B1 of tile t (staged at t-1's phases 2-3);
+        // newer in flight: A0,B0,A1 of t+1 = 6 instructions
+        if (t + 1 < nsteps)
+          asm volatile("s_waitcnt vmcnt(6)" ::: "memory");
+        else
+          asm volatile("s_waitcnt vmcnt(0)" ::: "memory");
+      }
+      __builtin_amdgcn_s_barrier();
+      asm volatile("s_waitcnt lgkmcnt(0)" ::: "memory");
+      __builtin_amdgcn_sched_barrier(0);
+      __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+      for (int mb = 0; mb < 4; ++mb)
+#pragma unroll
+        for (int nb = 0; nb < 4; ++nb)
+          acc[mh * 4 + mb][nb] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+              afr[mb], bfr[nb], acc[mh * 4 + mb][nb], 0, 0, 0);
+      __builtin_amdgcn_s_setprio(0);
+      __builtin_amdgcn_s_barrier();
+    }
+    // next step's phase 0 reads A0/B0 of t+1 — the FIRST 4 glds issued
+    // this step; its A1/B1 (4 newer) stay in flight across the barrier
+    // and are landed by the phase-2 vmcnt(6) above
+    if (t + 1 < nsteps)
+      asm volatile("s_waitcnt vmcnt(4)" ::: "memory");  // A0,B0 landed
+    __builtin_amdgcn_s_barrier();
+  }
+
+  const long long mbase = (long long)tm * BM + wr * 128;
+  const long long nbase = (long long)tn * BN + wc * 64;
+#pragma unroll
+  for (int mb = 0; mb < 8; ++mb)
+#pragma unroll
+    for (int r4 = 0; r4 < 4; ++r4) {
+      const long long m = mbase + mb * 16 + l4 * 4 + r4;
+#pragma unroll
+      for (int nb = 0; nb < 4; ++nb) {
+        const long long n = nbase + nb * 16 + l15;
+        const float v = ((float*)&acc[mb][nb])[r4];
+        if (splits > 1) atomicAdd(dw_acc + m * N + n, v);
+        else dw_acc[m * N + n] = v;
+      }
+    }
+}
+#endif  // VARIANT == 8
 
 #if VARIANT == 7
 __global__ __launch_bounds__(BLOCK) void wgrad2_v7_kernel(
@@ -798,6 +959,8 @@ int main(int argc, char** argv) {
       wgrad2_v6_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
 #elif VARIANT == 7
       wgrad2_v7_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
+#elif VARIANT == 8
+      wgrad2_v8_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
 #else
       wgrad2_kernel<<<grid, block, lds>>>(da, db, dacc, M, N, K, S);
 #endif
