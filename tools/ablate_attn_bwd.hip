@@ -882,6 +882,400 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_v4_kernel(
   }
 }
 
+
+// ---------------------------------------------------------------------------
+// VARIANT 5: 8-wave geometry — 2 waves/SIMD. V4's remaining PMC profile
+// is 47.8% barrier/wait-parked at ONE wave/SIMD (154 KiB LDS = 1 block
+// /CU, 501 VGPRs = 1 wave/SIMD): nothing covers a parked wave. Here a
+// block is 8 waves of 32-row chunks (per-wave registers halve: s/dp/
+// dqacc are [2][4]), so each SIMD holds TWO waves that cover each
+// other's stalls. Wave pairs share the 64-row chunk regions (qt/dot/
+// pds/ds2 keep their 128-B-row layouts and swizzles); the dV/dK key
+// slices split the MFMA K-dim by ks-half across wave halves, the upper
+// half writing to separate dacc2/dlook2 accumulators (plain stores
+// stay raceless; the finalize pass sums four sources instead of two).
+// ---------------------------------------------------------------------------
+
+#define V5_WAVES 8
+#define V5_BLOCK (V5_WAVES * WAVE)
+
+__global__ __launch_bounds__(V5_BLOCK) void attn_bwd_v5_kernel(
+    const short* __restrict__ dout, const short* __restrict__ qkv,
+    const short* __restrict__ out, const float* __restrict__ lse,
+    float* __restrict__ dacc, float* __restrict__ dlook,
+    float* __restrict__ dacc2, float* __restrict__ dlook2,
+    int B, int N, int H, int wsz) {
+  const int window = blockIdx.x;
+  const int head = blockIdx.y;
+  const int batch = blockIdx.z;
+
+  const int lane = threadIdx.x % WAVE;
+  const int wid = threadIdx.x / WAVE;   // 0..7
+  const int l15 = lane & 15;
+  const int l4 = lane >> 4;
+  const int pid = wid >> 1;             // pair region 0..3
+  const int phalf = wid & 1;            // row half within the pair
+
+  const long long HD3 = 3LL * H * DH;
+  const long long HD = (long long)H * DH;
+  const long long HD2 = 2LL * H * DH;
+  const long long qkv_bn = (long long)batch * N * HD3;
+  const long long o_bn = (long long)batch * N * HD;
+  const int q_off = head * DH;
+  const int k_off = H * DH + head * DH;
+  const int v_off = 2 * H * DH + head * DH;
+  const long long look_bn = (long long)batch * N * HD2;
+  const int lk_off = head * DH;
+  const int lv_off = H * DH + head * DH;
+
+  extern __shared__ __attribute__((aligned(16))) char smem[];
+  char* k_lds = smem;
+  char* kt_lds = smem + 8192;
+  char* v_lds = smem + 16384;
+  char* qt_base = smem + 24576;
+  char* dot_base = qt_base + 4 * 8192;
+  char* pds_base = dot_base + 4 * 8192;
+  char* ds2_base = pds_base + 4 * 8192;
+  float* d_base = (float*)(ds2_base + 4 * 8192);        // [pair][64]
+  float* lse_base = (float*)(ds2_base + 4 * 8192 + 1024);
+
+  char* qt_lds = qt_base + pid * 8192;
+  char* dot_lds = dot_base + pid * 8192;
+  char* pds_lds = pds_base + pid * 8192;
+  char* ds2_lds = ds2_base + pid * 8192;
+  float* d_lds = d_base + pid * 64;
+  float* lse_lds = lse_base + pid * 64;
+
+  const float scale = rsqrtf((float)DH);
+  const int tiles = 2 * wsz / KT;
+  const int chunks64 = wsz / 64;                 // pair-sized chunks
+  const int rounds = (chunks64 + 3) / 4;
+
+  // block-wide k/v tile staging: 512 threads cover 64 keys x 64 dh
+  const int su_key = (int)threadIdx.x >> 3;
+  const int su_d0 = ((int)threadIdx.x & 7) * 8;
+
+  for (int round = 0; round < rounds; ++round) {
+    const int c64 = round * 4 + pid;             // this pair's 64-chunk
+    const bool active = c64 < chunks64;
+    const int nactive = min(4, chunks64 - round * 4);
+    const int chunk_off = c64 * 64 + phalf * 32; // rows-in-window
+    const int q0 = window * wsz + chunk_off;
+    const int colbase = phalf * 32;              // in the pair region
+
+    bf16x8 qfrag[2][2];
+    f32x4 dqacc[2][4];
+#pragma unroll
+    for (int m = 0; m < 2; ++m)
+#pragma unroll
+      for (int d = 0; d < 4; ++d) dqacc[m][d] = (f32x4){0.f, 0.f, 0.f, 0.f};
+
+    if (active) {
+#pragma unroll
+      for (int m = 0; m < 2; ++m) {
+        const int row = q0 + m * 16 + l15;
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks) {
+          const int d0 = ks * 32 + 8 * l4;
+          bf16x8 v = *(const bf16x8*)(qkv + qkv_bn + (long long)row * HD3 +
+                                      q_off + d0);
+          bf16x8 o;
+#pragma unroll
+          for (int j = 0; j < 8; ++j)
+            ((short*)&o)[j] = f2bf(bf2f(((short*)&v)[j]) * scale);
+          qfrag[m][ks] = o;
+        }
+      }
+      // stage this wave's 32 rows into the pair region columns
+      // [colbase, colbase+32): 2 lanes per row, 4 d-groups each
+      {
+        const int row = lane >> 1;
+        const int col = colbase + row;
+        const long long gq = qkv_bn + (long long)(q0 + row) * HD3 + q_off;
+        const long long go = o_bn + (long long)(q0 + row) * HD + head * DH;
+        float dsum = 0.f;
+#pragma unroll
+        for (int gg = 0; gg < 4; ++gg) {
+          const int g = (lane & 1) * 4 + gg;
+          const int d0 = g * 8;
+          bf16x8 qv = *(const bf16x8*)(qkv + gq + d0);
+          bf16x8 ov = *(const bf16x8*)(out + go + d0);
+          bf16x8 dov = *(const bf16x8*)(dout + go + d0);
+#pragma unroll
+          for (int j = 0; j < 8; ++j) {
+            const int d = d0 + j;
+            *(short*)(qt_lds + d * 128 + swz(d, col * 2)) =
+                f2bf(bf2f(((short*)&qv)[j]) * scale);
+            *(short*)(dot_lds + d * 128 + swz(d, col * 2)) = ((short*)&dov)[j];
+            dsum += bf2f(((short*)&ov)[j]) * bf2f(((short*)&dov)[j]);
+          }
+        }
+        dsum += __shfl_xor(dsum, 1, 64);  // join the row's two lanes
+        if ((lane & 1) == 0) {
+          d_lds[col] = dsum;
+          lse_lds[col] = lse[((long long)batch * H + head) * N + q0 + row];
+        }
+      }
+    }
+    __syncthreads();
+
+    const int max_tile = active ? ((chunk_off + 31 + wsz) / KT) : -1;
+
+    bf16x8 kreg, vreg;
+    auto issue_loads = [&](int t) {
+      const int kpos = (window - 1) * wsz + t * KT + su_key;
+      if (kpos >= 0) {
+        const long long base = qkv_bn + (long long)kpos * HD3;
+        kreg = *(const bf16x8*)(qkv + base + k_off + su_d0);
+        vreg = *(const bf16x8*)(qkv + base + v_off + su_d0);
+      } else {
+#pragma unroll
+        for (int j = 0; j < 8; ++j) {
+          ((short*)&kreg)[j] = 0;
+          ((short*)&vreg)[j] = 0;
+        }
+      }
+    };
+    auto write_lds = [&]() {
+      *(bf16x8*)(k_lds + su_key * 128 + swz(su_key, su_d0 * 2)) = kreg;
+      *(bf16x8*)(v_lds + su_key * 128 + swz(su_key, su_d0 * 2)) = vreg;
+#pragma unroll
+      for (int j = 0; j < 8; ++j) {
+        const int d = su_d0 + j;
+        *(short*)(kt_lds + d * 128 + swz(d, su_key * 2)) =
+            f2bf(bf2f(((short*)&kreg)[j]) * scale);
+      }
+    };
+
+    issue_loads(0);
+    write_lds();
+    __syncthreads();
+
+    for (int t = 0; t < tiles; ++t) {
+      if (t + 1 < tiles) issue_loads(t + 1);
+      const int kb = t * KT;
+      const int c_min = max(0, (t * KT - wsz) / 64 - round * 4);
+      const bool i_compute = active && t <= max_tile;
+
+      // ---- phase 1 (wave-local) ----
+      if (i_compute) {
+        f32x4 s[2][4];
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) s[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            const int key = n * 16 + l15;
+            bf16x8 kf = *(const bf16x8*)(k_lds + key * 128 +
+                                         swz(key, (ks * 32 + 8 * l4) * 2));
+#pragma unroll
+            for (int m = 0; m < 2; ++m)
+              s[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  qfrag[m][ks], kf, s[m][n], 0, 0, 0);
+          }
+        __builtin_amdgcn_s_setprio(0);
+
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int rowiw = chunk_off + m * 16 + l4 * 4 + r;
+            const float l = lse_lds[colbase + m * 16 + l4 * 4 + r];
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int kpos_band = kb + n * 16 + l15;
+              float v = ((float*)&s[m][n])[r];
+              v = (kpos_band > rowiw + wsz) ? 0.f : __expf(v - l);
+              ((float*)&s[m][n])[r] = v;
+            }
+          }
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            const int key = n * 16 + l15;
+            const int row0 = colbase + m * 16 + l4 * 4;
+            short pk[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) pk[r] = f2bf(((float*)&s[m][n])[r]);
+            *(unsigned long long*)(pds_lds + key * 128 + swz(key, row0 * 2)) =
+                *(unsigned long long*)pk;
+          }
+
+        f32x4 dp[2][4];
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) dp[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int m = 0; m < 2; ++m) {
+            const int row = q0 + m * 16 + l15;
+            const int d0 = ks * 32 + 8 * l4;
+            bf16x8 dof = *(const bf16x8*)(dout + o_bn + (long long)row * HD +
+                                          head * DH + d0);
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int key = n * 16 + l15;
+              bf16x8 vf = *(const bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2));
+              dp[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  dof, vf, dp[m][n], 0, 0, 0);
+            }
+          }
+        __builtin_amdgcn_s_setprio(0);
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const float dval = d_lds[colbase + m * 16 + l4 * 4 + r];
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              float p = ((float*)&s[m][n])[r];
+              float d = ((float*)&dp[m][n])[r];
+              ((float*)&dp[m][n])[r] = p * (d - dval);
+            }
+          }
+#pragma unroll
+        for (int m = 0; m < 2; ++m)
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            const int key = n * 16 + l15;
+            const int row0 = colbase + m * 16 + l4 * 4;
+            short dk4[4];
+#pragma unroll
+            for (int r = 0; r < 4; ++r) dk4[r] = f2bf(((float*)&dp[m][n])[r]);
+            *(unsigned long long*)(ds2_lds + key * 128 +
+                                   ((row0 * 2) ^ (uk4(key) * 32))) =
+                *(unsigned long long*)dk4;
+          }
+
+        __builtin_amdgcn_s_setprio(1);
+#pragma unroll
+        for (int ks = 0; ks < 2; ++ks)
+#pragma unroll
+          for (int m = 0; m < 2; ++m) {
+            const int krow1 = ks * 32 + l4 * 8 + (l15 >> 2);
+            const int krow2 = krow1 + 4;
+            const int colb = (colbase + m * 16 + (l15 & 3) * 4) * 2;
+            auto p1 = (AS3 bf16x4t*)(ds2_lds + krow1 * 128 +
+                                     (colb ^ (uk4(krow1) * 32)));
+            auto p2 = (AS3 bf16x4t*)(ds2_lds + krow2 * 128 +
+                                     (colb ^ (uk4(krow2) * 32)));
+            bf16x4t f1 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+            bf16x4t f2 = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
+            bf16x8 dsf;
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              ((__bf16*)&dsf)[j] = f1[j];
+              ((__bf16*)&dsf)[j + 4] = f2[j];
+            }
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              const int d = n * 16 + l15;
+              bf16x8 kf = *(const bf16x8*)(kt_lds + d * 128 +
+                                           swz(d, (ks * 32 + 8 * l4) * 2));
+              dqacc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
+                  dsf, kf, dqacc[m][n], 0, 0, 0);
+            }
+          }
+        __builtin_amdgcn_s_setprio(0);
+      }
+      __syncthreads();  // barrier A
+
+      if (t + 1 < tiles) write_lds();
+
+      // ---- phase 2: dV+dK key slices, K-dim split by ks across wave
+      // halves (half 1 accumulates into dacc2/dlook2) ----
+      {
+        const int ks = wid >> 2;          // fixed ks half per wave
+        const int keyslot = (wid & 3) * 16 + l15;
+        f32x4 dv[4], dk[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          dv[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+          dk[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
+        }
+        __builtin_amdgcn_s_setprio(1);
+        const int r0 = ks * 32 + 8 * l4;
+        for (int c = c_min; c < nactive; ++c) {
+          char* pds_c = pds_base + c * 8192;
+          char* ds2_c = ds2_base + c * 8192;
+          char* dot_c = dot_base + c * 8192;
+          char* qt_c = qt_base + c * 8192;
+          bf16x8 pf = *(const bf16x8*)(pds_c + keyslot * 128 +
+                                       swz(keyslot, r0 * 2));
+          bf16x8 dsf = *(const bf16x8*)(ds2_c + keyslot * 128 +
+                                        ((r0 * 2) ^ (uk4(keyslot) * 32)));
+#pragma unroll
+          for (int n = 0; n < 4; ++n) {
+            const int d = n * 16 + l15;
+            bf16x8 dof = *(const bf16x8*)(dot_c + d * 128 + swz(d, r0 * 2));
+            bf16x8 qf = *(const bf16x8*)(qt_c + d * 128 + swz(d, r0 * 2));
+            dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
+                                                            0, 0, 0);
+            dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, dk[n],
+                                                            0, 0, 0);
+          }
+        }
+        __builtin_amdgcn_s_setprio(0);
+        const bool lookback = kb < wsz;
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int kpos = (window - 1) * wsz + kb + (wid & 3) * 16 + l4 * 4 + r;
+          if (kpos >= 0) {
+            float* dstv;
+            float* dstk;
+            if (ks == 0) {
+              dstv = lookback
+                  ? dlook + look_bn + (long long)kpos * HD2 + lv_off
+                  : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
+              dstk = lookback
+                  ? dlook + look_bn + (long long)kpos * HD2 + lk_off
+                  : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
+            } else {
+              float* base2 = lookback ? dlook2 : dacc2;
+              dstv = base2 + look_bn + (long long)kpos * HD2 + lv_off;
+              dstk = base2 + look_bn + (long long)kpos * HD2 + lk_off;
+            }
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              float vv = ((float*)&dv[n])[r];
+              float vk = ((float*)&dk[n])[r];
+              if (round > 0) {
+                vv += dstv[n * 16 + l15];
+                vk += dstk[n * 16 + l15];
+              }
+              dstv[n * 16 + l15] = vv;
+              dstk[n * 16 + l15] = vk;
+            }
+          }
+        }
+      }
+      __syncthreads();  // barrier B
+    }
+
+    if (active) {
+#pragma unroll
+      for (int m = 0; m < 2; ++m)
+#pragma unroll
+        for (int r = 0; r < 4; ++r) {
+          const int row = q0 + m * 16 + l4 * 4 + r;
+#pragma unroll
+          for (int n = 0; n < 4; ++n)
+            dacc[qkv_bn + (long long)row * HD3 + q_off + n * 16 + l15] =
+                ((float*)&dqacc[m][n])[r];
+        }
+    }
+    __syncthreads();
+  }
+}
+
 // ---------------------------------------------------------------------------
 // finalize: inverse rotary rotation on the fp32 accumulator -> bf16 dqkv
 // ---------------------------------------------------------------------------
@@ -968,6 +1362,14 @@ int main(int argc, char** argv) {
   hipMalloc(&dl, ln * 4);
   hipMalloc(&dacc, qn * 4);
   hipMalloc(&dlook, (long long)B * N * 2 * H * DH * 4);
+  float *dacc2 = nullptr, *dlook2 = nullptr;
+#if VARIANT == 5
+  hipMalloc(&dacc2, (long long)B * N * 2 * H * DH * 4);
+  hipMalloc(&dlook2, (long long)B * N * 2 * H * DH * 4);
+  hipMemset(dacc2, 0, (long long)B * N * 2 * H * DH * 4);
+  hipMemset(dlook2, 0, (long long)B * N * 2 * H * DH * 4);
+#endif
+  (void)dacc2; (void)dlook2;
   hipMemcpy(dq, hq.data(), qn * 2, hipMemcpyHostToDevice);
   hipMemcpy(ddo, hdo.data(), on * 2, hipMemcpyHostToDevice);
   hipMemcpy(dou, ho.data(), on * 2, hipMemcpyHostToDevice);
@@ -980,7 +1382,11 @@ int main(int argc, char** argv) {
   printf("LDS %zu KiB\n", lds / 1024);
 
   for (int i = 0; i < 10; ++i)
-#if VARIANT == 4
+#if VARIANT == 5
+    attn_bwd_v5_kernel<<<grid, dim3(V5_BLOCK), lds>>>(ddo, dq, dou, dl,
+                                                      dacc, dlook, dacc2,
+                                                      dlook2,
+#elif VARIANT == 4
     attn_bwd_v4_kernel<<<grid, block, lds>>>(ddo, dq, dou, dl, dacc, dlook,
 #else
     attn_bwd_kernel<<<grid, block, lds>>>(ddo, dq, dou, dl, dacc, dlook,
@@ -995,7 +1401,11 @@ int main(int argc, char** argv) {
   hipEventCreate(&e1);
   hipEventRecord(e0);
   for (int i = 0; i < iters; ++i)
-#if VARIANT == 4
+#if VARIANT == 5
+    attn_bwd_v5_kernel<<<grid, dim3(V5_BLOCK), lds>>>(ddo, dq, dou, dl,
+                                                      dacc, dlook, dacc2,
+                                                      dlook2,
+#elif VARIANT == 4
     attn_bwd_v4_kernel<<<grid, block, lds>>>(ddo, dq, dou, dl, dacc, dlook,
 #else
     attn_bwd_kernel<<<grid, block, lds>>>(ddo, dq, dou, dl, dacc, dlook,
