@@ -50,6 +50,16 @@ __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return (byte_in_row ^ ((row & 7) << 4));
 }
 
+// per-key 32-B XOR window for the LINEAR [key][dh] V image (128-B
+// rows): the PV B-fragments are ds_read_b64_tr_b16 transposed reads
+// whose half-wave covers key rows {kb..kb+3, kb+8..kb+11}; v(k) is a
+// bijection onto 0..3 within each row parity, so the 8 rows cover all
+// 64 banks (same derivation as attention_bwd.hip's dS^T image)
+__device__ __forceinline__ int uk4(int k) { return ((k & 2) >> 1) | ((k & 8) >> 2); }
+
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4t;
+#define AS3 __attribute__((address_space(3)))
+
 __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
     const short* __restrict__ qkv,   // (B, N, 3*H*DH) bf16, PRE-ROTATED
     const short* __restrict__ halo,  // (B, wsz, 2*H*DH) rotated [k|v]
@@ -174,14 +184,10 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         const int key = su_key[u];
         const int d0 = su_d0[u];
         *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
-        // NOTE: a lane-staggered write order was tried for bank spreading
-        // and measured 29% SLOWER (runtime-indexed vector extract goes to
-        // scratch; tools/ablate_attn.hip VARIANT 4) — keep static order.
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d = d0 + j;
-          *(short*)(v_lds + d * 128 + swz(d, key * 2)) = ((short*)&vreg[u])[j];
-        }
+        // V stored LINEAR [key][dh] (one 16-B write; the old [dh][key]
+        // scatter-transpose was 8 b16 writes per fragment) — the PV
+        // step reads it transposed with ds_read_b64_tr_b16 (T10)
+        *(bf16x8*)(v_lds + key * 128 + ((d0 * 2) ^ (uk4(key) * 32))) = vreg[u];
       }
     };
 
@@ -289,6 +295,26 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
         __builtin_amdgcn_s_setprio(1);  // favor the MFMA cluster (T5)
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
+          // V B-fragments by hardware transpose read: key rows
+          // kk0 + (l15>>2) (+4), column = the dh block
+          const int kr1 = ks * 32 + l4 * 8 + (l15 >> 2);
+          const int kr2 = kr1 + 4;
+          bf16x8 vfr[4];
+#pragma unroll
+          for (int d = 0; d < 4; ++d) {
+            const int colb = (d * 16 + (l15 & 3) * 4) * 2;
+            auto p1 = (AS3 bf16x4t*)(v_lds + kr1 * 128 +
+                                     (colb ^ (uk4(kr1) * 32)));
+            auto p2 = (AS3 bf16x4t*)(v_lds + kr2 * 128 +
+                                     (colb ^ (uk4(kr2) * 32)));
+            bf16x4t a = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+            bf16x4t b = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
+#pragma unroll
+            for (int j = 0; j < 4; ++j) {
+              ((__bf16*)&vfr[d])[j] = a[j];
+              ((__bf16*)&vfr[d])[j + 4] = b[j];
+            }
+          }
 #pragma unroll
           for (int m = 0; m < MF; ++m) {
             const int row = m * 16 + l15;
@@ -296,10 +322,8 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_fwd_kernel(
             bf16x8 pfrag = *(const bf16x8*)(p_lds + row * 128 + swz(row, kk0 * 2));
 #pragma unroll
             for (int d = 0; d < 4; ++d) {
-              const int dcol = d * 16 + l15;
-              bf16x8 vfrag = *(const bf16x8*)(v_lds + dcol * 128 + swz(dcol, kk0 * 2));
               oacc[m][d] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                  pfrag, vfrag, oacc[m][d], 0, 0, 0);
+                  pfrag, vfr[d], oacc[m][d], 0, 0, 0);
             }
           }
         }
