@@ -100,10 +100,14 @@ ENABLED = os.environ.get("PROGEN_FP8", "0") == "1"
 def scaled_mm(a: torch.Tensor, b_colmajor: torch.Tensor,
               out_dtype: torch.dtype = torch.bfloat16,
               bias: torch.Tensor = None) -> torch.Tensor:
-    """(M,K) row-major @ (K,N) column-major in e4m3; fp32 accumulate."""
+    """(M,K) row-major @ (K,N) column-major in e4m3; fp32 accumulate.
+
+    b is quantized through its row-major transpose view (the fused
+    quantize kernel contiguous()-izes its input, which would silently
+    relayout a column-major tensor) and transposed back."""
     qa, sa = quantize_e4m3(a)
-    qb, sb = quantize_e4m3(b_colmajor)
-    return torch._scaled_mm(qa, qb, scale_a=sa, scale_b=sb, bias=bias,
+    qbt, sb = quantize_e4m3(b_colmajor.t().contiguous())
+    return torch._scaled_mm(qa, qbt.t(), scale_a=sa, scale_b=sb, bias=bias,
                             out_dtype=out_dtype)
 
 
