@@ -1224,40 +1224,52 @@ __global__ __launch_bounds__(V5_BLOCK) void attn_bwd_v5_kernel(
           }
         }
         __builtin_amdgcn_s_setprio(0);
-        const bool lookback = kb < wsz;
+        __syncthreads();  // B1: slices done reading pds/ds2 everywhere
+        // half 1 parks its partial in the (now dead) ds2 region scratch:
+        // [slot = key-slice][key16 x dh64 f32] x2 (dv, dk)
+        float* scratch = (float*)(ds2_base) + (wid & 3) * 2048;
+        if (ks == 1) {
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int kpos = (window - 1) * wsz + kb + (wid & 3) * 16 + l4 * 4 + r;
-          if (kpos >= 0) {
-            float* dstv;
-            float* dstk;
-            if (ks == 0) {
-              dstv = lookback
+          for (int n = 0; n < 4; ++n)
+#pragma unroll
+            for (int r = 0; r < 4; ++r) {
+              scratch[(l4 * 4 + r) * 64 + n * 16 + l15] = ((float*)&dv[n])[r];
+              scratch[1024 + (l4 * 4 + r) * 64 + n * 16 + l15] =
+                  ((float*)&dk[n])[r];
+            }
+        }
+        __syncthreads();  // B2: partials visible
+        if (ks == 0) {
+          const bool lookback = kb < wsz;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int kpos = (window - 1) * wsz + kb + (wid & 3) * 16 + l4 * 4 + r;
+            if (kpos >= 0) {
+              float* dstv = lookback
                   ? dlook + look_bn + (long long)kpos * HD2 + lv_off
                   : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
-              dstk = lookback
+              float* dstk = lookback
                   ? dlook + look_bn + (long long)kpos * HD2 + lk_off
                   : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
-            } else {
-              float* base2 = lookback ? dlook2 : dacc2;
-              dstv = base2 + look_bn + (long long)kpos * HD2 + lv_off;
-              dstk = base2 + look_bn + (long long)kpos * HD2 + lk_off;
-            }
 #pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              float vv = ((float*)&dv[n])[r];
-              float vk = ((float*)&dk[n])[r];
-              if (round > 0) {
-                vv += dstv[n * 16 + l15];
-                vk += dstk[n * 16 + l15];
+              for (int n = 0; n < 4; ++n) {
+                float vv = ((float*)&dv[n])[r] +
+                           scratch[(l4 * 4 + r) * 64 + n * 16 + l15];
+                float vk = ((float*)&dk[n])[r] +
+                           scratch[1024 + (l4 * 4 + r) * 64 + n * 16 + l15];
+                if (round > 0) {
+                  vv += dstv[n * 16 + l15];
+                  vk += dstk[n * 16 + l15];
+                }
+                dstv[n * 16 + l15] = vv;
+                dstk[n * 16 + l15] = vk;
               }
-              dstv[n * 16 + l15] = vv;
-              dstk[n * 16 + l15] = vk;
             }
           }
         }
       }
-      __syncthreads();  // barrier B
+      __syncthreads();  // barrier B3: scratch reads done before t+1's
+                        // phase-1 overwrites ds2
     }
 
     if (active) {
