@@ -25,7 +25,10 @@ load_dotenv()  # reference: sample.py:1-2
 @click.option('--cached', default=False, is_flag=True,
               help='incremental decode with per-layer recurrent caches '
                    '(O(window) per token; identical semantics)')
-def main(seed, checkpoint_path, prime, fast, cached):
+@click.option('--graph', default=False, is_flag=True,
+              help='with --cached on GPU: replay the per-token step as one '
+                   'captured hipGraph (~2x the eager cached step)')
+def main(seed, checkpoint_path, prime, fast, cached, graph):
     _, get_last_checkpoint, _ = get_checkpoint_fns(checkpoint_path)
     last_checkpoint = get_last_checkpoint()
     if last_checkpoint is None:
@@ -66,7 +69,7 @@ def main(seed, checkpoint_path, prime, fast, cached):
     if cached:
         from progen_amd.decode import sample_cached
         sampled = sample_cached(module, prime_tensor, seq_len, top_k=25,
-                                add_bos=True, generator=g)
+                                add_bos=True, generator=g, graph=graph)
     elif fast:
         sampled = sample_fast(fwd, prime_tensor, seq_len, top_k=25,
                               add_bos=True, generator=g,
