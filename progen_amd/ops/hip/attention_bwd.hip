@@ -15,22 +15,28 @@
 // casts the fp32 accumulator to bf16 dqkv. Input qkv is PRE-ROTATED
 // (ops/hip/rope_qkv.hip).
 //
-// Geometry: block = one (batch, head, window), 4 waves, each wave owns a
-// 64-row q-chunk. Round-2 restructure (harness VARIANT 4,
-// tools/ablate_attn_bwd.hip — 1965 -> 1500 us/call at the production
-// grid B=64 H=24 wsz=256, bitwise-identical output; PMC showed the old
-// 5-barriers-per-tile schedule parked waves on barriers 50% of cycles):
-// per 64-key tile there are TWO barriers —
+// Geometry: block = one (batch, head, window), EIGHT waves of 32-row
+// q-chunks (round-2 ladder, tools/ablate_attn_bwd.hip: the V4 2-barrier
+// 4-wave restructure took 1965 -> 1500 us/call at the production grid;
+// this V5 8-wave geometry takes it to 1257 us, both bitwise-identical.
+// 8 waves halve the per-wave register state (s/dp/dqacc are [2][4]:
+// 240 VGPRs) so each SIMD runs TWO waves that cover each other's
+// stalls — the V4 PMC still showed 47.8% of cycles barrier/wait-parked
+// at 1 wave/SIMD). Wave pairs share the 64-row chunk LDS regions
+// (layouts and swizzles unchanged); the dV/dK key slices split their
+// MFMA K-dim by ks-half across the block's wave halves, the upper half
+// parking its partials in the (dead-by-then) dS^T region and the lower
+// half combining + storing (plain stores stay raceless, no extra
+// global buffers). Per 64-key tile —
 //   phase 1 (wave-local, no barrier): S, P (P^T b64-written), dP,
 //     dS = P o (dP - D) (dS^T b64-written), and dQ += dS k_s where the
 //     dS A-fragments come from ds_read_b64_tr_b16 transposed reads of
 //     the wave's OWN just-written dS^T region (same-wave DS ordering is
-//     program order). The old separate row-major dS image (dsrl, 16
-//     scattered b16 writes per m,n) is gone.
-//   barrier A; then tile t+1's k/v/kt staging (phase 2 reads none of
-//     them) overlaps with phase 2: the dV and dK 16-key OUTPUT SLICES,
-//     merged into one chunk-loop, whose MFMA K-dim spans ALL chunks'
-//     P^T/dS^T/dO^T/Q^T regions; barrier B.
+//     program order).
+//   barrier A; tile t+1's k/v/kt staging overlaps the dV/dK slices;
+//   barrier B1 (slice reads done) -> upper-half partials to LDS
+//   scratch; barrier B2 -> lower half combines + stores; barrier B3
+//   (scratch reads done before t+1's dS^T writes).
 // Every dV/dK element is produced by exactly ONE wave, so the stores
 // are PLAIN (no atomics): each window's own-band gradients go to dacc,
 // its lookback-band gradients to a separate dlook buffer, and
@@ -85,7 +91,10 @@ __device__ __forceinline__ int uk4(int k) { return ((k & 2) >> 1) | ((k & 8) >> 
 typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 bf16x4t;
 #define AS3 __attribute__((address_space(3)))
 
-__global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
+#define V5_WAVES 8
+#define V5_BLOCK (V5_WAVES * WAVE)
+
+__global__ __launch_bounds__(V5_BLOCK) void attn_bwd_kernel(
     const short* __restrict__ dout, const short* __restrict__ qkv,
     const short* __restrict__ halo,   // CP lookback for window 0 (see
                                       // attention_fwd.hip) or nullptr
@@ -93,25 +102,27 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
     float* __restrict__ dacc, float* __restrict__ dlook,
     float* __restrict__ dhalo,        // (B, wsz, 2*H*DH) fp32 window-0
                                       // lookback grads (CP) or nullptr
-                                      // (quirk path: grads discarded)
     int B, int N, int H, int wsz) {
   const int window = blockIdx.x;
   const int head = blockIdx.y;
   const int batch = blockIdx.z;
 
   const int lane = threadIdx.x % WAVE;
-  const int wid = threadIdx.x / WAVE;
+  const int wid = threadIdx.x / WAVE;   // 0..7
   const int l15 = lane & 15;
   const int l4 = lane >> 4;
+  const int pid = wid >> 1;             // pair region 0..3
+  const int phalf = wid & 1;            // row half within the pair
 
   const long long HD3 = 3LL * H * DH;
   const long long HD = (long long)H * DH;
+  const long long HD2 = 2LL * H * DH;
   const long long qkv_bn = (long long)batch * N * HD3;
   const long long o_bn = (long long)batch * N * HD;
   const int q_off = head * DH;
   const int k_off = H * DH + head * DH;
   const int v_off = 2 * H * DH + head * DH;
-  const long long look_bn = (long long)batch * N * (2LL * H * DH);
+  const long long look_bn = (long long)batch * N * HD2;
   const int lk_off = head * DH;
   const int lv_off = H * DH + head * DH;
 
@@ -119,45 +130,47 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
   char* k_lds = smem;
   char* kt_lds = smem + 8192;
   char* v_lds = smem + 16384;
-  char* qt_base = smem + 24576;                  // 32 KiB
-  char* dot_base = qt_base + 4 * 8192;           // 32 KiB
-  char* pds_base = dot_base + 4 * 8192;          // 32 KiB (P^T)
-  char* ds2_base = pds_base + 4 * 8192;          // 32 KiB (dS^T, XOR-u)
-  float* d_lds = (float*)(ds2_base + 4 * 8192 + wid * 256);
-  float* lse_lds = (float*)(ds2_base + 4 * 8192 + 1024 + wid * 256);
+  char* qt_base = smem + 24576;
+  char* dot_base = qt_base + 4 * 8192;
+  char* pds_base = dot_base + 4 * 8192;
+  char* ds2_base = pds_base + 4 * 8192;
+  float* d_base = (float*)(ds2_base + 4 * 8192);        // [pair][64]
+  float* lse_base = (float*)(ds2_base + 4 * 8192 + 1024);
 
-  char* qt_lds = qt_base + wid * 8192;
-  char* dot_lds = dot_base + wid * 8192;
-  char* pds_lds = pds_base + wid * 8192;
-  char* ds2_lds = ds2_base + wid * 8192;
+  char* qt_lds = qt_base + pid * 8192;
+  char* dot_lds = dot_base + pid * 8192;
+  char* pds_lds = pds_base + pid * 8192;
+  char* ds2_lds = ds2_base + pid * 8192;
+  float* d_lds = d_base + pid * 64;
+  float* lse_lds = lse_base + pid * 64;
 
   const float scale = rsqrtf((float)DH);
   const int tiles = 2 * wsz / KT;
-  const int chunks = wsz / 64;
-  const int rounds = (chunks + 3) / 4;
+  const int chunks64 = wsz / 64;                 // pair-sized chunks
+  const int rounds = (chunks64 + 3) / 4;
 
-  const int su_key[2] = {(int)threadIdx.x >> 3,
-                         (int)(threadIdx.x + ATTN_BLOCK) >> 3};
-  const int su_d0[2] = {((int)threadIdx.x & 7) * 8,
-                        (((int)threadIdx.x + ATTN_BLOCK) & 7) * 8};
+  // block-wide k/v tile staging: 512 threads cover 64 keys x 64 dh
+  const int su_key = (int)threadIdx.x >> 3;
+  const int su_d0 = ((int)threadIdx.x & 7) * 8;
 
   for (int round = 0; round < rounds; ++round) {
-    const int chunk = round * 4 + wid;
-    const bool active = chunk < chunks;
-    const int nactive = min(4, chunks - round * 4);
-    const int chunk_off = chunk * 64;
+    const int c64 = round * 4 + pid;             // this pair's 64-chunk
+    const bool active = c64 < chunks64;
+    const int nactive = min(4, chunks64 - round * 4);
+    const int chunk_off = c64 * 64 + phalf * 32; // rows-in-window
     const int q0 = window * wsz + chunk_off;
+    const int colbase = phalf * 32;              // in the pair region
 
-    bf16x8 qfrag[4][2];
-    f32x4 dqacc[4][4];
+    bf16x8 qfrag[2][2];
+    f32x4 dqacc[2][4];
 #pragma unroll
-    for (int m = 0; m < 4; ++m)
+    for (int m = 0; m < 2; ++m)
 #pragma unroll
       for (int d = 0; d < 4; ++d) dqacc[m][d] = (f32x4){0.f, 0.f, 0.f, 0.f};
 
     if (active) {
 #pragma unroll
-      for (int m = 0; m < 4; ++m) {
+      for (int m = 0; m < 2; ++m) {
         const int row = q0 + m * 16 + l15;
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks) {
@@ -171,13 +184,17 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           qfrag[m][ks] = o;
         }
       }
+      // stage this wave's 32 rows into the pair region columns
+      // [colbase, colbase+32): 2 lanes per row, 4 d-groups each
       {
-        const int row = lane;
+        const int row = lane >> 1;
+        const int col = colbase + row;
         const long long gq = qkv_bn + (long long)(q0 + row) * HD3 + q_off;
         const long long go = o_bn + (long long)(q0 + row) * HD + head * DH;
         float dsum = 0.f;
 #pragma unroll
-        for (int g = 0; g < 8; ++g) {
+        for (int gg = 0; gg < 4; ++gg) {
+          const int g = (lane & 1) * 4 + gg;
           const int d0 = g * 8;
           bf16x8 qv = *(const bf16x8*)(qkv + gq + d0);
           bf16x8 ov = *(const bf16x8*)(out + go + d0);
@@ -185,57 +202,52 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
 #pragma unroll
           for (int j = 0; j < 8; ++j) {
             const int d = d0 + j;
-            *(short*)(qt_lds + d * 128 + swz(d, row * 2)) =
+            *(short*)(qt_lds + d * 128 + swz(d, col * 2)) =
                 f2bf(bf2f(((short*)&qv)[j]) * scale);
-            *(short*)(dot_lds + d * 128 + swz(d, row * 2)) = ((short*)&dov)[j];
+            *(short*)(dot_lds + d * 128 + swz(d, col * 2)) = ((short*)&dov)[j];
             dsum += bf2f(((short*)&ov)[j]) * bf2f(((short*)&dov)[j]);
           }
         }
-        d_lds[row] = dsum;
-        lse_lds[row] = lse[((long long)batch * H + head) * N + q0 + row];
+        dsum += __shfl_xor(dsum, 1, 64);  // join the row's two lanes
+        if ((lane & 1) == 0) {
+          d_lds[col] = dsum;
+          lse_lds[col] = lse[((long long)batch * H + head) * N + q0 + row];
+        }
       }
     }
     __syncthreads();
 
-    const int max_tile = active ? ((chunk_off + 63 + wsz) / KT) : -1;
+    const int max_tile = active ? ((chunk_off + 31 + wsz) / KT) : -1;
 
-    bf16x8 kreg[2], vreg[2];
+    bf16x8 kreg, vreg;
     auto issue_loads = [&](int t) {
+      const int kpos = (window - 1) * wsz + t * KT + su_key;
+      if (kpos >= 0) {
+        const long long base = qkv_bn + (long long)kpos * HD3;
+        kreg = *(const bf16x8*)(qkv + base + k_off + su_d0);
+        vreg = *(const bf16x8*)(qkv + base + v_off + su_d0);
+      } else if (halo != nullptr) {
+        const long long hb =
+            ((long long)batch * wsz + (kpos + wsz)) * (2LL * H * DH);
+        kreg = *(const bf16x8*)(halo + hb + head * DH + su_d0);
+        vreg = *(const bf16x8*)(halo + hb + (long long)H * DH + head * DH +
+                                su_d0);
+      } else {
 #pragma unroll
-      for (int u = 0; u < 2; ++u) {
-        const int kpos = (window - 1) * wsz + t * KT + su_key[u];
-        if (kpos >= 0) {
-          const long long base = qkv_bn + (long long)kpos * HD3;
-          kreg[u] = *(const bf16x8*)(qkv + base + k_off + su_d0[u]);
-          vreg[u] = *(const bf16x8*)(qkv + base + v_off + su_d0[u]);
-        } else if (halo != nullptr) {
-          const long long hb =
-              ((long long)batch * wsz + (kpos + wsz)) * (2LL * H * DH);
-          kreg[u] = *(const bf16x8*)(halo + hb + head * DH + su_d0[u]);
-          vreg[u] = *(const bf16x8*)(halo + hb + (long long)H * DH +
-                                     head * DH + su_d0[u]);
-        } else {
-#pragma unroll
-          for (int j = 0; j < 8; ++j) {
-            ((short*)&kreg[u])[j] = 0;
-            ((short*)&vreg[u])[j] = 0;
-          }
+        for (int j = 0; j < 8; ++j) {
+          ((short*)&kreg)[j] = 0;
+          ((short*)&vreg)[j] = 0;
         }
       }
     };
     auto write_lds = [&]() {
+      *(bf16x8*)(k_lds + su_key * 128 + swz(su_key, su_d0 * 2)) = kreg;
+      *(bf16x8*)(v_lds + su_key * 128 + swz(su_key, su_d0 * 2)) = vreg;
 #pragma unroll
-      for (int u = 0; u < 2; ++u) {
-        const int key = su_key[u];
-        const int d0 = su_d0[u];
-        *(bf16x8*)(k_lds + key * 128 + swz(key, d0 * 2)) = kreg[u];
-        *(bf16x8*)(v_lds + key * 128 + swz(key, d0 * 2)) = vreg[u];
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d = d0 + j;
-          *(short*)(kt_lds + d * 128 + swz(d, key * 2)) =
-              f2bf(bf2f(((short*)&kreg[u])[j]) * scale);
-        }
+      for (int j = 0; j < 8; ++j) {
+        const int d = su_d0 + j;
+        *(short*)(kt_lds + d * 128 + swz(d, su_key * 2)) =
+            f2bf(bf2f(((short*)&kreg)[j]) * scale);
       }
     };
 
@@ -249,11 +261,11 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
       const int c_min = max(0, (t * KT - wsz) / 64 - round * 4);
       const bool i_compute = active && t <= max_tile;
 
-      // ---- phase 1 (wave-local): S, P, P^T, dP, dS, dS^T, dQ ----
+      // ---- phase 1 (wave-local) ----
       if (i_compute) {
-        f32x4 s[4][4];
+        f32x4 s[2][4];
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int m = 0; m < 2; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n) s[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
         __builtin_amdgcn_s_setprio(1);
@@ -265,18 +277,18 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             bf16x8 kf = *(const bf16x8*)(k_lds + key * 128 +
                                          swz(key, (ks * 32 + 8 * l4) * 2));
 #pragma unroll
-            for (int m = 0; m < 4; ++m)
+            for (int m = 0; m < 2; ++m)
               s[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
                   qfrag[m][ks], kf, s[m][n], 0, 0, 0);
           }
         __builtin_amdgcn_s_setprio(0);
 
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int m = 0; m < 2; ++m)
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
             const int rowiw = chunk_off + m * 16 + l4 * 4 + r;
-            const float l = lse_lds[m * 16 + l4 * 4 + r];
+            const float l = lse_lds[colbase + m * 16 + l4 * 4 + r];
 #pragma unroll
             for (int n = 0; n < 4; ++n) {
               const int kpos_band = kb + n * 16 + l15;
@@ -286,11 +298,11 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
             }
           }
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int m = 0; m < 2; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n) {
             const int key = n * 16 + l15;
-            const int row0 = m * 16 + l4 * 4;
+            const int row0 = colbase + m * 16 + l4 * 4;
             short pk[4];
 #pragma unroll
             for (int r = 0; r < 4; ++r) pk[r] = f2bf(((float*)&s[m][n])[r]);
@@ -298,17 +310,16 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
                 *(unsigned long long*)pk;
           }
 
-        // dP = dO V'^T (wave-local: v_lds staged, dout from global)
-        f32x4 dp[4][4];
+        f32x4 dp[2][4];
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int m = 0; m < 2; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n) dp[m][n] = (f32x4){0.f, 0.f, 0.f, 0.f};
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
-          for (int m = 0; m < 4; ++m) {
+          for (int m = 0; m < 2; ++m) {
             const int row = q0 + m * 16 + l15;
             const int d0 = ks * 32 + 8 * l4;
             bf16x8 dof = *(const bf16x8*)(dout + o_bn + (long long)row * HD +
@@ -323,24 +334,23 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           }
         __builtin_amdgcn_s_setprio(0);
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int m = 0; m < 2; ++m)
 #pragma unroll
           for (int r = 0; r < 4; ++r) {
-            const float dval = d_lds[m * 16 + l4 * 4 + r];
+            const float dval = d_lds[colbase + m * 16 + l4 * 4 + r];
 #pragma unroll
             for (int n = 0; n < 4; ++n) {
               float p = ((float*)&s[m][n])[r];
               float d = ((float*)&dp[m][n])[r];
-              ((float*)&dp[m][n])[r] = p * (d - dval);  // now dS
+              ((float*)&dp[m][n])[r] = p * (d - dval);
             }
           }
-        // write dS^T once: b64 into [key][row] image, col ^= u(key)*32
 #pragma unroll
-        for (int m = 0; m < 4; ++m)
+        for (int m = 0; m < 2; ++m)
 #pragma unroll
           for (int n = 0; n < 4; ++n) {
             const int key = n * 16 + l15;
-            const int row0 = m * 16 + l4 * 4;
+            const int row0 = colbase + m * 16 + l4 * 4;
             short dk4[4];
 #pragma unroll
             for (int r = 0; r < 4; ++r) dk4[r] = f2bf(((float*)&dp[m][n])[r]);
@@ -349,16 +359,14 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
                 *(unsigned long long*)dk4;
           }
 
-        // dQ += dS k_s: A-frags by tr reads of the wave's OWN dS^T
-        // (same-wave DS ordering; no barrier needed)
         __builtin_amdgcn_s_setprio(1);
 #pragma unroll
         for (int ks = 0; ks < 2; ++ks)
 #pragma unroll
-          for (int m = 0; m < 4; ++m) {
+          for (int m = 0; m < 2; ++m) {
             const int krow1 = ks * 32 + l4 * 8 + (l15 >> 2);
             const int krow2 = krow1 + 4;
-            const int colb = (m * 16 + (l15 & 3) * 4) * 2;
+            const int colb = (colbase + m * 16 + (l15 & 3) * 4) * 2;
             auto p1 = (AS3 bf16x4t*)(ds2_lds + krow1 * 128 +
                                      (colb ^ (uk4(krow1) * 32)));
             auto p2 = (AS3 bf16x4t*)(ds2_lds + krow2 * 128 +
@@ -382,13 +390,15 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           }
         __builtin_amdgcn_s_setprio(0);
       }
-      __syncthreads();  // barrier A: P^T/dS^T ready; k/v/kt reads done
+      __syncthreads();  // barrier A
 
-      // stage t+1 into k/v/kt while phase 2 runs (phase 2 reads none)
       if (t + 1 < tiles) write_lds();
 
-      // ---- phase 2: merged dV + dK key-slices ----
+      // ---- phase 2: dV+dK key slices, K-dim split by ks across wave
+      // halves (half 1 accumulates into dacc2/dlook2) ----
       {
+        const int ks = wid >> 2;          // fixed ks half per wave
+        const int keyslot = (wid & 3) * 16 + l15;
         f32x4 dv[4], dk[4];
 #pragma unroll
         for (int n = 0; n < 4; ++n) {
@@ -396,71 +406,88 @@ __global__ __launch_bounds__(ATTN_BLOCK) void attn_bwd_kernel(
           dk[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
         }
         __builtin_amdgcn_s_setprio(1);
+        const int r0 = ks * 32 + 8 * l4;
         for (int c = c_min; c < nactive; ++c) {
           char* pds_c = pds_base + c * 8192;
           char* ds2_c = ds2_base + c * 8192;
           char* dot_c = dot_base + c * 8192;
           char* qt_c = qt_base + c * 8192;
+          bf16x8 pf = *(const bf16x8*)(pds_c + keyslot * 128 +
+                                       swz(keyslot, r0 * 2));
+          bf16x8 dsf = *(const bf16x8*)(ds2_c + keyslot * 128 +
+                                        ((r0 * 2) ^ (uk4(keyslot) * 32)));
 #pragma unroll
-          for (int ks = 0; ks < 2; ++ks) {
-            const int key = wid * 16 + l15;
-            const int r0 = ks * 32 + 8 * l4;
-            bf16x8 pf = *(const bf16x8*)(pds_c + key * 128 + swz(key, r0 * 2));
-            bf16x8 dsf = *(const bf16x8*)(ds2_c + key * 128 +
-                                          ((r0 * 2) ^ (uk4(key) * 32)));
-#pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              const int d = n * 16 + l15;
-              bf16x8 dof = *(const bf16x8*)(dot_c + d * 128 + swz(d, r0 * 2));
-              bf16x8 qf = *(const bf16x8*)(qt_c + d * 128 + swz(d, r0 * 2));
-              dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
-                                                              0, 0, 0);
-              dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, dk[n],
-                                                              0, 0, 0);
-            }
+          for (int n = 0; n < 4; ++n) {
+            const int d = n * 16 + l15;
+            bf16x8 dof = *(const bf16x8*)(dot_c + d * 128 + swz(d, r0 * 2));
+            bf16x8 qf = *(const bf16x8*)(qt_c + d * 128 + swz(d, r0 * 2));
+            dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
+                                                            0, 0, 0);
+            dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, dk[n],
+                                                            0, 0, 0);
           }
         }
         __builtin_amdgcn_s_setprio(0);
-        const bool lookback = kb < wsz;
+        __syncthreads();  // B1: slices done reading pds/ds2 everywhere
+        // half 1 parks its partial in the (now dead) ds2 region scratch:
+        // [slot = key-slice][key16 x dh64 f32] x2 (dv, dk)
+        float* scratch = (float*)(ds2_base) + (wid & 3) * 2048;
+        if (ks == 1) {
 #pragma unroll
-        for (int r = 0; r < 4; ++r) {
-          const int kpos = (window - 1) * wsz + kb + wid * 16 + l4 * 4 + r;
-          float* dstv;
-          float* dstk;
-          if (kpos >= 0) {
-            dstv = lookback
-                ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lv_off
-                : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
-            dstk = lookback
-                ? dlook + look_bn + (long long)kpos * (2LL * H * DH) + lk_off
-                : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
-          } else if (dhalo != nullptr) {
-            const long long hb =
-                ((long long)batch * wsz + (kpos + wsz)) * (2LL * H * DH);
-            dstv = dhalo + hb + lv_off;
-            dstk = dhalo + hb + lk_off;
-          } else {
-            continue;  // window-0 zero-pad quirk: grads discarded
-          }
+          for (int n = 0; n < 4; ++n)
 #pragma unroll
-          for (int n = 0; n < 4; ++n) {
-            float vv = ((float*)&dv[n])[r];
-            float vk = ((float*)&dk[n])[r];
-            if (round > 0) {
-              vv += dstv[n * 16 + l15];
-              vk += dstk[n * 16 + l15];
+            for (int r = 0; r < 4; ++r) {
+              scratch[(l4 * 4 + r) * 64 + n * 16 + l15] = ((float*)&dv[n])[r];
+              scratch[1024 + (l4 * 4 + r) * 64 + n * 16 + l15] =
+                  ((float*)&dk[n])[r];
             }
-            dstv[n * 16 + l15] = vv;
-            dstk[n * 16 + l15] = vk;
+        }
+        __syncthreads();  // B2: partials visible
+        if (ks == 0) {
+          const bool lookback = kb < wsz;
+#pragma unroll
+          for (int r = 0; r < 4; ++r) {
+            const int kpos = (window - 1) * wsz + kb + (wid & 3) * 16 + l4 * 4 + r;
+            float* dstv;
+            float* dstk;
+            if (kpos >= 0) {
+              dstv = lookback
+                  ? dlook + look_bn + (long long)kpos * HD2 + lv_off
+                  : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
+              dstk = lookback
+                  ? dlook + look_bn + (long long)kpos * HD2 + lk_off
+                  : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
+            } else if (dhalo != nullptr) {
+              const long long hb =
+                  ((long long)batch * wsz + (kpos + wsz)) * HD2;
+              dstv = dhalo + hb + lv_off;
+              dstk = dhalo + hb + lk_off;
+            } else {
+              continue;  // window-0 zero-pad quirk: grads discarded
+            }
+#pragma unroll
+            for (int n = 0; n < 4; ++n) {
+              float vv = ((float*)&dv[n])[r] +
+                         scratch[(l4 * 4 + r) * 64 + n * 16 + l15];
+              float vk = ((float*)&dk[n])[r] +
+                         scratch[1024 + (l4 * 4 + r) * 64 + n * 16 + l15];
+              if (round > 0) {
+                vv += dstv[n * 16 + l15];
+                vk += dstk[n * 16 + l15];
+              }
+              dstv[n * 16 + l15] = vv;
+              dstk[n * 16 + l15] = vk;
+            }
           }
         }
       }
-      __syncthreads();  // barrier B: slices done; t+1 staged
+      __syncthreads();  // barrier B3: scratch reads done before t+1's
+                        // phase-1 overwrites ds2
     }
 
     if (active) {
 #pragma unroll
-      for (int m = 0; m < 4; ++m)
+      for (int m = 0; m < 2; ++m)
 #pragma unroll
         for (int r = 0; r < 4; ++r) {
           const int row = q0 + m * 16 + l4 * 4 + r;
@@ -529,7 +556,7 @@ void attn_bwd_launch(const void* dout, const void* qkv, const void* halo,
                      const float* lse, float* dacc, float* dlook,
                      float* dhalo, void* dqkv, int B, int N,
                      int H, int wsz, hipStream_t stream) {
-  dim3 grid(N / wsz, H, B), block(ATTN_BLOCK);
+  dim3 grid(N / wsz, H, B), block(V5_BLOCK);  // 8 waves: 2/SIMD
   size_t lds = 24576 + 131072 + 2048;  // 154 KiB
   attn_bwd_kernel<<<grid, block, lds, stream>>>(
       (const short*)dout, (const short*)qkv, (const short*)halo,
