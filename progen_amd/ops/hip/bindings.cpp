@@ -401,6 +401,16 @@ uint32_t crc32c_host(py::bytes data) {
   return c ^ 0xFFFFFFFFu;
 }
 
+at::Tensor colsum(const at::Tensor& dy) {
+  TORCH_CHECK(dy.is_cuda() && dy.is_contiguous() && dy.dim() == 2 &&
+              dy.scalar_type() == at::kBFloat16);
+  TORCH_CHECK(dy.size(1) % 8 == 0, "colsum: C % 8");
+  auto out = at::zeros({dy.size(1)}, dy.options().dtype(at::kFloat));
+  colsum_launch(dy.data_ptr(), out.data_ptr<float>(), dy.size(0),
+                (int)dy.size(1), cur_stream());
+  return out;
+}
+
 at::Tensor fp8_quantize(const at::Tensor& t, const at::Tensor& scale) {
   TORCH_CHECK(t.is_cuda() && t.is_contiguous() &&
               t.scalar_type() == at::kBFloat16);
@@ -456,6 +466,7 @@ PYBIND11_MODULE(TORCH_EXTENSION_NAME, m) {
         py::arg("out"), py::arg("lse"), py::arg("heads"), py::arg("window"),
         py::arg("halo") = py::none());
   m.def("crc32c", &crc32c_host, "CRC-32C (slicing-by-8, host)");
+  m.def("colsum", &colsum, "bf16 column sum (bias grads), fp32 accum");
   m.def("fp8_quantize", &fp8_quantize, "fused bf16 -> e4m3 at device scale");
   m.def("fp8_quantize_t", &fp8_quantize_t,
         "fused transpose-quantize (N,K) bf16 -> (K,N) e4m3");

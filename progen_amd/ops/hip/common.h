@@ -57,13 +57,24 @@ __device__ __forceinline__ float group16_max(float v) {
   return v;
 }
 
+// fast tanh on v_exp_f32 (libm tanhf is a slow ocml polynomial — the
+// round-2 profile put glu_bwd at 56% of HBM peak purely on VALU cost;
+// v_exp is ~2 cycles beside other work). (e-1)/(e+1) with the argument
+// clamped so exp never overflows to inf (inf/inf = NaN); |x| >= 15 is
+// tanh = +-1 to beyond fp32 precision anyway.
+__device__ __forceinline__ float fast_tanh(float x) {
+  float xc = fminf(fmaxf(x, -15.0f), 15.0f);
+  float e = __expf(2.0f * xc);
+  return (e - 1.0f) / (e + 1.0f);
+}
+
 // tanh-approximation GELU (matches jax.nn.gelu approximate=True and
 // torch F.gelu(approximate="tanh"))
 __device__ __forceinline__ float gelu_tanh(float x) {
   const float k0 = 0.7978845608028654f;  // sqrt(2/pi)
   const float k1 = 0.044715f;
   float inner = k0 * (x + k1 * x * x * x);
-  return 0.5f * x * (1.0f + tanhf(inner));
+  return 0.5f * x * (1.0f + fast_tanh(inner));
 }
 
 __device__ __forceinline__ float gelu_tanh_grad(float x) {
@@ -71,9 +82,23 @@ __device__ __forceinline__ float gelu_tanh_grad(float x) {
   const float k1 = 0.044715f;
   float x2 = x * x;
   float inner = k0 * (x + k1 * x * x2);
-  float t = tanhf(inner);
+  float t = fast_tanh(inner);
   float dinner = k0 * (1.0f + 3.0f * k1 * x2);
   return 0.5f * (1.0f + t) + 0.5f * x * (1.0f - t * t) * dinner;
+}
+
+// value AND derivative from ONE tanh (the backward needs both; two
+// separate calls were the other half of glu_bwd's VALU bill)
+__device__ __forceinline__ void gelu_tanh_both(float x, float* val,
+                                               float* grad) {
+  const float k0 = 0.7978845608028654f;
+  const float k1 = 0.044715f;
+  float x2 = x * x;
+  float inner = k0 * (x + k1 * x * x2);
+  float t = fast_tanh(inner);
+  float dinner = k0 * (1.0f + 3.0f * k1 * x2);
+  *val = 0.5f * x * (1.0f + t);
+  *grad = 0.5f * (1.0f + t) + 0.5f * x * (1.0f - t * t) * dinner;
 }
 
 #define HIP_CHECK_LAST()                                                      \

@@ -49,8 +49,10 @@ __global__ __launch_bounds__(GLU_BLOCK) void glu_bwd_kernel(
       float a = IS_BF16 ? bf2f(((short*)&va)[j]) : ((float*)&va)[j];
       float g = IS_BF16 ? bf2f(((short*)&vg)[j]) : ((float*)&vg)[j];
       float d = IS_BF16 ? bf2f(((short*)&vdy)[j]) : ((float*)&vdy)[j];
-      float rda = d * gelu_tanh(g);
-      float rdg = d * a * gelu_tanh_grad(g);
+      float gv, gg;
+      gelu_tanh_both(g, &gv, &gg);
+      float rda = d * gv;
+      float rdg = d * a * gg;
       if (IS_BF16) { ((short*)&da)[j] = f2bf(rda); ((short*)&dg)[j] = f2bf(rdg); }
       else { ((float*)&da)[j] = rda; ((float*)&dg)[j] = rdg; }
     }

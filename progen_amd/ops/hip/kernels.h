@@ -40,6 +40,8 @@ void fused_adamw_launch(float* master, void* params, const void* grads,
                         const float* clip_coef, bool is_bf16,
                         long long shard_off, hipStream_t stream);
 
+void colsum_launch(const void* dy, float* out, long long R, int C,
+                   hipStream_t stream);
 void quant_e4m3_launch(const void* in, void* out, const float* scale,
                        long long numel, hipStream_t stream);
 void quant_e4m3_t_launch(const void* in, void* out, const float* scale,

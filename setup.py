@@ -25,6 +25,7 @@ HIP_SOURCES = [
     "progen_amd/ops/hip/attention_bwd.hip",
     "progen_amd/ops/hip/sgu.hip",
     "progen_amd/ops/hip/fp8_quant.hip",
+    "progen_amd/ops/hip/colsum.hip",
 ]
 
 setup(

@@ -493,3 +493,13 @@ def test_cached_decode_matches_hip_forward():
     inc = torch.cat(rows, dim=0).float()
     # bf16 GEMV vs HIP MFMA paths across 3 layers
     assert rel_err(inc, full) < 6e-2
+
+
+def test_colsum_matches_torch():
+    """Bias-grad column sum (ops/hip/colsum.hip) vs fp32 torch."""
+    torch.manual_seed(23)
+    for R, C in [(65536, 1536), (4096, 12288), (1000, 8), (64, 256)]:
+        dy = (torch.randn(R, C, device=dev()) * 0.3).to(torch.bfloat16)
+        got = dispatch.ext().colsum(dy)
+        want = dy.float().sum(dim=0)
+        assert rel_err(got, want) < 1e-3, (R, C)
