@@ -30,6 +30,13 @@ import torch.nn.functional as F
 # an experiment flag for configurations with sparser backward phases.
 ENABLED = os.environ.get("PROGEN_OVERLAP_WGRAD", "0") == "1"
 
+# Route GPU linears through the custom Function even without the wgrad
+# side stream (gives the colsum dbias kernel; the backward GEMM
+# signatures differ from F.linear's autograd, so the TunableOp CSV must
+# cover them — progen_amd/tuned/). PROGEN_FN_LINEAR=0 reverts to plain
+# F.linear for A/B measurements.
+FN_LINEAR = os.environ.get("PROGEN_FN_LINEAR", "1") == "1"
+
 
 class WgradQueue:
     """Process-wide side stream for weight-gradient work."""
@@ -108,6 +115,6 @@ def overlap_linear(x: torch.Tensor, weight: torch.nn.Parameter,
     """F.linear through the custom Function on GPU (side-stream wgrad if
     PROGEN_OVERLAP_WGRAD=1, fast colsum dbias always); plain F.linear on
     CPU."""
-    if not x.is_cuda:
+    if not x.is_cuda or not (ENABLED or FN_LINEAR):
         return F.linear(x, weight, bias)
     return _OverlapLinearFn.apply(x, weight, bias)
