@@ -35,10 +35,13 @@ extern "C" {
 void colsum_launch(const void* dy, float* out, long long R, int C,
                    hipStream_t stream) {
   const int gx = (C + CS_COLS - 1) / CS_COLS;
-  // enough row-slices to fill the chip; atomics stay cheap (gy per col)
-  int gy = 1024 / gx;
-  if (gy < 1) gy = 1;
-  if (gy > 256) gy = 256;
+  // >> 256 blocks to fill the chip (the first cut capped at 256 = one
+  // 4-wave block per CU and ran latency-bound, SLOWER than torch's
+  // reduce); gy row-slices bound the fp32 atomics at gy per column
+  int gy = 4096 / gx;
+  if (gy < 64) gy = 64;
+  if (gy > 1024) gy = 1024;
+  if (gy > (int)R) gy = (int)R;
   dim3 grid(gx, gy);
   colsum_kernel<<<grid, CS_BLOCK, 0, stream>>>((const short*)dy, out, R, C);
 }
