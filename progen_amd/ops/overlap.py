@@ -30,12 +30,15 @@ import torch.nn.functional as F
 # an experiment flag for configurations with sparser backward phases.
 ENABLED = os.environ.get("PROGEN_OVERLAP_WGRAD", "0") == "1"
 
-# Route GPU linears through the custom Function even without the wgrad
-# side stream (gives the colsum dbias kernel; the backward GEMM
-# signatures differ from F.linear's autograd, so the TunableOp CSV must
-# cover them — progen_amd/tuned/). PROGEN_FN_LINEAR=0 reverts to plain
-# F.linear for A/B measurements.
-FN_LINEAR = os.environ.get("PROGEN_FN_LINEAR", "1") == "1"
+# Routing every GPU linear through the custom Function (for the colsum
+# dbias kernel) MEASURED 15% SLOWER end-to-end (108.4k -> 93.3k
+# tokens/s, gpurun_out/r02_call22.log) even though per-kernel time was
+# within 8 ms/step of the native path — torch's fused linear_backward
+# schedules the dgrad/wgrad pair better than the Function-built graph
+# replays. Default OFF; PROGEN_FN_LINEAR=1 re-enables for experiments
+# (the colsum kernel stays in use on the PROGEN_OVERLAP_WGRAD=1 side
+# stream, where the Function is required anyway).
+FN_LINEAR = os.environ.get("PROGEN_FN_LINEAR", "0") == "1"
 
 
 class WgradQueue:
