@@ -7,6 +7,16 @@ cd /root/repo
 mkdir -p gpurun_out
 {
 python tools/make_synthetic_fasta.py 20000
+cat > configs/data/synth.toml <<'TOML'
+read_from = "./synthetic.fasta"
+write_to = "./train_data"
+num_samples = 20000
+max_seq_len = 1024
+prob_invert_seq_annotation = 0.5
+fraction_valid_data = 0.025
+num_sequences_per_file = 100000
+sort_annotations = true
+TOML
 python generate_data.py --name synth
 echo "=== train 120 steps (eager, checkpoints) ==="
 timeout 700 python train.py --model_name small --mixed_precision \
