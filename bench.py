@@ -100,8 +100,13 @@ def main():
 
     graphed = None
     # hipGraph step: pure-replay only (profiles/r01_graph_interleave_bug.md).
-    # Single-GPU by default; PROGEN_GRAPH_DP=1 opts the DP step (incl. the
-    # RCCL all-reduce) into the capture.
+    # The DP step (incl. the RCCL all-reduce) is captured BY DEFAULT in the
+    # bench: the collective sequence is one fixed-size all-reduce per
+    # replay, symmetric across ranks by construction, and capture failure
+    # on any rank degrades every rank to eager via the MIN-reduce
+    # agreement below. PROGEN_GRAPH_DP=0 is the kill-switch (falls back
+    # to eager bucketed-overlap DP, the gloo-tested default of train.py).
+    os.environ.setdefault("PROGEN_GRAPH_DP", "1")
     want_graph = world == 1 or os.environ.get("PROGEN_GRAPH_DP") == "1"
     if on_gpu and want_graph and os.environ.get("PROGEN_NO_GRAPH") != "1":
         from progen_amd.runtime import GraphedTrainStep
