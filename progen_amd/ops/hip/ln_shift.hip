@@ -15,6 +15,37 @@
 #define LN_BLOCK 256
 #define LN_WAVES (LN_BLOCK / WAVE)
 
+// block-reduce two per-thread partial sums through a LN_WAVES-float
+// shared scratch (all threads must reach this together)
+__device__ __forceinline__ void ln_block_reduce2(float& a, float& b,
+                                                 float* red, int wid,
+                                                 int lane) {
+  a = wave_sum(a);
+  b = wave_sum(b);
+  if (lane == 0) red[wid] = a;
+  __syncthreads();
+  if (wid == 0) {
+    float t = (lane < LN_WAVES) ? red[lane] : 0.f;
+#pragma unroll
+    for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
+    if (lane == 0) red[0] = t;
+  }
+  __syncthreads();
+  a = red[0];
+  __syncthreads();
+  if (lane == 0) red[wid] = b;
+  __syncthreads();
+  if (wid == 0) {
+    float t = (lane < LN_WAVES) ? red[lane] : 0.f;
+#pragma unroll
+    for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
+    if (lane == 0) red[0] = t;
+  }
+  __syncthreads();
+  b = red[0];
+  __syncthreads();
+}
+
 // ---------------------------------------------------------------------------
 // forward
 // ---------------------------------------------------------------------------
@@ -36,6 +67,77 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_fwd_kernel(
   const long long base = (long long)row * Dv;
 
   __shared__ float red[LN_WAVES];
+  const int wid = threadIdx.x / WAVE;
+  const int lane = threadIdx.x % WAVE;
+  const int D = Dv * VLEN;
+  const int halfv = (D / 2) / VLEN;  // D even, half % VLEN == 0 (host-checked)
+
+  if (Dv <= LN_BLOCK) {
+    // fast path (every production D): each thread owns <= 1 vector, so
+    // the row stays in REGISTERS between the stats pass and the
+    // normalize pass — no second global read
+    const int i = threadIdx.x;
+    const bool act = i < Dv;
+    VEC v;
+    float s = 0.f, ss = 0.f;
+    if (act) {
+      v = x[base + i];
+      if (RES) {
+        VEC rv = res[base + i];
+        VEC sv;
+#pragma unroll
+        for (int j = 0; j < VLEN; ++j) {
+          if (IS_BF16) {
+            ((short*)&sv)[j] = f2bf(bf2f(((short*)&v)[j]) + bf2f(((short*)&rv)[j]));
+          } else {
+            ((float*)&sv)[j] = ((float*)&v)[j] + ((float*)&rv)[j];
+          }
+        }
+        s_out[base + i] = sv;
+        v = sv;
+      }
+#pragma unroll
+      for (int j = 0; j < VLEN; ++j) {
+        float f = IS_BF16 ? bf2f(((short*)&v)[j]) : ((float*)&v)[j];
+        s += f;
+        ss += f * f;
+      }
+    }
+    ln_block_reduce2(s, ss, red, wid, lane);
+    const float mu = s / D;
+    const float var = fmaxf(ss / D - mu * mu, 0.f);
+    const float rs = rsqrtf(var + eps);
+    if (threadIdx.x == 0) {
+      mean[row] = mu;
+      rstd[row] = rs;
+    }
+    if (act) {
+      VEC gw = g[i];
+      VEC o;
+#pragma unroll
+      for (int j = 0; j < VLEN; ++j) {
+        float f = IS_BF16 ? bf2f(((short*)&v)[j]) : ((float*)&v)[j];
+        float gj = IS_BF16 ? bf2f(((short*)&gw)[j]) : ((float*)&gw)[j];
+        float r = (f - mu) * rs * gj;
+        if (IS_BF16) ((short*)&o)[j] = f2bf(r);
+        else ((float*)&o)[j] = r;
+      }
+      if (!SHIFT || i >= halfv) {
+        y[base + i] = o;
+      } else if (n + 1 < N) {
+        y[base + Dv + i] = o;
+      }
+    }
+    if (SHIFT && n == 0) {
+      VEC z;
+#pragma unroll
+      for (int j = 0; j < VLEN; ++j) {
+        if (IS_BF16) ((short*)&z)[j] = 0; else ((float*)&z)[j] = 0.f;
+      }
+      for (int k = threadIdx.x; k < halfv; k += LN_BLOCK) y[base + k] = z;
+    }
+    return;
+  }
 
   float s = 0.f, ss = 0.f;
   for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
@@ -61,33 +163,8 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_fwd_kernel(
       ss += f * f;
     }
   }
-  s = wave_sum(s);
-  ss = wave_sum(ss);
-  const int wid = threadIdx.x / WAVE;
-  const int lane = threadIdx.x % WAVE;
-  if (lane == 0) red[wid] = s;
-  __syncthreads();
-  if (wid == 0) {
-    float t = (lane < LN_WAVES) ? red[lane] : 0.f;
-#pragma unroll
-    for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
-    if (lane == 0) red[0] = t;
-  }
-  __syncthreads();
-  s = red[0];
-  __syncthreads();
-  if (lane == 0) red[wid] = ss;
-  __syncthreads();
-  if (wid == 0) {
-    float t = (lane < LN_WAVES) ? red[lane] : 0.f;
-#pragma unroll
-    for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
-    if (lane == 0) red[0] = t;
-  }
-  __syncthreads();
-  ss = red[0];
+  ln_block_reduce2(s, ss, red, wid, lane);
 
-  const int D = Dv * VLEN;
   const float mu = s / D;
   const float var = fmaxf(ss / D - mu * mu, 0.f);
   const float rs = rsqrtf(var + eps);
@@ -96,7 +173,6 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_fwd_kernel(
     rstd[row] = rs;
   }
 
-  const int halfv = (D / 2) / VLEN;  // D even, half % VLEN == 0 (checked host-side)
   for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
     // with RES the summed row was just written: the re-read hits L1/L2
     VEC v = RES ? s_out[base + i] : x[base + i];
@@ -163,6 +239,58 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_bwd_kernel(
     const long long base = (long long)row * Dv;
     const float mu = mean[row];
     const float rs = rstd[row];
+    const float inv_d = 1.0f / D;
+
+    if (Dv <= LN_BLOCK) {
+      // fast path: each thread owns <= 1 vector; e/x/g stay in
+      // registers between the two phases — no second global read
+      const int i = threadIdx.x;
+      const bool act = i < Dv;
+      VEC e, v, gw, dsv;
+      bool have_e = false;
+      float a = 0.f, bsum = 0.f;
+      if (act) {
+        const bool shifted = SHIFT && (i < halfv);
+        have_e = true;
+        if (shifted) {
+          if (n + 1 < N) e = dy[base + Dv + i];
+          else have_e = false;
+        } else {
+          e = dy[base + i];
+        }
+        v = x[base + i];
+        gw = g[i];
+        if (DS) dsv = ds[base + i];
+#pragma unroll
+        for (int j = 0; j < VLEN; ++j) {
+          float ej = have_e ? (IS_BF16 ? bf2f(((short*)&e)[j]) : ((float*)&e)[j]) : 0.f;
+          float f = IS_BF16 ? bf2f(((short*)&v)[j]) : ((float*)&v)[j];
+          float gj = IS_BF16 ? bf2f(((short*)&gw)[j]) : ((float*)&gw)[j];
+          float xh = (f - mu) * rs;
+          a += ej * gj * xh;
+          bsum += ej * gj;
+        }
+      }
+      ln_block_reduce2(a, bsum, red, wid, lane);
+      if (act) {
+        VEC o;
+#pragma unroll
+        for (int j = 0; j < VLEN; ++j) {
+          float ej = have_e ? (IS_BF16 ? bf2f(((short*)&e)[j]) : ((float*)&e)[j]) : 0.f;
+          float f = IS_BF16 ? bf2f(((short*)&v)[j]) : ((float*)&v)[j];
+          float gj = IS_BF16 ? bf2f(((short*)&gw)[j]) : ((float*)&gw)[j];
+          float xh = (f - mu) * rs;
+          float dxv = rs * (ej * gj - bsum * inv_d - xh * a * inv_d);
+          if (DS) dxv += IS_BF16 ? bf2f(((short*)&dsv)[j]) : ((float*)&dsv)[j];
+          if (IS_BF16) ((short*)&o)[j] = f2bf(dxv);
+          else ((float*)&o)[j] = dxv;
+          dw_lds[i * VLEN + j] += ej * xh;  // thread-private index: no race
+        }
+        dx[base + i] = o;
+      }
+      __syncthreads();
+      continue;
+    }
 
     float a = 0.f, bsum = 0.f;  // a = sum(e*g*xhat), bsum = sum(e*g)
     for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
@@ -187,31 +315,8 @@ __global__ __launch_bounds__(LN_BLOCK) void ln_shift_bwd_kernel(
         bsum += ej * gj;
       }
     }
-    a = wave_sum(a);
-    bsum = wave_sum(bsum);
-    if (lane == 0) red[wid] = a;
-    __syncthreads();
-    if (wid == 0) {
-      float t = (lane < LN_WAVES) ? red[lane] : 0.f;
-#pragma unroll
-      for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
-      if (lane == 0) red[0] = t;
-    }
-    __syncthreads();
-    a = red[0];
-    __syncthreads();
-    if (lane == 0) red[wid] = bsum;
-    __syncthreads();
-    if (wid == 0) {
-      float t = (lane < LN_WAVES) ? red[lane] : 0.f;
-#pragma unroll
-      for (int off = LN_WAVES / 2; off > 0; off >>= 1) t += __shfl_xor(t, off, 64);
-      if (lane == 0) red[0] = t;
-    }
-    __syncthreads();
-    bsum = red[0];
+    ln_block_reduce2(a, bsum, red, wid, lane);
 
-    const float inv_d = 1.0f / D;
     for (int i = threadIdx.x; i < Dv; i += LN_BLOCK) {
       const bool shifted = SHIFT && (i < halfv);
       VEC e;
