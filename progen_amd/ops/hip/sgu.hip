@@ -29,6 +29,31 @@
 #define SGU_WAVES 4
 #define SGU_BLOCK (SGU_WAVES * WAVE)
 
+// XOR windows for tr-read images (derivations: wgrad probe header for
+// 512-B rows, attention_bwd.hip for 128-B rows)
+__device__ __forceinline__ int uk512(int k) { return (k & 3) | ((k & 8) >> 1); }
+__device__ __forceinline__ int uk128(int k) { return ((k & 2) >> 1) | ((k & 8) >> 2); }
+
+typedef __attribute__((__vector_size__(4 * sizeof(__bf16)))) __bf16 sgu_bf16x4t;
+#define SGU_AS3 __attribute__((address_space(3)))
+
+// one MFMA fragment from two ds_read_b64_tr_b16 of a row-major image
+__device__ __forceinline__ bf16x8 sgu_frag_tr(const char* img, int rowbytes,
+                                              int r1, int r2, int colb1,
+                                              int colb2) {
+  auto p1 = (SGU_AS3 sgu_bf16x4t*)(img + r1 * rowbytes + colb1);
+  auto p2 = (SGU_AS3 sgu_bf16x4t*)(img + r2 * rowbytes + colb2);
+  sgu_bf16x4t a = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p1);
+  sgu_bf16x4t b = __builtin_amdgcn_ds_read_tr16_b64_v4bf16(p2);
+  bf16x8 o;
+#pragma unroll
+  for (int j = 0; j < 4; ++j) {
+    ((__bf16*)&o)[j] = a[j];
+    ((__bf16*)&o)[j + 4] = b[j];
+  }
+  return o;
+}
+
 __device__ __forceinline__ int swz(int row, int byte_in_row) {
   return (byte_in_row ^ ((row & 7) << 4));
 }
@@ -161,8 +186,11 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_dgate_kernel(
   const long long bND = (long long)batch * N * D;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* tt_lds = smem;          // [d=64][m=64] transposed t tile, 8 KiB
-  char* wt_lds = smem + 8192 + wid * 8192;  // per-wave [k=64][m=64], 8 KiB
+  // LINEAR images read with ds_read_b64_tr_b16 (the old scatter-
+  // transposed staging was 64 b16 writes per thread per tile and ran
+  // the kernel at 71 TF/s — the wgrad-gen-1 disease):
+  char* tt_lds = smem;          // [m=64][d=64] linear, 128-B rows, 8 KiB
+  char* wt_lds = smem + 8192;   // [m=64][k=256] linear, 512-B rows, 32 KiB
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -175,6 +203,7 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_dgate_kernel(
   for (int t = mstart; t < N / 64; ++t) {
     __syncthreads();
     {
+      // t tile: one 16-B write per unit into the linear [m][d] image
 #pragma unroll
       for (int pass = 0; pass < 2; ++pass) {
         const int u = pass * SGU_BLOCK + (int)threadIdx.x;
@@ -182,19 +211,12 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_dgate_kernel(
         const int dd = (u & 7) * 8;
         bf16x8 v = *(const bf16x8*)(t_in + bND + (long long)(t * 64 + m) * D +
                                     d0 + dd);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d = dd + j;
-          *(short*)(tt_lds + d * 128 + swz(d, m * 2)) = ((short*)&v)[j];
-        }
+        *(bf16x8*)(tt_lds + m * 128 + ((dd * 2) ^ (uk128(m) * 32))) = v;
       }
     }
-    // W^T tiles for all 4 waves, staged cooperatively: the block reads
-    // W[t*64+mm][kblk*256 .. +256] with coalesced 16-B loads and
-    // scatter-transposes into the per-wave [k][m] images (the triu mask
-    // W[m][k]=0 for m<k applied at the write).
     {
-      // 64 m-rows x 32 k-groups(8) = 2048 units / 256 threads = 8 iters
+      // W tile (all 256 k for the 4 waves), triu mask at the write:
+      // one masked 16-B write per unit into the linear [m][k] image
 #pragma unroll
       for (int it = 0; it < 8; ++it) {
         const int u = it * SGU_BLOCK + (int)threadIdx.x;
@@ -203,13 +225,10 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_dgate_kernel(
         const int m = t * 64 + mm;
         const int kglob = kblk * 256 + k8;
         bf16x8 wv = *(const bf16x8*)(w + (long long)m * N + kglob);
-        char* wt_region = smem + 8192 + (k8 / 64) * 8192;
 #pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int k = (k8 + j) & 63;  // k index within the wave region
-          short v = (m >= kglob + j) ? ((short*)&wv)[j] : (short)0;
-          *(short*)(wt_region + k * 128 + swz(k, mm * 2)) = v;
-        }
+        for (int j = 0; j < 8; ++j)
+          if (m < kglob + j) ((short*)&wv)[j] = 0;
+        *(bf16x8*)(wt_lds + mm * 512 + ((k8 * 2) ^ (uk512(mm) * 32))) = wv;
       }
     }
     __syncthreads();
@@ -217,19 +236,26 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_dgate_kernel(
     if (t * 64 >= k0) {
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
+        const int r1 = ks * 32 + l4 * 8 + (l15 >> 2);
+        const int r2 = r1 + 4;
+        // B-fragments (t image, [d][m-contig]) hoisted: reused by the
+        // 4 A-fragments
+        bf16x8 bfr[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const int cb = (n * 16 + (l15 & 3) * 4) * 2;
+          bfr[n] = sgu_frag_tr(tt_lds, 128, r1, r2,
+                               cb ^ (uk128(r1) * 32), cb ^ (uk128(r2) * 32));
+        }
 #pragma unroll
         for (int m = 0; m < 4; ++m) {
-          const int krow = m * 16 + l15;
-          bf16x8 af = *(const bf16x8*)(wt_lds + krow * 128 +
-                                       swz(krow, (ks * 32 + 8 * l4) * 2));
+          const int cb = (wid * 64 + m * 16 + (l15 & 3) * 4) * 2;
+          bf16x8 af = sgu_frag_tr(wt_lds, 512, r1, r2,
+                                  cb ^ (uk512(r1) * 32), cb ^ (uk512(r2) * 32));
 #pragma unroll
-          for (int n = 0; n < 4; ++n) {
-            const int d = n * 16 + l15;
-            bf16x8 bf = *(const bf16x8*)(tt_lds + d * 128 +
-                                         swz(d, (ks * 32 + 8 * l4) * 2));
+          for (int n = 0; n < 4; ++n)
             acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af, bf, acc[m][n], 0, 0, 0);
-          }
+                af, bfr[n], acc[m][n], 0, 0, 0);
         }
       }
     }
@@ -329,7 +355,7 @@ void sgu_fwd_launch(const void* xa, const void* g_ln, const void* w,
 void sgu_dgate_launch(const void* t_in, const void* w, void* dg, int B, int N,
                       int D, hipStream_t stream) {
   dim3 grid(N / 256, D / 64, B), block(SGU_BLOCK);
-  size_t lds = 8192 + (size_t)SGU_WAVES * 8192;
+  size_t lds = 8192 + 32768;  // linear [m][d] t image + [m][k] W image
   sgu_dgate_kernel<<<grid, block, lds, stream>>>(
       (const short*)t_in, (const short*)w, (short*)dg, B, N, D);
 }
