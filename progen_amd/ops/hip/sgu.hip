@@ -84,7 +84,9 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_fwd_kernel(
   const long long bND = (long long)batch * N * D;
 
   extern __shared__ __attribute__((aligned(16))) char smem[];
-  char* gt_lds = smem;  // [d=64][k=64] bf16 transposed g tile, 8 KiB
+  // LINEAR [k=64][d=64] image (128-B rows), B-fragments by tr read —
+  // replaces the 16-b16-scatter-per-unit transposed staging
+  char* gt_lds = smem;  // 8 KiB
 
   f32x4 acc[4][4];
 #pragma unroll
@@ -107,11 +109,7 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_fwd_kernel(
         const int dd = (u & 7) * 8;
         bf16x8 v = *(const bf16x8*)(g_ln + bND + (long long)(t * 64 + k) * D +
                                     d0 + dd);
-#pragma unroll
-        for (int j = 0; j < 8; ++j) {
-          const int d = dd + j;
-          *(short*)(gt_lds + d * 128 + swz(d, k * 2)) = ((short*)&v)[j];
-        }
+        *(bf16x8*)(gt_lds + k * 128 + ((dd * 2) ^ (uk128(k) * 32))) = v;
       }
     }
     __syncthreads();
@@ -120,6 +118,17 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_fwd_kernel(
       const bool diag = (t * 64) > m0 - 64 && (t * 64) <= m0 + 63;  // overlaps
 #pragma unroll
       for (int ks = 0; ks < 2; ++ks) {
+        // B-fragments [d][k-contig] by tr read of the linear [k][d]
+        // image, hoisted for reuse across the 4 A-fragments
+        const int r1 = ks * 32 + l4 * 8 + (l15 >> 2);
+        const int r2 = r1 + 4;
+        bf16x8 bfr[4];
+#pragma unroll
+        for (int n = 0; n < 4; ++n) {
+          const int cb = (n * 16 + (l15 & 3) * 4) * 2;
+          bfr[n] = sgu_frag_tr(gt_lds, 128, r1, r2,
+                               cb ^ (uk128(r1) * 32), cb ^ (uk128(r2) * 32));
+        }
 #pragma unroll
         for (int m = 0; m < 4; ++m) {
           // A-fragment: W[m0+m*16+l15][t*64 + ks*32 + 8*l4 .. +8] bf16
@@ -132,13 +141,9 @@ __global__ __launch_bounds__(SGU_BLOCK) void sgu_fwd_kernel(
               if (kk0 + j > row) ((short*)&af)[j] = 0;
           }
 #pragma unroll
-          for (int n = 0; n < 4; ++n) {
-            const int d = n * 16 + l15;
-            bf16x8 bf = *(const bf16x8*)(gt_lds + d * 128 +
-                                         swz(d, (ks * 32 + 8 * l4) * 2));
+          for (int n = 0; n < 4; ++n)
             acc[m][n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(
-                af, bf, acc[m][n], 0, 0, 0);
-          }
+                af, bfr[n], acc[m][n], 0, 0, 0);
         }
       }
     }
