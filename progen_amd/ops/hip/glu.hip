@@ -7,15 +7,16 @@
 
 #define GLU_BLOCK 256
 
+// 2-D grid (columns x row-stripes): the flat-index form cost two
+// 64-bit integer divisions per iteration — a long VALU sequence on a
+// kernel this loop-dense
 template <typename VEC, bool IS_BF16>
 __global__ __launch_bounds__(GLU_BLOCK) void glu_fwd_kernel(
     const VEC* __restrict__ h, VEC* __restrict__ y, long long rows, int Hv) {
   constexpr int VLEN = IS_BF16 ? 8 : 4;
-  const long long total = rows * Hv;
-  for (long long idx = blockIdx.x * (long long)GLU_BLOCK + threadIdx.x;
-       idx < total; idx += (long long)gridDim.x * GLU_BLOCK) {
-    const long long row = idx / Hv;
-    const int i = (int)(idx % Hv);
+  const int i = blockIdx.x * GLU_BLOCK + (int)threadIdx.x;
+  if (i >= Hv) return;
+  for (long long row = blockIdx.y; row < rows; row += gridDim.y) {
     VEC va = h[row * 2 * Hv + i];
     VEC vg = h[row * 2 * Hv + Hv + i];
     VEC o;
@@ -26,7 +27,7 @@ __global__ __launch_bounds__(GLU_BLOCK) void glu_fwd_kernel(
       float r = a * gelu_tanh(g);
       if (IS_BF16) ((short*)&o)[j] = f2bf(r); else ((float*)&o)[j] = r;
     }
-    y[idx] = o;
+    y[row * Hv + i] = o;
   }
 }
 
@@ -35,14 +36,12 @@ __global__ __launch_bounds__(GLU_BLOCK) void glu_bwd_kernel(
     const VEC* __restrict__ dy, const VEC* __restrict__ h,
     VEC* __restrict__ dh, long long rows, int Hv) {
   constexpr int VLEN = IS_BF16 ? 8 : 4;
-  const long long total = rows * Hv;
-  for (long long idx = blockIdx.x * (long long)GLU_BLOCK + threadIdx.x;
-       idx < total; idx += (long long)gridDim.x * GLU_BLOCK) {
-    const long long row = idx / Hv;
-    const int i = (int)(idx % Hv);
+  const int i = blockIdx.x * GLU_BLOCK + (int)threadIdx.x;
+  if (i >= Hv) return;
+  for (long long row = blockIdx.y; row < rows; row += gridDim.y) {
     VEC va = h[row * 2 * Hv + i];
     VEC vg = h[row * 2 * Hv + Hv + i];
-    VEC vdy = dy[idx];
+    VEC vdy = dy[row * Hv + i];
     VEC da, dg;
 #pragma unroll
     for (int j = 0; j < VLEN; ++j) {
@@ -106,17 +105,27 @@ static inline int glu_grid(long long total) {
   return (int)g;
 }
 
+// 2-D grid for the split (glu) kernels: x covers the Hv columns, y
+// row-stripes sized so x*y lands near 4096 blocks
+static inline dim3 glu_grid2(long long rows, int Hv) {
+  int gx = (Hv + GLU_BLOCK - 1) / GLU_BLOCK;
+  long long gy = 4096 / gx;
+  if (gy < 1) gy = 1;
+  if (gy > rows) gy = rows;
+  return dim3(gx, (unsigned)gy);
+}
+
 extern "C" {
 
 void glu_fwd_launch(const void* h, void* y, long long rows, int H,
                     bool is_bf16, hipStream_t stream) {
   if (is_bf16) {
     int Hv = H / 8;
-    glu_fwd_kernel<bf16x8, true><<<glu_grid(rows * Hv), GLU_BLOCK, 0, stream>>>(
+    glu_fwd_kernel<bf16x8, true><<<glu_grid2(rows, Hv), GLU_BLOCK, 0, stream>>>(
         (const bf16x8*)h, (bf16x8*)y, rows, Hv);
   } else {
     int Hv = H / 4;
-    glu_fwd_kernel<f32x4, false><<<glu_grid(rows * Hv), GLU_BLOCK, 0, stream>>>(
+    glu_fwd_kernel<f32x4, false><<<glu_grid2(rows, Hv), GLU_BLOCK, 0, stream>>>(
         (const f32x4*)h, (f32x4*)y, rows, Hv);
   }
 }
@@ -125,11 +134,11 @@ void glu_bwd_launch(const void* dy, const void* h, void* dh, long long rows,
                     int H, bool is_bf16, hipStream_t stream) {
   if (is_bf16) {
     int Hv = H / 8;
-    glu_bwd_kernel<bf16x8, true><<<glu_grid(rows * Hv), GLU_BLOCK, 0, stream>>>(
+    glu_bwd_kernel<bf16x8, true><<<glu_grid2(rows, Hv), GLU_BLOCK, 0, stream>>>(
         (const bf16x8*)dy, (const bf16x8*)h, (bf16x8*)dh, rows, Hv);
   } else {
     int Hv = H / 4;
-    glu_bwd_kernel<f32x4, false><<<glu_grid(rows * Hv), GLU_BLOCK, 0, stream>>>(
+    glu_bwd_kernel<f32x4, false><<<glu_grid2(rows, Hv), GLU_BLOCK, 0, stream>>>(
         (const f32x4*)dy, (const f32x4*)h, (f32x4*)dh, rows, Hv);
   }
 }
