@@ -511,15 +511,16 @@ __global__ __launch_bounds__(256) void attn_bwd_finalize_kernel(
     short* __restrict__ dqkv, int B, int N, int H, int wsz) {
   const long long HD3 = 3LL * H * DH;
   const long long HD2 = 2LL * H * DH;
-  const long long total = (long long)B * N * 3 * H * (DH / 8);
-  for (long long idx = blockIdx.x * 256LL + threadIdx.x; idx < total;
-       idx += (long long)gridDim.x * 256) {
-    const int g = idx % (DH / 8);
-    const long long rest = idx / (DH / 8);
-    const int hslot = rest % (3 * H);
-    const long long bn = rest / (3 * H);
-    const int n = bn % N;
-    const int d0 = g * 8;
+  // 2-D grid: y = hslot (3H values), x stripes (bn, g) — the flat-index
+  // form paid two 64-bit divisions per iteration (dividing by the
+  // non-power-of-two 3H), a long VALU chain on a memory-bound kernel
+  const int hslot = blockIdx.y;
+  const long long BN = (long long)B * N;
+  const int g = (int)threadIdx.x & 7;
+  const int d0 = g * 8;
+  for (long long bn = blockIdx.x * 32LL + ((int)threadIdx.x >> 3); bn < BN;
+       bn += (long long)gridDim.x * 32) {
+    const int n = (int)(bn % N);
 
     const long long off = bn * HD3 + (long long)hslot * DH + d0;
     float x[8], sv[8], cv[8];
@@ -561,9 +562,9 @@ void attn_bwd_launch(const void* dout, const void* qkv, const void* halo,
   attn_bwd_kernel<<<grid, block, lds, stream>>>(
       (const short*)dout, (const short*)qkv, (const short*)halo,
       (const short*)out, lse, dacc, dlook, dhalo, B, N, H, wsz);
-  long long total = (long long)B * N * 3 * H * (DH / 8);
-  int fin_grid = (int)((total + 255) / 256);
-  if (fin_grid > 2048) fin_grid = 2048;
+  long long bn_stripes = ((long long)B * N + 31) / 32;
+  int gx = (int)(bn_stripes < 512 ? bn_stripes : 512);
+  dim3 fin_grid(gx, 3 * H);
   attn_bwd_finalize_kernel<<<fin_grid, 256, 0, stream>>>(
       dacc, dlook, rsin, rcos, (short*)dqkv, B, N, H, wsz);
 }

@@ -16,18 +16,17 @@ __global__ __launch_bounds__(256) void rope_qkv_kernel(
     const float* __restrict__ rsin, const float* __restrict__ rcos,
     short* __restrict__ qkv_rot, int B, int N, int H) {
   const long long HD3 = 3LL * H * DH;
-  const long long total = (long long)B * N * 3 * H * (DH / 8);
-  for (long long idx = blockIdx.x * 256LL + threadIdx.x; idx < total;
-       idx += (long long)gridDim.x * 256) {
-    const int g = idx % (DH / 8);
-    const long long rest = idx / (DH / 8);
-    const int hslot = rest % (3 * H);
-    const long long bn = rest / (3 * H);
-    const int n = bn % N;
-    const int d0 = g * 8;
-    (void)hslot;
+  // 2-D grid (y = hslot, x stripes bn x g): the flat-index form paid
+  // two 64-bit divisions by the non-power-of-two 3H per iteration
+  const int hslot = blockIdx.y;
+  const long long BN = (long long)B * N;
+  const int g = (int)threadIdx.x & 7;
+  const int d0 = g * 8;
+  for (long long bn = blockIdx.x * 32LL + ((int)threadIdx.x >> 3); bn < BN;
+       bn += (long long)gridDim.x * 32) {
+    const int n = (int)(bn % N);
 
-    const long long off = bn * HD3 + (long long)(rest % (3 * H)) * DH + d0;
+    const long long off = bn * HD3 + (long long)hslot * DH + d0;
     bf16x8 v = *(const bf16x8*)(qkv + off);
     float x[8], sv[8], cv[8];
 #pragma unroll
@@ -54,9 +53,9 @@ extern "C" {
 
 void rope_qkv_launch(const void* qkv, const float* rsin, const float* rcos,
                      void* qkv_rot, int B, int N, int H, hipStream_t stream) {
-  long long total = (long long)B * N * 3 * H * (DH / 8);
-  int grid = (int)((total + 255) / 256);
-  if (grid > 4096) grid = 4096;
+  long long bn_stripes = ((long long)B * N + 31) / 32;
+  int gx = (int)(bn_stripes < 512 ? bn_stripes : 512);
+  dim3 grid(gx, 3 * H);
   rope_qkv_kernel<<<grid, 256, 0, stream>>>((const short*)qkv, rsin, rcos,
                                             (short*)qkv_rot, B, N, H);
 }
