@@ -511,15 +511,16 @@ __global__ __launch_bounds__(256) void attn_bwd_finalize_kernel(
     short* __restrict__ dqkv, int B, int N, int H, int wsz) {
   const long long HD3 = 3LL * H * DH;
   const long long HD2 = 2LL * H * DH;
-  // 2-D grid: y = hslot (3H values), x stripes (bn, g) — the flat-index
-  // form paid two 64-bit divisions per iteration (dividing by the
-  // non-power-of-two 3H), a long VALU chain on a memory-bound kernel
-  const int hslot = blockIdx.y;
-  const long long BN = (long long)B * N;
-  const int g = (int)threadIdx.x & 7;
+  // grid.x covers one row's 3H*8 vector units contiguously (hslot/g by
+  // shift+mask: no 64-bit division, no scattered access — see
+  // rope_qkv.hip), grid.y stripes the (B,N) rows
+  const int u = blockIdx.x * 256 + (int)threadIdx.x;
+  if (u >= 3 * H * (DH / 8)) return;
+  const int hslot = u >> 3;
+  const int g = u & 7;
   const int d0 = g * 8;
-  for (long long bn = blockIdx.x * 32LL + ((int)threadIdx.x >> 3); bn < BN;
-       bn += (long long)gridDim.x * 32) {
+  const long long BN = (long long)B * N;
+  for (long long bn = blockIdx.y; bn < BN; bn += gridDim.y) {
     const int n = (int)(bn % N);
 
     const long long off = bn * HD3 + (long long)hslot * DH + d0;
@@ -562,9 +563,10 @@ void attn_bwd_launch(const void* dout, const void* qkv, const void* halo,
   attn_bwd_kernel<<<grid, block, lds, stream>>>(
       (const short*)dout, (const short*)qkv, (const short*)halo,
       (const short*)out, lse, dacc, dlook, dhalo, B, N, H, wsz);
-  long long bn_stripes = ((long long)B * N + 31) / 32;
-  int gx = (int)(bn_stripes < 512 ? bn_stripes : 512);
-  dim3 fin_grid(gx, 3 * H);
+  int gx = (3 * H * (DH / 8) + 255) / 256;
+  long long bn_tot = (long long)B * N;
+  int gy = (int)(bn_tot < 2048 ? bn_tot : 2048);
+  dim3 fin_grid(gx, gy);
   attn_bwd_finalize_kernel<<<fin_grid, 256, 0, stream>>>(
       dacc, dlook, rsin, rcos, (short*)dqkv, B, N, H, wsz);
 }

@@ -16,14 +16,19 @@ __global__ __launch_bounds__(256) void rope_qkv_kernel(
     const float* __restrict__ rsin, const float* __restrict__ rcos,
     short* __restrict__ qkv_rot, int B, int N, int H) {
   const long long HD3 = 3LL * H * DH;
-  // 2-D grid (y = hslot, x stripes bn x g): the flat-index form paid
-  // two 64-bit divisions by the non-power-of-two 3H per iteration
-  const int hslot = blockIdx.y;
-  const long long BN = (long long)B * N;
-  const int g = (int)threadIdx.x & 7;
+  // grid.x covers one (B,N)-row's 3H*8 vector units CONTIGUOUSLY
+  // (hslot/g by shift+mask — the old flat form paid two 64-bit
+  // divisions by the non-power-of-two 3H per iteration; a first 2-D
+  // rework put g in the low lane bits and measured 23% SLOWER from the
+  // scattered access, so the in-row order stays sequential), grid.y
+  // stripes the rows
+  const int u = blockIdx.x * 256 + (int)threadIdx.x;
+  if (u >= 3 * H * (DH / 8)) return;
+  const int hslot = u >> 3;
+  const int g = u & 7;
   const int d0 = g * 8;
-  for (long long bn = blockIdx.x * 32LL + ((int)threadIdx.x >> 3); bn < BN;
-       bn += (long long)gridDim.x * 32) {
+  const long long BN = (long long)B * N;
+  for (long long bn = blockIdx.y; bn < BN; bn += gridDim.y) {
     const int n = (int)(bn % N);
 
     const long long off = bn * HD3 + (long long)hslot * DH + d0;
@@ -53,9 +58,10 @@ extern "C" {
 
 void rope_qkv_launch(const void* qkv, const float* rsin, const float* rcos,
                      void* qkv_rot, int B, int N, int H, hipStream_t stream) {
-  long long bn_stripes = ((long long)B * N + 31) / 32;
-  int gx = (int)(bn_stripes < 512 ? bn_stripes : 512);
-  dim3 grid(gx, 3 * H);
+  int gx = (3 * H * (DH / 8) + 255) / 256;
+  long long bn = (long long)B * N;
+  int gy = (int)(bn < 2048 ? bn : 2048);
+  dim3 grid(gx, gy);
   rope_qkv_kernel<<<grid, 256, 0, stream>>>((const short*)qkv, rsin, rcos,
                                             (short*)qkv_rot, B, N, H);
 }
