@@ -33,10 +33,12 @@
 //     dS A-fragments come from ds_read_b64_tr_b16 transposed reads of
 //     the wave's OWN just-written dS^T region (same-wave DS ordering is
 //     program order).
-//   barrier A; tile t+1's k/v/kt staging overlaps the dV/dK slices;
-//   barrier B1 (slice reads done) -> upper-half partials to LDS
-//   scratch; barrier B2 -> lower half combines + stores; barrier B3
-//   (scratch reads done before t+1's dS^T writes).
+//   barrier A; tile t+1's k/v/kt staging overlaps the dV/dK slices
+//   (each wave owns a 16-key x 32-dh output slice — the 8 waves split
+//   a slice by DH-half, summing the full K-dim, so outputs stay
+//   disjoint with no cross-wave combine); barrier B. Two barriers per
+//   tile (an earlier ks-half split needed an LDS partial combine and
+//   four barriers; PMC showed 53% of cycles parked on them).
 // Every dV/dK element is produced by exactly ONE wave, so the stores
 // are PLAIN (no atomics): each window's own-band gradients go to dacc,
 // its lookback-band gradients to a separate dlook buffer, and
@@ -394,95 +396,83 @@ __global__ __launch_bounds__(V5_BLOCK) void attn_bwd_kernel(
 
       if (t + 1 < tiles) write_lds();
 
-      // ---- phase 2: dV+dK key slices, K-dim split by ks across wave
-      // halves (half 1 accumulates into dacc2/dlook2) ----
+      // ---- phase 2: dV+dK key slices. The 8 waves split each 16-key
+      // slice by DH-HALF (32 channels each), every wave summing the
+      // FULL K-dim — outputs are disjoint, so the stores stay plain
+      // and no cross-wave partial combine (or its two extra barriers)
+      // is needed (the earlier ks-half split parked 53% of cycles on
+      // its 4-barriers-per-tile schedule). ----
       {
-        const int ks = wid >> 2;          // fixed ks half per wave
+        const int dh0 = (wid >> 2) * 2;   // wave's 2 d-blocks (32 dh)
         const int keyslot = (wid & 3) * 16 + l15;
-        f32x4 dv[4], dk[4];
+        f32x4 dv[2], dk[2];
 #pragma unroll
-        for (int n = 0; n < 4; ++n) {
+        for (int n = 0; n < 2; ++n) {
           dv[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
           dk[n] = (f32x4){0.f, 0.f, 0.f, 0.f};
         }
         __builtin_amdgcn_s_setprio(1);
-        const int r0 = ks * 32 + 8 * l4;
         for (int c = c_min; c < nactive; ++c) {
           char* pds_c = pds_base + c * 8192;
           char* ds2_c = ds2_base + c * 8192;
           char* dot_c = dot_base + c * 8192;
           char* qt_c = qt_base + c * 8192;
-          bf16x8 pf = *(const bf16x8*)(pds_c + keyslot * 128 +
-                                       swz(keyslot, r0 * 2));
-          bf16x8 dsf = *(const bf16x8*)(ds2_c + keyslot * 128 +
-                                        ((r0 * 2) ^ (uk4(keyslot) * 32)));
 #pragma unroll
-          for (int n = 0; n < 4; ++n) {
-            const int d = n * 16 + l15;
-            bf16x8 dof = *(const bf16x8*)(dot_c + d * 128 + swz(d, r0 * 2));
-            bf16x8 qf = *(const bf16x8*)(qt_c + d * 128 + swz(d, r0 * 2));
-            dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
-                                                            0, 0, 0);
-            dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, dk[n],
-                                                            0, 0, 0);
+          for (int ks = 0; ks < 2; ++ks) {
+            const int r0 = ks * 32 + 8 * l4;
+            bf16x8 pf = *(const bf16x8*)(pds_c + keyslot * 128 +
+                                         swz(keyslot, r0 * 2));
+            bf16x8 dsf = *(const bf16x8*)(ds2_c + keyslot * 128 +
+                                          ((r0 * 2) ^ (uk4(keyslot) * 32)));
+#pragma unroll
+            for (int n = 0; n < 2; ++n) {
+              const int d = (dh0 + n) * 16 + l15;
+              bf16x8 dof = *(const bf16x8*)(dot_c + d * 128 + swz(d, r0 * 2));
+              bf16x8 qf = *(const bf16x8*)(qt_c + d * 128 + swz(d, r0 * 2));
+              dv[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(pf, dof, dv[n],
+                                                              0, 0, 0);
+              dk[n] = __builtin_amdgcn_mfma_f32_16x16x32_bf16(dsf, qf, dk[n],
+                                                              0, 0, 0);
+            }
           }
         }
         __builtin_amdgcn_s_setprio(0);
-        __syncthreads();  // B1: slices done reading pds/ds2 everywhere
-        // half 1 parks its partial in the (now dead) ds2 region scratch:
-        // [slot = key-slice][key16 x dh64 f32] x2 (dv, dk)
-        float* scratch = (float*)(ds2_base) + (wid & 3) * 2048;
-        if (ks == 1) {
+        const bool lookback = kb < wsz;
 #pragma unroll
-          for (int n = 0; n < 4; ++n)
+        for (int r = 0; r < 4; ++r) {
+          const int kpos = (window - 1) * wsz + kb + (wid & 3) * 16 + l4 * 4 + r;
+          float* dstv;
+          float* dstk;
+          if (kpos >= 0) {
+            dstv = lookback
+                ? dlook + look_bn + (long long)kpos * HD2 + lv_off
+                : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
+            dstk = lookback
+                ? dlook + look_bn + (long long)kpos * HD2 + lk_off
+                : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
+          } else if (dhalo != nullptr) {
+            const long long hb =
+                ((long long)batch * wsz + (kpos + wsz)) * HD2;
+            dstv = dhalo + hb + lv_off;
+            dstk = dhalo + hb + lk_off;
+          } else {
+            continue;  // window-0 zero-pad quirk: grads discarded
+          }
 #pragma unroll
-            for (int r = 0; r < 4; ++r) {
-              scratch[(l4 * 4 + r) * 64 + n * 16 + l15] = ((float*)&dv[n])[r];
-              scratch[1024 + (l4 * 4 + r) * 64 + n * 16 + l15] =
-                  ((float*)&dk[n])[r];
+          for (int n = 0; n < 2; ++n) {
+            const int dcol = (dh0 + n) * 16 + l15;
+            float vv = ((float*)&dv[n])[r];
+            float vk = ((float*)&dk[n])[r];
+            if (round > 0) {
+              vv += dstv[dcol];
+              vk += dstk[dcol];
             }
-        }
-        __syncthreads();  // B2: partials visible
-        if (ks == 0) {
-          const bool lookback = kb < wsz;
-#pragma unroll
-          for (int r = 0; r < 4; ++r) {
-            const int kpos = (window - 1) * wsz + kb + (wid & 3) * 16 + l4 * 4 + r;
-            float* dstv;
-            float* dstk;
-            if (kpos >= 0) {
-              dstv = lookback
-                  ? dlook + look_bn + (long long)kpos * HD2 + lv_off
-                  : dacc + qkv_bn + (long long)kpos * HD3 + v_off;
-              dstk = lookback
-                  ? dlook + look_bn + (long long)kpos * HD2 + lk_off
-                  : dacc + qkv_bn + (long long)kpos * HD3 + k_off;
-            } else if (dhalo != nullptr) {
-              const long long hb =
-                  ((long long)batch * wsz + (kpos + wsz)) * HD2;
-              dstv = dhalo + hb + lv_off;
-              dstk = dhalo + hb + lk_off;
-            } else {
-              continue;  // window-0 zero-pad quirk: grads discarded
-            }
-#pragma unroll
-            for (int n = 0; n < 4; ++n) {
-              float vv = ((float*)&dv[n])[r] +
-                         scratch[(l4 * 4 + r) * 64 + n * 16 + l15];
-              float vk = ((float*)&dk[n])[r] +
-                         scratch[1024 + (l4 * 4 + r) * 64 + n * 16 + l15];
-              if (round > 0) {
-                vv += dstv[n * 16 + l15];
-                vk += dstk[n * 16 + l15];
-              }
-              dstv[n * 16 + l15] = vv;
-              dstk[n * 16 + l15] = vk;
-            }
+            dstv[dcol] = vv;
+            dstk[dcol] = vk;
           }
         }
       }
-      __syncthreads();  // barrier B3: scratch reads done before t+1's
-                        // phase-1 overwrites ds2
+      __syncthreads();  // barrier B: slices done before t+1's phase 1
     }
 
     if (active) {
