@@ -21,13 +21,11 @@ __device__ __forceinline__ float bf2f(short u) {
 }
 
 __device__ __forceinline__ short f2bf(float f) {
-  union { float f; unsigned int i; } c;
-  c.f = f;
-  unsigned int x = c.i;
-  // round-to-nearest-even like CDNA v_cvt and torch
-  unsigned int lsb = (x >> 16) & 1u;
-  x += 0x7fffu + lsb;
-  return (short)(x >> 16);
+  // __float2bfloat16 lowers to ONE v_cvt_pk_bf16_f32 (RNE, same
+  // rounding as torch); the manual bit-math RNE this replaces was
+  // three VALU ops in every kernel's store path
+  __hip_bfloat16 b = __float2bfloat16(f);
+  return *(short*)&b;
 }
 
 // wave-wide reductions (64 lanes)
