@@ -265,15 +265,19 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
             if wandb is not None:
                 wandb.log({'loss': loss_val, 'tokens_per_sec': toks_per_sec})
 
-        # checkpointing: a live hipGraph forbids eager work between
-        # replays (pure-replay rule), so the graph is dropped first and
-        # re-captured after the save (--graph used to skip
-        # checkpoints entirely, losing all progress on interruption)
+        # checkpointing / validation / sampling are eager work, and a
+        # live hipGraph forbids eager kernels between replays (pure-replay
+        # rule) — so the graph is dropped once, all due eager sections
+        # run, and ONE re-capture follows (--graph used to
+        # skip checkpoints entirely, losing all progress on interruption)
         do_ckpt = i % checkpoint_every == 0 and (graphed is None or i > 0)
+        do_valid = i % validate_every == 0 and (graphed is None or i > 0)
+        do_sample = i % sample_every == 0 and (graphed is None or i > 0)
+        was_graphed = graphed is not None
+        if (do_ckpt or do_valid or do_sample) and was_graphed:
+            torch.cuda.synchronize()
+            graphed = None  # free replay state before eager/D2H work
         if do_ckpt:
-            if graphed is not None:
-                torch.cuda.synchronize()
-                graphed = None  # free replay state before eager/D2H work
             # ZeRO-1's state_dict all-gathers shards: collective call
             optim_sd = optim.state_dict() \
                 if (is_main or getattr(optim, 'state_dict_is_collective',
@@ -290,10 +294,8 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
                 save_checkpoint(package, checkpoint_keep_n)
                 print(f"checkpoint to start at sequence index of "
                       f"{package['next_seq_index']}")
-            if graph_ok:
-                graphed = make_graphed()
 
-        if graphed is None and i % validate_every == 0:
+        if do_valid:
             valid_data = my_shard(next(valid_dataset))
             with torch.no_grad():
                 vloss = compute_loss(module, valid_data)
@@ -303,7 +305,7 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
                 if wandb is not None:
                     wandb.log({'valid_loss': vloss_val})
 
-        if graphed is None and i % sample_every == 0 and is_main:
+        if do_sample and is_main:
             valid_data = my_shard(next(valid_dataset))[0]
             prime = valid_data[:prime_length]
             prime_str = decode_tokens(prime.cpu().numpy())
@@ -324,6 +326,9 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
                         f'<div style="overflow-wrap: break-word;">'
                         f"{sampled_str}</div>")
                 wandb.log({'samples': wandb.Html(html)})
+
+        if was_graphed and graphed is None and graph_ok:
+            graphed = make_graphed()
 
     # synchronized teardown: without the barrier a fast rank can exit and
     # close its gloo/RCCL connections while a slower rank is still inside
