@@ -90,3 +90,29 @@ def test_train_cli_world2_gloo(tmp_path):
     assert "loss:" not in outs[1]
     ckpts = list((tmp_path / "ckpts").glob("ckpt_*.pkl"))
     assert len(ckpts) >= 1
+
+
+@pytest.mark.timeout(300)
+def test_graph_flag_says_why_disabled(tmp_path):
+    """--graph must NEVER silently run eager: the r02 evidence scripts
+    lost the graphed path for days because --grad_accum_every defaults
+    to 4 and graph_ok quietly became False."""
+    repo = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+    _make_data(tmp_path)
+    subprocess.run(
+        [sys.executable, os.path.join(repo, "generate_data.py"),
+         "--data_dir", "./configs/data", "--name", "tiny"],
+        cwd=tmp_path, check=True, capture_output=True, timeout=120)
+    out = subprocess.run(
+        [sys.executable, os.path.join(repo, "train.py"),
+         "--config_path", "./configs/model", "--model_name", "tiny",
+         "--data_path", "./train_data", "--batch_size", "2",
+         "--graph", "--wandb_off", "--yes",
+         "--checkpoint_every", "100000", "--validate_every", "100000",
+         "--sample_every", "100000", "--max_steps", "1"],
+        cwd=tmp_path, env=dict(os.environ, PYTHONPATH=repo),
+        capture_output=True, text=True, timeout=240)
+    assert out.returncode == 0, out.stdout[-2000:] + out.stderr[-2000:]
+    assert "--graph disabled (" in out.stdout
+    assert "no GPU" in out.stdout
+    assert "--grad_accum_every 4 != 1" in out.stdout

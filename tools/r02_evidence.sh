@@ -30,7 +30,7 @@ timeout 400 python train.py --model_name small --mixed_precision \
   --data_path ./train_data 2>&1 | grep -E "loss|starting|params" | head -12
 echo "=== graphed train WITH in-loop checkpoint (drop+recapture) ==="
 timeout 500 python train.py --model_name small --mixed_precision \
-  --batch_size 32 --max_steps 25 --checkpoint_every 10 --graph \
+  --batch_size 32 --max_steps 25 --checkpoint_every 10 --graph --grad_accum_every 1 \
   --validate_every 1000 --sample_every 1000 --wandb_off \
   --data_path ./train_data 2>&1 | grep -E "loss|checkpoint|captured|starting" | head -30
 echo "=== sample from trained ckpt (graphed decode) ==="

@@ -19,7 +19,7 @@ TOML
 python generate_data.py --name synth
 echo "=== graphed train: ckpt_every=8 valid_every=5 sample_every=10 ==="
 timeout 500 python train.py --model_name small --mixed_precision \
-  --batch_size 32 --max_steps 22 --checkpoint_every 8 --graph \
+  --batch_size 32 --max_steps 22 --checkpoint_every 8 --graph --grad_accum_every 1 \
   --validate_every 5 --sample_every 10 --wandb_off --yes --new \
   --data_path ./train_data 2>&1 | \
   grep -E "loss|checkpoint|captured|valid_loss|\*{10}|starting" | head -60

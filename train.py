@@ -183,6 +183,20 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
     # fwd+bwd(+optimizer) sequence as one graph when shapes are static
     graph_ok = graph and device.type == 'cuda' and grad_accum_every == 1 \
         and grad_accum_mode == 'sum' and world == 1
+    if graph and not graph_ok and is_main:
+        # say WHY, loudly: --graph silently running eager hides a 15%+
+        # perf loss (found the hard way: --grad_accum_every defaults to 4)
+        why = []
+        if device.type != 'cuda':
+            why.append('no GPU')
+        if grad_accum_every != 1:
+            why.append(f'--grad_accum_every {grad_accum_every} != 1')
+        if grad_accum_mode != 'sum':
+            why.append(f"--grad_accum_mode {grad_accum_mode} != 'sum'")
+        if world != 1:
+            why.append(f'world {world} > 1 (graphed DP lives in bench.py '
+                       f'behind PROGEN_GRAPH_DP=1)')
+        print(f"--graph disabled ({', '.join(why)}); running eager")
 
     def make_graphed():
         from progen_amd.runtime import GraphedTrainStep
@@ -194,6 +208,8 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
             return g
         except Exception as e:  # noqa: BLE001
             if is_main:
+                import traceback
+                traceback.print_exc()
                 print(f'hipGraph capture failed ({e}); running eager')
             return None
 
