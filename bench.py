@@ -16,6 +16,7 @@ Rank 0 prints ONE JSON line with the whole-job aggregate tokens/sec.
 
 import argparse
 import json
+import math
 import os
 import time
 
@@ -160,9 +161,17 @@ def main():
 
     t0 = time.perf_counter()
     for _ in range(args.steps):
-        step()
+        last_loss = step()
     barrier_sync()
     elapsed = time.perf_counter() - t0
+    # read AFTER the timed region: a silent NaN trajectory would make the
+    # throughput number meaningless as training evidence
+    final_loss = float(ddp.all_reduce_scalar(last_loss.detach()).item())
+    if rank == 0 and not math.isfinite(final_loss):
+        import sys
+        print(f"[bench] WARNING: final loss is {final_loss} — the step "
+              f"work is still representative but the trajectory is not",
+              file=sys.stderr)
 
     # MAX over ranks
     if world > 1:
@@ -189,6 +198,7 @@ def main():
             "dtype": ("bf16+fp8" if args.fp8 else
                       "bf16" if dtype == torch.bfloat16 else "fp32"),
             "data": "synthetic",
+            "final_loss": round(final_loss, 4),
             "config": {
                 "model": "ProGen-1.2B" if args.model == "progen-1.2b" else args.model,
                 "global_batch": B * world,

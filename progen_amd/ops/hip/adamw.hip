@@ -15,8 +15,19 @@
 // correction — required for hipGraph capture (a host-baked step would be
 // frozen at capture time). step_inc_kernel runs just before on the same
 // stream.
-__global__ void step_inc_kernel(int* step) {
-  if (threadIdx.x == 0 && blockIdx.x == 0) *step += 1;
+// Non-finite-gradient skip (GradScaler semantics): inf/NaN grads make
+// the global norm inf (-> clip coef 0) or NaN (-> coef NaN); such a
+// step is SKIPPED outright — no moment update, no param update, no
+// step-count advance — instead of poisoning the master weights with
+// inf*0 = NaN. Healthy steps always have coef in (0, 1], so this is a
+// no-op on them. (Also the mitigation for the open graphed-replay
+// corruption issue: profiles/r02_graphed_nan_investigation.md.)
+__device__ __forceinline__ bool step_ok(float clip) {
+  return clip > 0.f && !__builtin_isinf(clip) && !__builtin_isnan(clip);
+}
+
+__global__ void step_inc_kernel(int* step, const float* clip_coef) {
+  if (threadIdx.x == 0 && blockIdx.x == 0 && step_ok(*clip_coef)) *step += 1;
 }
 
 template <bool IS_BF16>
@@ -32,6 +43,7 @@ __global__ __launch_bounds__(ADAMW_BLOCK) void fused_adamw_kernel(
   // [lo, hi) slice of the flat space; chunk tables are pre-clipped to
   // that range and the fp32 state is indexed at (i - shard_off).
   const float clip = *clip_coef;
+  if (!step_ok(clip)) return;  // skip the whole update (see step_ok)
   const int step = *step_ptr;
   const float bc1 = 1.f - powf(b1, (float)step);
   const float bc2 = 1.f - powf(b2, (float)step);
@@ -114,7 +126,7 @@ void fused_adamw_launch(float* master, void* params, const void* grads,
                         int* step_dev, float grad_scale,
                         const float* clip_coef, bool is_bf16,
                         long long shard_off, hipStream_t stream) {
-  step_inc_kernel<<<1, 1, 0, stream>>>(step_dev);
+  step_inc_kernel<<<1, 1, 0, stream>>>(step_dev, clip_coef);
   int grid = nchunks < 2048 ? nchunks : 2048;
   if (is_bf16)
     fused_adamw_kernel<true><<<grid, ADAMW_BLOCK, 0, stream>>>(

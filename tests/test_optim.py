@@ -137,3 +137,32 @@ def test_state_dict_roundtrip():
     np.testing.assert_allclose(opt2.master.numpy(), opt.master.numpy())
     np.testing.assert_allclose(opt2.exp_avg.numpy(), opt.exp_avg.numpy())
     assert opt2.step_count == opt.step_count
+
+
+def test_nonfinite_grad_step_is_skipped():
+    """GradScaler semantics: inf/NaN grads -> the whole update is skipped
+    (no master/moment/param change, no step-count advance) instead of
+    poisoning the weights with inf*0 = NaN. Parity with the fused
+    kernel's step_ok guard (ops/hip/adamw.hip)."""
+    m1, _ = _models()
+    opt = ProGenAdamW(m1, lr=1e-3, max_grad_norm=0.5)
+    data = torch.randint(0, 32, (2, 33))
+    compute_loss(m1, data).backward()
+    opt.space.flat_grad[7] = float("inf")
+    before = opt.master.clone()
+    opt.step()
+    assert opt.step_count == 0
+    np.testing.assert_array_equal(opt.master.numpy(), before.numpy())
+    assert torch.isfinite(opt.master).all()
+    assert opt.exp_avg.abs().sum().item() == 0.0
+    # NaN grads skip too
+    opt.space.flat_grad[7] = float("nan")
+    opt.step()
+    np.testing.assert_array_equal(opt.master.numpy(), before.numpy())
+    # and a healthy step afterwards applies normally
+    opt.zero_grad()
+    compute_loss(m1, data).backward()
+    opt.step()
+    assert opt.step_count == 1
+    assert not torch.equal(opt.master, before)
+    assert torch.isfinite(opt.master).all()

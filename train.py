@@ -273,6 +273,15 @@ def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
         # NOTE: like the reference (train.py:192), the logged loss is the
         # last micro-batch's
         loss_val = ddp.all_reduce_scalar(loss).item()
+        if graphed is not None and loss_val != loss_val:
+            # tripwire for the open graphed-replay NaN issue
+            # (profiles/r02_graphed_nan_investigation.md): the eager path
+            # is trajectory-stable on the same data — fail loudly instead
+            # of silently training on NaNs
+            raise RuntimeError(
+                f'loss went NaN at step {i} under --graph; re-run without '
+                f'--graph (eager path). Known open issue: see '
+                f'profiles/r02_graphed_nan_investigation.md')
         if is_main:
             toks_per_sec = effective_batch_size * seq_len / step_times[-1]
             print(f'loss: {loss_val}')
