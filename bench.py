@@ -100,16 +100,21 @@ def main():
     data[:, -8:] = 0  # pad tail so the EOS-mask path is exercised
 
     graphed = None
-    # hipGraph step: pure-replay only (profiles/r01_graph_interleave_bug.md).
-    # The DP step (incl. the RCCL all-reduce) is captured BY DEFAULT in the
-    # bench: the collective sequence is one fixed-size all-reduce per
-    # replay, symmetric across ranks by construction, and capture failure
-    # on any rank degrades every rank to eager via the MIN-reduce
-    # agreement below. PROGEN_GRAPH_DP=0 is the kill-switch (falls back
-    # to eager bucketed-overlap DP, the gloo-tested default of train.py).
-    os.environ.setdefault("PROGEN_GRAPH_DP", "1")
-    want_graph = world == 1 or os.environ.get("PROGEN_GRAPH_DP") == "1"
-    if on_gpu and want_graph and os.environ.get("PROGEN_NO_GRAPH") != "1":
+    # hipGraph step: OPT-IN via PROGEN_GRAPH=1 since the late-r2 findings
+    # (profiles/r02_graphed_nan_investigation.md): the graph's measured
+    # launch-overhead win at the flagship config is ~0% once the
+    # trajectory is finite, multi-step replay has an open probabilistic
+    # corruption issue (mitigated by the AdamW non-finite skip guard),
+    # and the r1 "graph is 12% faster" delta turned out to be the
+    # NaN-poisoned weights clocking the power-capped chip higher. Eager
+    # is the trajectory-honest default. Under world>1 the opt-in also
+    # captures the RCCL all-reduce (one fixed-size collective per
+    # replay, symmetric by construction); capture failure on any rank
+    # degrades every rank to eager via the MIN-reduce agreement below.
+    want_graph = os.environ.get("PROGEN_GRAPH") == "1"
+    if want_graph:
+        os.environ.setdefault("PROGEN_GRAPH_DP", "1")
+    if on_gpu and want_graph:
         from progen_amd.runtime import GraphedTrainStep
         try:
             graphed = GraphedTrainStep(module, optim, ddp, B, N, device)
