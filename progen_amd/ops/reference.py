@@ -175,7 +175,10 @@ def local_attention(
     sim = torch.einsum("bhwid,bhwjd->bhwij", q, k) * scale
 
     mask = torch.ones(wsz, 2 * wsz, dtype=torch.bool, device=qkv.device).tril(wsz)
-    sim = torch.where(mask, sim, torch.tensor(ATTN_MASK_VALUE, dtype=sim.dtype, device=sim.device))
+    # masked_fill with a python scalar (not torch.tensor(...): that is an
+    # H2D op, illegal under hipGraph capture — needed by the
+    # PROGEN_FORCE_EAGER-in-graph bisect path of the replay investigation)
+    sim = sim.masked_fill(~mask, ATTN_MASK_VALUE)
 
     sim = sim - sim.amax(dim=-1, keepdim=True).detach()
     attn = sim.softmax(dim=-1)

@@ -207,13 +207,19 @@ class ProGenAdamW:
     def _step_eager(self, grad_scale: float) -> None:
         g32 = self.space.flat_grad.float() * grad_scale
         coef = self._clip_coef(g32, grad_scale)
-        if self.max_grad_norm is not None and not (
-                bool(torch.isfinite(coef)) and float(coef) > 0.0):
-            # inf/NaN gradients: skip the step outright (GradScaler
-            # semantics; parity with the fused kernel's step_ok guard)
-            self.step_count -= 1
-            return
-        g32 *= coef
+        if self.max_grad_norm is not None:
+            if g32.is_cuda and torch.cuda.is_current_stream_capturing():
+                # capture-clean variant (no host branch) for the
+                # FORCE_EAGER-in-graph bisect path: zero non-finite
+                # updates instead of skipping the step
+                g32 = (g32 * coef).nan_to_num(0.0, 0.0, 0.0)
+            elif not (bool(torch.isfinite(coef)) and float(coef) > 0.0):
+                # inf/NaN gradients: skip the step outright (GradScaler
+                # semantics; parity with the fused kernel's step_ok guard)
+                self.step_count -= 1
+                return
+            else:
+                g32 *= coef
         self.master.sub_(self._adamw_update(g32))
         if self.is_low_precision:
             self.space.flat.copy_(self.master.to(self.space.flat.dtype))
