@@ -503,3 +503,44 @@ def test_colsum_matches_torch():
         got = dispatch.ext().colsum(dy)
         want = dy.float().sum(dim=0)
         assert rel_err(got, want) < 1e-3, (R, C)
+
+
+def test_fused_adamw_nonfinite_skip():
+    """Kernel-side step_ok guard (ops/hip/adamw.hip): an inf/NaN grad
+    buffer must skip the whole update — params, master, moments and the
+    DEVICE step counter all unchanged — then a healthy step applies.
+    Mitigation for the open graphed-replay corruption issue
+    (profiles/r02_graphed_nan_investigation.md)."""
+    from progen_amd import ProGenBase, ProGenConfig
+    from progen_amd.optim import ProGenAdamW
+
+    cfg = ProGenConfig(num_tokens=64, dim=64, seq_len=64, depth=2,
+                       window_size=64, global_mlp_depth=1, heads=1, dim_head=64)
+    torch.manual_seed(9)
+    m = ProGenBase(cfg).to(device=dev(), dtype=torch.bfloat16)
+    o = ProGenAdamW(m, lr=1e-3, max_grad_norm=0.5)
+
+    torch.manual_seed(10)
+    fake = torch.randn(o.space.flat_grad.shape, device=dev())
+    for bad in (float("inf"), float("nan")):
+        o.space.flat_grad.copy_(fake.to(o.space.flat_grad.dtype))
+        o.space.flat_grad[3] = bad
+        master0 = o.master.clone()
+        flat0 = o.space.flat.clone()
+        v0 = o.exp_avg_sq.clone()
+        step0 = int(o.step_dev.item())
+        o.step()
+        torch.cuda.synchronize()
+        assert int(o.step_dev.item()) == step0, bad
+        assert torch.equal(o.master, master0), bad
+        assert torch.equal(o.space.flat, flat0), bad
+        assert torch.equal(o.exp_avg_sq, v0), bad
+
+    # healthy step still applies and advances the device counter
+    o.space.flat_grad.copy_(fake.to(o.space.flat_grad.dtype))
+    master0 = o.master.clone()
+    o.step()
+    torch.cuda.synchronize()
+    assert int(o.step_dev.item()) == 1
+    assert not torch.equal(o.master, master0)
+    assert torch.isfinite(o.master).all()
