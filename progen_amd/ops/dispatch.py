@@ -56,11 +56,22 @@ def has_ext() -> bool:
     return _EXT is not None
 
 
-def use_hip(t) -> bool:
-    """True when tensor t lives on a ROCm GPU (→ HIP kernels are mandatory)."""
+def use_hip(t, op: str = None) -> bool:
+    """True when tensor t lives on a ROCm GPU (→ HIP kernels are mandatory).
+
+    Debug-only escape hatches (never the default on GPU):
+    - PROGEN_FORCE_EAGER=1 routes EVERY op to its torch reference;
+    - PROGEN_EAGER_OPS="attn,sgu" routes only the named ops eager —
+      the per-kernel bisect lever for the graphed-replay investigation
+      (profiles/r02_graphed_nan_investigation.md). Tags: ln, attn, glu,
+      sgu, ce, adamw.
+    """
     if not t.is_cuda:
         return False
     if os.environ.get("PROGEN_FORCE_EAGER") == "1":
-        # debugging escape hatch only; never the default on GPU
         return False
+    if op is not None:
+        eager = os.environ.get("PROGEN_EAGER_OPS")
+        if eager and op in {x.strip() for x in eager.split(",")}:
+            return False
     return True

@@ -42,7 +42,7 @@ def ln_shift(x: torch.Tensor, weight: torch.Tensor, shift: bool = True,
              eps: float = 1e-5) -> torch.Tensor:
     """LN (scale-only, reference: progen.py:22) + optional token shift
     (reference: progen.py:43-46), fused on GPU."""
-    if dispatch.use_hip(x):
+    if dispatch.use_hip(x, "ln"):
         return _LnShiftFn.apply(x, weight, shift, eps)
     return reference.ln_shift(x, weight, shift, eps)
 
@@ -80,7 +80,7 @@ def ln_shift_res(x: torch.Tensor, res, weight: torch.Tensor,
     """Residual add + LN + shift in one pass: returns (y, s) with
     s = x + res (the updated residual stream) and y = ln_shift(s).
     ``res=None`` degenerates to plain ln_shift with s = x."""
-    if dispatch.use_hip(x):
+    if dispatch.use_hip(x, "ln"):
         if res is None:
             return _LnShiftFn.apply(x, weight, shift, eps), x
         return _LnShiftResFn.apply(x, res.contiguous(), weight, shift, eps)
@@ -137,7 +137,7 @@ def local_attention(qkv: torch.Tensor, sin: torch.Tensor, cos: torch.Tensor,
     ``halo``: optional (B, wsz, 2*H*dh) ROTATED [k|v] band that replaces
     window 0's zero lookback (context parallelism, parallel/cp.py); its
     gradient (same shape/space) is returned to the autograd graph."""
-    if dispatch.use_hip(qkv):
+    if dispatch.use_hip(qkv, "attn"):
         return _LocalAttnFn.apply(qkv, sin, cos, heads, window_size, halo)
     assert halo is None, "halo is a kernel-path (GPU) feature"
     return reference.local_attention(qkv, sin, cos, heads, window_size)
@@ -164,7 +164,7 @@ class _GluGeluFn(torch.autograd.Function):
 
 def glu_gelu(h: torch.Tensor) -> torch.Tensor:
     """x, gate = split(h, 2); x * gelu(gate)  (reference: progen.py:139-141)."""
-    if dispatch.use_hip(h):
+    if dispatch.use_hip(h, "glu"):
         return _GluGeluFn.apply(h)
     return reference.glu_gelu(h)
 
@@ -185,7 +185,7 @@ class _GeluFn(torch.autograd.Function):
 
 
 def gelu(h: torch.Tensor) -> torch.Tensor:
-    if dispatch.use_hip(h):
+    if dispatch.use_hip(h, "glu"):
         return _GeluFn.apply(h)
     return reference.gelu(h)
 
@@ -221,7 +221,7 @@ def cross_entropy(logits: torch.Tensor, targets: torch.Tensor,
     """Masked CE with first-pad-as-EOS (reference: utils.py:45-59); the
     per-sequence masked mean then batch mean reduction order is preserved
     (reference: utils.py:67,75-76)."""
-    if dispatch.use_hip(logits):
+    if dispatch.use_hip(logits, "ce"):
         targets = targets.long().contiguous()
         nll = _CERowFn.apply(logits.contiguous(), targets)
         mask = targets != ignore_index
@@ -290,7 +290,7 @@ def sgu_gate(x: torch.Tensor, norm_weight: torch.Tensor,
     spatial-matmul kernels (ops/hip/sgu.hip — the tril mask is baked into
     the tile iteration). Sequences shorter than 256 (toy configs) use the
     hipBLASLt composite path."""
-    if dispatch.use_hip(x):
+    if dispatch.use_hip(x, "sgu"):
         xa, gate = x.chunk(2, dim=-1)
         gate_ln = ln_shift(gate.contiguous(), norm_weight, shift=False, eps=eps)
         n = x.shape[1]

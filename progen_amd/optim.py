@@ -174,7 +174,7 @@ class ProGenAdamW:
             WgradQueue.sync()  # join side-stream wgrads before the update
         self.step_count += 1
         g = self.space.flat_grad
-        if dispatch.use_hip(g):
+        if dispatch.use_hip(g, "adamw"):
             self._step_hip(grad_scale)
         else:
             self._step_eager(grad_scale)

@@ -273,3 +273,24 @@ def test_sgu_causality_and_bias_init():
     m = 2
     want = xa[m] * ((W[m, : m + 1].unsqueeze(0) @ ln[: m + 1]).squeeze(0) + 1.0)
     np.testing.assert_allclose(out[0, m].numpy(), want.numpy(), atol=1e-8)
+
+
+def test_eager_ops_env_gates_dispatch(monkeypatch):
+    """PROGEN_EAGER_OPS routes only the named ops to the torch reference
+    (the per-kernel bisect lever for the graphed-replay investigation)."""
+    from progen_amd.ops import dispatch
+
+    class FakeCuda:
+        is_cuda = True
+
+    t = FakeCuda()
+    monkeypatch.delenv("PROGEN_FORCE_EAGER", raising=False)
+    monkeypatch.setenv("PROGEN_EAGER_OPS", "attn, sgu")
+    assert dispatch.use_hip(t, "attn") is False
+    assert dispatch.use_hip(t, "sgu") is False
+    assert dispatch.use_hip(t, "ln") is True
+    assert dispatch.use_hip(t) is True
+    monkeypatch.delenv("PROGEN_EAGER_OPS")
+    assert dispatch.use_hip(t, "attn") is True
+    monkeypatch.setenv("PROGEN_FORCE_EAGER", "1")
+    assert dispatch.use_hip(t, "ln") is False
