@@ -25,6 +25,14 @@ opt-in).
 
 Warmup runs on a side stream; optimizer/param state perturbed by warmup
 and capture is snapshotted and restored.
+
+KNOWN OPEN ISSUE (late r2): even pure back-to-back replays corrupt
+gradients probabilistically (~25%/replay at ProGen-small scale) on this
+stack — most likely the same runtime bug class as the interleave
+corruption above. The fused AdamW's non-finite step skip keeps the
+trajectory finite, and eager is the default everywhere; full
+investigation and the bisect plan live in
+profiles/r02_graphed_nan_investigation.md.
 """
 
 from __future__ import annotations

@@ -75,11 +75,14 @@ def _wandb(wandb_off):
 @click.option('--new', default=False, is_flag=True)
 @click.option('--max_steps', default=0, help='stop after N effective batches (0 = full epoch)')
 @click.option('--graph/--no-graph', default=False,
-              help='hipGraph-capture the training step (single-GPU, pure replay '
-                   'only: disables in-loop validation/sampling — eager kernels '
-                   'between replays corrupt replay state on this ROCm stack; '
-                   'checkpoints still happen: the graph is dropped and '
-                   're-captured around each save)')
+              help='hipGraph-capture the training step (single-GPU, '
+                   'grad_accum_every=1, pure replay). Checkpoints, validation '
+                   'AND sampling still run in-loop: the graph is dropped and '
+                   're-captured around them. KNOWN OPEN ISSUE: multi-step '
+                   'replay can corrupt gradients on this ROCm stack (the '
+                   'AdamW non-finite skip guard contains it, and the loop '
+                   'aborts on a NaN loss) — eager is the safe default; see '
+                   'profiles/r02_graphed_nan_investigation.md')
 @click.option('--yes', default=False, is_flag=True, help='skip the --new confirmation prompt')
 def main(seed, batch_size, grad_accum_every, learning_rate, weight_decay,
          data_parallel, max_grad_norm, validate_every, sample_every,
