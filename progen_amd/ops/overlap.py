@@ -64,9 +64,11 @@ class WgradQueue:
 
 
 def _fast_dbias(dy2: torch.Tensor) -> torch.Tensor:
-    """Column sum of the output grad. torch's bf16 dim-0 reduce measured
-    ~50x the traffic bound on these shapes (round-2 profile); the colsum
-    kernel is one vectorized pass with fp32 accumulation."""
+    """Column sum of the output grad: one vectorized pass with fp32
+    accumulation. (The round-2 "torch reduce is 50x traffic" reading was
+    a profiler misattribution — measured properly, torch's reduce is
+    competitive at these shapes; this whole path only runs under the
+    off-by-default PROGEN_OVERLAP_WGRAD side-stream mode.)"""
     if dy2.is_cuda and dy2.dtype == torch.bfloat16 and dy2.shape[1] % 8 == 0:
         from . import dispatch
         return dispatch.ext().colsum(dy2)
