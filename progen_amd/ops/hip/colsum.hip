@@ -1,10 +1,11 @@
 // Column sum for bias gradients: out[c] = sum_r dy[r][c], fp32 accum.
 //
-// The round-2 step profile showed torch's bf16 dim-0 reduce taking
-// ~15 ms/step across the ~105 per-step dbias reductions — ~50x the
-// traffic bound (the reduction axis is the MAJOR axis, which
-// at::native::reduce_kernel handles a row-chunk at a time without
-// vectorizing the bf16 loads). Here: each thread owns 8 consecutive
+// (The initial round-2 reading that torch's reduce was "~50x the
+// traffic bound / ~15 ms per step" was a profiler misattribution — a
+// sloppy kernel-name match pulled in unrelated reduce_kernel launches.
+// Measured properly, torch's bf16 dim-0 reduce is competitive at these
+// shapes; this kernel only serves the off-by-default
+// PROGEN_OVERLAP_WGRAD side-stream path.) Each thread owns 8 consecutive
 // columns (one bf16x8 load per row), blocks stride the rows, partials
 // combine with fp32 atomics (gridDim.y partials per column).
 
