@@ -22,7 +22,10 @@ Endpoints:
 Requests are serialized through one lock (single model instance; the
 batch dimension inside ONE request is where serving throughput comes
 from — 2.9k tok/s aggregate at batch 128 on ProGen-1.2B, see
-profiles/r02_fp8_and_decode.md).
+profiles/r02_fp8_and_decode.md). With --graph the per-token step is
+captured per request (the decoder manages its own capture); the capture
+cost amortizes over the generated length — persistent cross-request
+capture keyed on batch shape is the next step if request shapes repeat.
 """
 
 import threading
