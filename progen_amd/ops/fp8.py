@@ -3,11 +3,13 @@
 
 MI355X doubles MFMA throughput at fp8 (≈5 PFLOP/s dense vs ≈2.5 bf16),
 so the 6B projections move to e4m3 GEMMs with per-tensor scales and fp32
-accumulation. This module is the dtype plumbing — amax-based scaling,
-quantize/dequantize, and a simulated-fp8 matmul used by the CPU tests to
-pin the quantization error envelope. The GPU GEMM itself (hipBLASLt fp8
-or `mfma_f32_16x16x32_fp8_fp8` tiles in the hand-written kernels) is
-round-2 work and is gated on measurement (TODO.md).
+accumulation. This module holds both halves: the dtype plumbing —
+amax-based scaling, quantize/dequantize, a simulated-fp8 matmul the CPU
+tests pin the error envelope against — and (round 2) the REAL GEMM path:
+hipBLASLt e4m3 via torch._scaled_mm with fused quantize kernels
+(ops/hip/fp8_quant.hip), measured 1.77-2.06x bf16 on the 1.2B/6B
+projection shapes (profiles/r02_fp8_and_decode.md; end-to-end wiring is
+PROGEN_FP8=1, currently off by default — measured negative, see TODO.md).
 
 E4M3 facts used here (OCP FP8, the `torch.float8_e4m3fn` variant):
 max normal 448, no inf (S.1111.111 is NaN), ~3-bit mantissa → worst-case
