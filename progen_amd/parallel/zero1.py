@@ -18,7 +18,7 @@ Design (v1, opt-in via PROGEN_ZERO1=1 in train.py/bench.py):
 A reduce-scatter of the gradients (instead of all-reduce) would also
 halve the gradient communication — that needs bucket/shard alignment in
 the DDP overlap path and RCCL reduce_scatter (gloo, used by the CPU
-tests, lacks it), so it stays on the round-2 list (TODO.md).
+tests, lacks it), so it stays on the future list (TODO.md).
 """
 
 from __future__ import annotations
