@@ -80,3 +80,11 @@ def test_greedy_matches_direct_decoder():
     with torch.no_grad():
         want = sample_cached_batch(module, [prime], 32, top_k=None)
     assert r["tokens"][0] == want[0][prime.shape[0]:].tolist()
+
+
+def test_generate_empty_prime_is_unconditional():
+    client = TestClient(_app())
+    r = client.post("/generate", json={"prime": "", "num_tokens": 16,
+                                       "seed": 3})
+    assert r.status_code == 200, r.text
+    assert len(r.json()["tokens"][0]) == 15  # BOS + 15 generated
