@@ -192,9 +192,10 @@ class ProGenAdamW:
         # optax clip_by_global_norm: g * max_norm / max(norm, max_norm)
         return self.max_grad_norm / torch.clamp_min(norm, self.max_grad_norm)
 
-    def _adamw_update(self, g32: torch.Tensor) -> torch.Tensor:
+    def _adamw_update(self, g32: torch.Tensor, t=None) -> torch.Tensor:
         b1, b2 = self.betas
-        t = self.step_count
+        if t is None:
+            t = self.step_count
         self.exp_avg.mul_(b1).add_(g32, alpha=1 - b1)
         self.exp_avg_sq.mul_(b2).addcmul_(g32, g32, value=1 - b2)
         mhat = self.exp_avg / (1 - b1 ** t)
@@ -207,6 +208,14 @@ class ProGenAdamW:
     def _step_eager(self, grad_scale: float) -> None:
         g32 = self.space.flat_grad.float() * grad_scale
         coef = self._clip_coef(g32, grad_scale)
+        # on GPU the bias-correction step count lives on device (as in
+        # the fused kernel) so a captured eager step stays correct under
+        # hipGraph replay (the FORCE_EAGER/PROGEN_EAGER_OPS bisect path —
+        # a python step_count would be frozen at its capture-time value)
+        t_dev = None
+        if self.step_dev is not None and g32.is_cuda:
+            self.step_dev += 1
+            t_dev = self.step_dev.float()
         if self.max_grad_norm is not None:
             if g32.is_cuda and torch.cuda.is_current_stream_capturing():
                 # capture-clean variant (no host branch) for the
@@ -217,10 +226,12 @@ class ProGenAdamW:
                 # inf/NaN gradients: skip the step outright (GradScaler
                 # semantics; parity with the fused kernel's step_ok guard)
                 self.step_count -= 1
+                if t_dev is not None:
+                    self.step_dev -= 1
                 return
             else:
                 g32 *= coef
-        self.master.sub_(self._adamw_update(g32))
+        self.master.sub_(self._adamw_update(g32, t_dev))
         if self.is_low_precision:
             self.space.flat.copy_(self.master.to(self.space.flat.dtype))
 
