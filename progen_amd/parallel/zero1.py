@@ -91,6 +91,14 @@ class Zero1AdamW(ProGenAdamW):
     def _step_eager(self, grad_scale: float) -> None:
         g_full = self.space.flat_grad.float() * grad_scale
         coef = self._clip_coef(g_full, grad_scale)  # full-grad norm: no comm
+        if self.max_grad_norm is not None and not (
+                bool(torch.isfinite(coef)) and float(coef) > 0.0):
+            # inf/NaN gradients: skip the whole step (GradScaler
+            # semantics; the fused kernel's step_ok guard does the same
+            # on the HIP path). The coef comes from the FULL grad norm,
+            # identical on every rank, so all ranks skip together.
+            self.step_count -= 1
+            return
         g32 = g_full[self.lo:self.hi] * coef
         self.master.sub_(self._adamw_update(g32))
         if self.is_low_precision:

@@ -243,3 +243,53 @@ def test_zero1_state_roundtrip_and_guard():
     for p in ps:
         p.join(timeout=60)
     assert all(msg == "ok" for _, msg in results), results
+
+
+def _skip_worker(rank, world, port, q):
+    import torch.distributed as dist
+    os.environ.update(MASTER_ADDR="127.0.0.1", MASTER_PORT=str(port))
+    dist.init_process_group("gloo", rank=rank, world_size=world)
+    from progen_amd.parallel.zero1 import Zero1AdamW
+    try:
+        torch.manual_seed(17)
+        m = torch.nn.Linear(3, 4).double()
+        opt = Zero1AdamW(m, lr=1e-2, max_grad_norm=0.5)
+        x = torch.randn(8, 3, dtype=torch.float64,
+                        generator=torch.Generator().manual_seed(5))
+        (m(x) ** 2).mean().backward()
+        opt.space.flat_grad[1] = float("inf")
+        before = opt.space.flat.detach().clone()
+        opt.step()  # must skip on every rank together (full-grad norm)
+        assert torch.equal(opt.space.flat, before)
+        assert opt.exp_avg.abs().sum().item() == 0.0
+        # healthy step afterwards applies
+        opt.zero_grad()
+        (m(x) ** 2).mean().backward()
+        opt.step()
+        assert not torch.equal(opt.space.flat, before)
+        assert torch.isfinite(opt.space.flat).all()
+        q.put((rank, "ok"))
+    except Exception as e:
+        import traceback
+        q.put((rank, f"fail: {e}\n{traceback.format_exc()[-1000:]}"))
+    dist.barrier()
+    dist.destroy_process_group()
+
+
+@pytest.mark.timeout(120)
+def test_zero1_nonfinite_grad_step_is_skipped():
+    """Parity with the base optimizer's GradScaler-style skip: inf grads
+    leave params/moments untouched on every rank (the clip coef comes
+    from the full-grad norm, so the skip decision is rank-consistent)."""
+    world = 2
+    port = _free_port()
+    ctx = mp.get_context("spawn")
+    q = ctx.Queue()
+    ps = [ctx.Process(target=_skip_worker, args=(r, world, port, q))
+          for r in range(world)]
+    for p in ps:
+        p.start()
+    results = [q.get(timeout=90) for _ in range(world)]
+    for p in ps:
+        p.join(timeout=60)
+    assert all(msg == "ok" for _, msg in results), results
