@@ -238,7 +238,9 @@ class ProGenAdamW:
     def _apply_every_micro(self) -> None:
         """Reference apply_every semantics: moments advance per micro-batch,
         updates accumulate, applied every k micro-batches
-        (reference: train.py:117-121,185-191)."""
+        (reference: train.py:117-121,185-191). Deliberately NO
+        non-finite skip here: optax apply_every has none either, and
+        this mode exists for reference parity."""
         self.step_count += 1
         g32 = self.space.flat_grad.float()
         g32 *= self._clip_coef(g32)
