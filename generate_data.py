@@ -104,8 +104,21 @@ def fasta_to_tmp_files(config):
     print(f'wrote {written} tmp sequences from {count} fasta records')
 
 
-def files_to_tfrecords(config):
-    """(reference: generate_data.py:107-153)"""
+def _write_shard(task):
+    """One output TFRecord file (picklable worker for --workers > 1)."""
+    out_path, index_list, tmp_names = task
+    with with_tfrecord_writer(out_path) as write:
+        for index in index_list:
+            with gzip.open(tmp_names[index], 'rb') as f:
+                write(f.read())
+    return out_path
+
+
+def files_to_tfrecords(config, workers: int = 1):
+    """(reference: generate_data.py:107-153). Output shards are
+    independent, so ``workers > 1`` writes them in parallel — the file
+    set and every byte in it are identical to the serial run (each shard
+    is still written sequentially by one process)."""
     filenames = sorted(TMP_DIR.glob('**/*'), key=lambda p: int(p.name))
     num_samples = len(filenames)
     num_valids = ceil(config['fraction_valid_data'] * num_samples)
@@ -121,29 +134,37 @@ def files_to_tfrecords(config):
     write_to_path = Path(write_to)
     clear_directory_(write_to_path)
 
+    tasks = []
     for seq_type, seqs in (('train', train_seqs), ('valid', valid_seqs)):
         if seqs.shape[0] == 0:
             continue
         num_split = ceil(seqs.shape[0] / num_sequences_per_file)
         for file_index, indices in enumerate(np.array_split(seqs, num_split)):
-            num_sequences = len(indices)
-            name = f'{file_index}.{num_sequences}.{seq_type}.tfrecord.gz'
-            with with_tfrecord_writer(str(write_to_path / name)) as write:
-                for index in indices:
-                    with gzip.open(filenames[index], 'rb') as f:
-                        write(f.read())
-            print(f'wrote {write_to_path / name}')
+            name = f'{file_index}.{len(indices)}.{seq_type}.tfrecord.gz'
+            tasks.append((str(write_to_path / name), indices.tolist(),
+                          filenames))
+    if workers > 1 and len(tasks) > 1:
+        from multiprocessing import Pool
+        with Pool(min(workers, len(tasks))) as pool:
+            for out in pool.imap_unordered(_write_shard, tasks):
+                print(f'wrote {out}')
+    else:
+        for task in tasks:
+            print(f'wrote {_write_shard(task)}')
 
 
 @click.command()
 @click.option('--data_dir', default='./configs/data')
 @click.option('--name', default='default')
-def main(data_dir, name):
+@click.option('--workers', default=1,
+              help='parallel TFRecord shard writers (output is byte-'
+                   'identical to the serial run)')
+def main(data_dir, name, workers):
     config_path = Path(data_dir) / f'{name}.toml'
     assert config_path.exists(), f'config does not exist at {config_path}'
     config = tomllib.loads(config_path.read_text())
     fasta_to_tmp_files(config)
-    files_to_tfrecords(config)
+    files_to_tfrecords(config, workers=workers)
 
 
 if __name__ == '__main__':
