@@ -84,6 +84,8 @@ def create_app(module: ProGenBase, cfg: ProGenConfig, *,
         length = req.num_tokens or cfg.seq_len
         if not 1 <= length <= cfg.seq_len:
             raise HTTPException(400, f"num_tokens must be in [1, {cfg.seq_len}]")
+        if req.top_k is not None and req.top_k < 0:
+            raise HTTPException(400, "top_k must be >= 0 (0/null = greedy)")
         top_k = req.top_k if req.top_k else None
         # explicit BOS column, as the samplers' add_bos does (the byte
         # tokenizer reserves 0 for BOS/pad)

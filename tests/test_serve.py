@@ -61,6 +61,9 @@ def test_generate_validation_errors():
     assert client.post("/generate",
                        json={"prime": "X" * 40,
                              "num_tokens": 16}).status_code == 400
+    assert client.post("/generate",
+                       json={"prime": "# M", "num_tokens": 16,
+                             "top_k": -2}).status_code == 400
 
 
 def test_greedy_matches_direct_decoder():
