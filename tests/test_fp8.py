@@ -53,3 +53,27 @@ def test_sim_gemm_scale_invariance():
     r1 = fp8.matmul_sim_fp8(a, b)
     r2 = fp8.matmul_sim_fp8(a * 1000, b * 0.001)
     torch.testing.assert_close(r1, r2, rtol=1e-5, atol=1e-5)
+
+
+def test_cached_weight_q_version_invalidation():
+    """The quantized-weight cache returns the same objects while the
+    weight is untouched (grad-accum micros / inference) and requantizes
+    after any in-place update (optimizer step through the flat view)."""
+    import torch
+
+    from progen_amd.ops.fp8 import _cached_weight_q, dequantize
+
+    w = torch.nn.Parameter(torch.randn(32, 16))
+    q1, s1 = _cached_weight_q(w, transposed=False)
+    q2, s2 = _cached_weight_q(w, transposed=False)
+    assert q1 is q2 and s1 is s2  # cache hit
+    qt1, _ = _cached_weight_q(w, transposed=True)
+    qt2, _ = _cached_weight_q(w, transposed=True)
+    assert qt1 is qt2
+    # in-place update through a VIEW (how the flat optimizer writes)
+    with torch.no_grad():
+        w.view(-1).mul_(2.0)
+    q3, s3 = _cached_weight_q(w, transposed=False)
+    assert q3 is not q1
+    err = (dequantize(q3, s3) - w.detach()).abs().max()
+    assert err < 0.07 * w.detach().abs().max()  # e4m3 envelope
