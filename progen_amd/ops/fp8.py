@@ -120,7 +120,10 @@ def _cached_weight_q(weight: torch.Tensor, transposed: bool):
     weights per projection call. The optimizer updates params in place
     through the flat buffer, and views share the version counter, so
     any real update invalidates the cache. Bypassed during hipGraph
-    capture (a baked cache would freeze the weights into the graph)."""
+    capture (a baked cache would freeze the weights into the graph).
+    Memory: +1 byte/param per cached form (normal + transposed = +2
+    bytes/param while PROGEN_FP8=1 — vs the 2-byte bf16 weights; fine
+    in 288 GB)."""
     if weight.is_cuda and torch.cuda.is_current_stream_capturing():
         ent = None
     else:
