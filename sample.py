@@ -29,6 +29,8 @@ load_dotenv()  # reference: sample.py:1-2
               help='with --cached on GPU: replay the per-token step as one '
                    'captured hipGraph (~2x the eager cached step)')
 def main(seed, checkpoint_path, prime, fast, cached, graph):
+    from progen_amd.tuning import enable_tuned_gemms
+    enable_tuned_gemms()
     _, get_last_checkpoint, _ = get_checkpoint_fns(checkpoint_path)
     last_checkpoint = get_last_checkpoint()
     if last_checkpoint is None:

@@ -145,6 +145,8 @@ def load_module(checkpoint_path: str):
                    "captured hipGraph (2.1x the eager cached step)")
 def main(checkpoint_path, host, port, graph):
     load_dotenv()
+    from progen_amd.tuning import enable_tuned_gemms
+    enable_tuned_gemms()
     import uvicorn
     module, cfg, meta = load_module(checkpoint_path)
     app = create_app(module, cfg, graph=graph, meta=meta)
