@@ -91,3 +91,29 @@ def test_generate_empty_prime_is_unconditional():
                                        "seed": 3})
     assert r.status_code == 200, r.text
     assert len(r.json()["tokens"][0]) == 15  # BOS + 15 generated
+
+
+def test_load_module_from_checkpoint(tmp_path, monkeypatch):
+    """serve.load_module rebuilds the model from the checkpoint's stored
+    model_config (sample.py's recipe) and reports trained_sequences."""
+    from progen_amd.checkpoint import get_checkpoint_fns, tensors_to_numpy
+    from serve import load_module
+
+    cfg_kwargs = dict(num_tokens=256, dim=16, depth=2, dim_head=8, heads=2,
+                      window_size=8, seq_len=64, global_mlp_depth=1)
+    torch.manual_seed(9)
+    m = ProGenBase(ProGenConfig(**cfg_kwargs))
+    _, _, save = get_checkpoint_fns(str(tmp_path / "ckpts"))
+    save({
+        "next_seq_index": 1234,
+        "params": tensors_to_numpy(dict(m.state_dict())),
+        "optim_state": None,
+        "model_config": cfg_kwargs,
+        "run_id": None,
+    }, keep_last_n=1)
+    module, cfg, meta = load_module(str(tmp_path / "ckpts"))
+    assert cfg.seq_len == 64 and cfg.depth == 2
+    assert meta["trained_sequences"] == 1234
+    # weights actually loaded, not re-initialized
+    got = dict(module.state_dict())["embed.weight"]
+    torch.testing.assert_close(got, m.state_dict()["embed.weight"])
